@@ -1,0 +1,149 @@
+"""Flagship benchmark: Llama-3 8B training step, tokens/sec whole-job.
+
+Driver contract (BASELINE.json): tokens/sec (whole node), Llama-3-8B,
+TP=N, ZeRO-1, bf16, synthetic data, random-init weights. Strong scaling:
+the global batch is fixed as N grows (TP shards the model across the xGMI
+clique).
+
+    python bench.py --gpus N --steps K --warmup W
+    (N>1 launched by the driver via torch.distributed.run, one rank/GPU)
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=8)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--gbs", type=int, default=None, help="global batch (sequences)")
+    ap.add_argument("--seq", type=int, default=8192)
+    ap.add_argument("--layers", type=int, default=32)
+    args = ap.parse_args()
+
+    n = args.gpus
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    use_gpu = torch.cuda.is_available()
+    if world > 1:
+        assert world == n, f"WORLD_SIZE {world} != --gpus {n}"
+        if use_gpu:
+            torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl" if use_gpu else "gloo")
+
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+
+    tp = world if world > 1 else 1
+    ps.initialize_model_parallel(tensor_model_parallel_size=tp)
+    torch.manual_seed(1234)
+
+    gbs = args.gbs if args.gbs is not None else 4
+    mbs = 1
+    seq = args.seq
+    cfg = {
+        "data": {"global_batch_size": gbs, "micro_batch_size": mbs, "seq_length": seq},
+        "distributed_strategy": {
+            "tensor_model_parallel_size": tp,
+            "sequence_parallel": tp > 1,
+            "zero1": True,
+        },
+        "model": {
+            "vocab_size": 128256,
+            "hidden_size": 4096,
+            "intermediate_size": 14336,
+            "num_layers": args.layers,
+            "num_attention_heads": 32,
+            "num_kv_heads": 8,
+            "rope_theta": 500000.0,
+            "grad_clip": 1.0,
+            "optim": {"lr": 3.0e-4, "sched": {"warmup_steps": 10}},
+        },
+        "precision": {"type": "mixed_precision"},
+        "exp_manager": {"log_gradient_norm": False, "log_parameter_norm": False},
+    }
+
+    module = LlamaModule(cfg)
+    module.setup()
+    module.configure_optimizers(max_steps=args.steps + args.warmup)
+
+    vocab = cfg["model"]["vocab_size"]
+    n_micro = module.num_microbatches
+
+    def make_micro():
+        ids = torch.randint(0, vocab, (mbs, seq))
+        return {"input_ids": ids, "labels": ids.clone()}
+
+    micros = [make_micro() for _ in range(n_micro)]
+
+    def one_step():
+        module.optimizer.zero_grad()
+        module.forward_backward_step(iter(micros))
+        module.optimizer.step()
+        module.scheduler.step()
+
+    for _ in range(args.warmup):
+        one_step()
+
+    if world > 1:
+        dist.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+    if use_gpu:
+        torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    t1 = time.perf_counter()
+
+    elapsed = torch.tensor([t1 - t0], dtype=torch.float64)
+    if world > 1:
+        dist.all_reduce(elapsed, op=dist.ReduceOp.MAX)
+    ms_per_step = float(elapsed[0]) / args.steps * 1000.0
+    tokens_per_step = gbs * seq
+    tok_s = tokens_per_step / (ms_per_step / 1000.0)
+
+    if rank == 0:
+        print(
+            json.dumps(
+                {
+                    "metric": "tokens_per_sec",
+                    "value": tok_s,
+                    "unit": "tokens/s",
+                    "n_gpus": n,
+                    "steps": args.steps,
+                    "warmup": args.warmup,
+                    "ms_per_step": ms_per_step,
+                    "higher_is_better": True,
+                    "scaling": "strong",
+                    "vs_baseline": None,
+                    "dtype": "bf16" if use_gpu else "float32",
+                    "data": "synthetic",
+                    "config": {
+                        "model": "llama3-8B",
+                        "global_batch": gbs,
+                        "seq_len": seq,
+                        "parallelism": f"tp{tp}_zero1" + ("_sp" if tp > 1 else ""),
+                    },
+                }
+            ),
+            flush=True,
+        )
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,35 @@
+// Shared MFMA fragment helpers for the CDNA4 attention kernels.
+//
+// mfma_f32_16x16x32_bf16 operand maps (verified on gfx950 by
+// tests/test_gpu_kernels.py::test_mfma_layout_probe):
+//   A (16×32): lane holds row = lane&15, k = (lane>>4)*8 + j   (j = 0..7)
+//   B (32×16): lane holds col = lane&15, k = (lane>>4)*8 + j
+//   C/D (16×16): lane holds col = lane&15, row = (lane>>4)*4 + r (r = 0..3)
+#pragma once
+#include "common.h"
+
+typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
+typedef float f32x4_t __attribute__((ext_vector_type(4)));
+
+#define MFMA_16x16x32(A, B, C) \
+  __builtin_amdgcn_mfma_f32_16x16x32_bf16((A), (B), (C), 0, 0, 0)
+
+// Load an A-fragment from a row-major LDS/global tile with row stride
+// `stride` (in elements): rows = 16 consecutive, k-window of 32 at `k0`.
+DEVINL bf16x8_t load_frag_a(const __bf16* base, int stride, int k0, int lane) {
+  const __bf16* p = base + (lane & 15) * stride + k0 + (lane >> 4) * 8;
+  bf16x8_t f;
+  *(int4*)&f = *(const int4*)p;  // 8 contiguous bf16 = 16 B
+  return f;
+}
+
+// Load a B-fragment where the tile is stored TRANSPOSED as [N][K] row-major
+// (i.e. B[k][j] lives at tile[j*stride + k]): col j = lane&15 picks the row,
+// the 8 k's are contiguous. This covers K^T reads from k_lds[key][d] etc.
+DEVINL bf16x8_t load_frag_b_rowmajorT(const __bf16* base, int stride, int k0,
+                                      int lane) {
+  const __bf16* p = base + (lane & 15) * stride + k0 + (lane >> 4) * 8;
+  bf16x8_t f;
+  *(int4*)&f = *(const int4*)p;
+  return f;
+}

@@ -1,0 +1,29 @@
+#include "hip/hip_runtime.h"
+// MFMA layout probe: one wave computes C[16,16] = A[16,32] @ B[32,16]
+// using the fragment maps in attn_common.h. The GPU test compares against
+// torch.matmul — a failure means the lane→element maps are wrong and every
+// attention kernel is too (asymmetric B catches transposes).
+#include "attn_common.h"
+
+__global__ void mfma_probe_kernel(const bf16* __restrict__ A,
+                                  const bf16* __restrict__ B,
+                                  float* __restrict__ C) {
+  const int lane = threadIdx.x & 63;
+  bf16x8_t af, bf;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    af[j] = (__bf16)A[(lane & 15) * 32 + (lane >> 4) * 8 + j];
+    bf[j] = (__bf16)B[((lane >> 4) * 8 + j) * 16 + (lane & 15)];
+  }
+  f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
+  acc = MFMA_16x16x32(af, bf, acc);
+#pragma unroll
+  for (int r = 0; r < 4; ++r)
+    C[((lane >> 4) * 4 + r) * 16 + (lane & 15)] = acc[r];
+}
+
+extern "C" void launch_mfma_probe(const void* a, const void* b, void* c,
+                                  hipStream_t stream) {
+ hipLaunchKernelGGL(( mfma_probe_kernel), dim3(1), dim3(64), 0, stream, (const bf16*)a, (const bf16*)b,
+                                          (float*)c);
+}

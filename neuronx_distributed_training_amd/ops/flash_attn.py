@@ -1,0 +1,98 @@
+"""Causal flash attention, hand-written CDNA4 HIP (MFMA + LDS online softmax).
+
+Replaces the reference's NKI flash kernel contract
+(``nki_flash_attn_func``, call site modeling_llama.py:486): causal,
+bf16, GQA via kv-head grouping (no repeat_kv materialization — the kernel
+indexes kv_head = q_head // group), seq up to 8k+, head_dim 128.
+
+Layout: q [b, hq, s, d], k/v [b, hkv, s, d] — all bf16 contiguous.
+Returns o [b, hq, s, d]; saves per-row LSE [b, hq, s] fp32 for the
+recompute backward (FlashAttention-2 style two-kernel dKV/dQ backward).
+
+CPU path: exact fp32 reference via scaled_dot_product_attention (tests
+compare the HIP kernel against this).
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn.functional as F
+
+from . import kernels_for
+
+
+def _cpu_ref_fwd(q, k, v, causal, scale):
+    hq, hkv = q.size(1), k.size(1)
+    if hq != hkv:
+        k = k.repeat_interleave(hq // hkv, dim=1)
+        v = v.repeat_interleave(hq // hkv, dim=1)
+    qf, kf, vf = q.float(), k.float(), v.float()
+    s = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+    if causal:
+        mask = torch.ones(q.size(-2), k.size(-2), dtype=torch.bool, device=q.device).triu(1)
+        s = s.masked_fill(mask, float("-inf"))
+    lse = torch.logsumexp(s, dim=-1)
+    p = torch.softmax(s, dim=-1)
+    o = torch.matmul(p, vf)
+    return o.to(q.dtype), lse
+
+
+class _FlashAttnFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, causal: bool, scale: float):
+        kern = kernels_for(q)
+        if kern is not None:
+            o, lse = kern.flash_attn_fwd(
+                q.contiguous(), k.contiguous(), v.contiguous(), causal, scale
+            )
+        else:
+            o, lse = _cpu_ref_fwd(q, k, v, causal, scale)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.causal = causal
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse = ctx.saved_tensors
+        kern = kernels_for(q)
+        if kern is not None:
+            dq, dk, dv = kern.flash_attn_bwd(
+                do.contiguous(), q, k, v, o, lse, ctx.causal, ctx.scale
+            )
+            return dq, dk, dv, None, None
+        # CPU reference backward (fp32, explicit)
+        hq, hkv = q.size(1), k.size(1)
+        g = hq // hkv
+        kx = k.repeat_interleave(g, dim=1).float()
+        vx = v.repeat_interleave(g, dim=1).float()
+        qf, dof = q.float(), do.float()
+        s = torch.matmul(qf, kx.transpose(-1, -2)) * ctx.scale
+        if ctx.causal:
+            mask = torch.ones(q.size(-2), k.size(-2), dtype=torch.bool, device=q.device).triu(1)
+            s = s.masked_fill(mask, float("-inf"))
+        p = torch.softmax(s, dim=-1)
+        dv = torch.matmul(p.transpose(-1, -2), dof)
+        dp = torch.matmul(dof, vx.transpose(-1, -2))
+        delta = (dof * o.float()).sum(-1, keepdim=True)
+        ds = p * (dp - delta) * ctx.scale
+        dq = torch.matmul(ds, kx)
+        dk = torch.matmul(ds.transpose(-1, -2), qf)
+        if g > 1:
+            dk = dk.reshape(dk.size(0), hkv, g, dk.size(-2), dk.size(-1)).sum(2)
+            dv = dv.reshape(dv.size(0), hkv, g, dv.size(-2), dv.size(-1)).sum(2)
+        return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), None, None
+
+
+def flash_attn_func(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    causal: bool = True,
+    scale: float | None = None,
+) -> torch.Tensor:
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.size(-1))
+    return _FlashAttnFn.apply(q, k, v, causal, scale)

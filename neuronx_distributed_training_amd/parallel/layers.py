@@ -1,0 +1,344 @@
+"""Tensor-parallel layers for MI355X (RCCL over xGMI).
+
+Capability parity with the reference's NxD layer contracts
+(``ColumnParallelLinear`` / ``RowParallelLinear`` / ``ParallelEmbedding`` /
+``GQAQKVColumnParallelLinear``; call sites pinned in
+/root/reference src/.../models/hf_models/modeling_llama.py:185-357 and
+models/megatron/transformer.py:876-955), designed MI355X-first:
+
+- the GEMM itself is ``F.linear`` → hipBLASLt on ROCm (plain library GEMM);
+- TP collectives are eager RCCL ops via :mod:`.mappings`, placed so a Row
+  fwd all-reduce (or SP reduce-scatter) can overlap the next GEMM on a
+  separate HIP stream (overlap handled by the trainer's comm stream);
+- sharded weight init happens on a CPU master tensor with a shared seed so
+  every TP rank slices the same full matrix (numerics-testable vs dense).
+
+Sequence-parallel convention: activations are [s, b, h] with dim 0 sharded
+by TP when ``sequence_parallel=True`` (matches reference layout switches at
+modeling_llama.py:398-400).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Callable, Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from . import state as ps
+from .mappings import (
+    copy_to_tensor_model_parallel_region,
+    gather_from_sequence_parallel_region,
+    gather_from_tensor_model_parallel_region,
+    reduce_from_tensor_model_parallel_region,
+    reduce_scatter_to_sequence_parallel_region,
+    scatter_to_tensor_model_parallel_region,
+)
+
+__all__ = [
+    "ColumnParallelLinear",
+    "RowParallelLinear",
+    "ParallelEmbedding",
+    "GQAQKVColumnParallelLinear",
+]
+
+
+def _default_init(weight: torch.Tensor) -> None:
+    nn.init.kaiming_uniform_(weight, a=math.sqrt(5))
+
+
+def _shard_master(
+    full_shape,
+    partition_dim: int,
+    stride: int,
+    init_method: Callable,
+    dtype: torch.dtype,
+    seed: Optional[int] = None,
+) -> torch.Tensor:
+    """Init a full-size master weight on CPU and return this rank's shard.
+
+    ``stride`` > 1 means the full output dim is the concatenation of
+    ``stride`` logically distinct matrices (e.g. fused gate_up stride=2,
+    fused qkv stride=3); each is sharded independently so rank r holds
+    [gate_r | up_r] rather than a contiguous slice of [gate | up].
+    """
+    world = ps.get_tensor_model_parallel_world_size()
+    rank = ps.get_tensor_model_parallel_rank()
+    if seed is not None:
+        g_state = torch.random.get_rng_state()
+        torch.manual_seed(seed)
+    master = torch.empty(*full_shape, dtype=torch.float32)
+    init_method(master)
+    if seed is not None:
+        torch.random.set_rng_state(g_state)
+    if world == 1:
+        return master.to(dtype)
+    dim_size = full_shape[partition_dim]
+    assert dim_size % (world * stride) == 0, (
+        f"dim {dim_size} not divisible by tp*stride {world}*{stride}"
+    )
+    pieces = torch.chunk(master, world * stride, dim=partition_dim)
+    # piece layout: [p0_r0..p0_rW, p1_r0..p1_rW, ...]; rank r takes pK_rR
+    mine = [pieces[k * world + rank] for k in range(stride)]
+    return torch.cat(mine, dim=partition_dim).contiguous().to(dtype)
+
+
+class ColumnParallelLinear(nn.Module):
+    """Y = X A^T with A row-sharded over TP (output dim split).
+
+    Forward has no collective when ``gather_output=False``; backward
+    all-reduces dX (or reduce-scatters under SP).
+    """
+
+    def __init__(
+        self,
+        input_size: int,
+        output_size: int,
+        bias: bool = False,
+        gather_output: bool = False,
+        sequence_parallel: bool = False,
+        stride: int = 1,
+        init_method: Callable = _default_init,
+        dtype: torch.dtype = torch.float32,
+        init_seed: Optional[int] = None,
+    ):
+        super().__init__()
+        world = ps.get_tensor_model_parallel_world_size()
+        assert output_size % world == 0
+        self.input_size = input_size
+        self.output_size = output_size
+        self.output_size_per_partition = output_size // world
+        self.gather_output = gather_output
+        self.sequence_parallel = sequence_parallel and world > 1
+        self.weight = nn.Parameter(
+            _shard_master((output_size, input_size), 0, stride, init_method, dtype, init_seed)
+        )
+        self.weight.tensor_model_parallel = True
+        self.weight.partition_dim = 0
+        self.weight.partition_stride = stride
+        if bias:
+            self.bias = nn.Parameter(
+                torch.zeros(self.output_size_per_partition, dtype=dtype)
+            )
+            self.bias.tensor_model_parallel = True
+            self.bias.partition_dim = 0
+        else:
+            self.register_parameter("bias", None)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if self.sequence_parallel:
+            x = gather_from_sequence_parallel_region(x)
+        else:
+            x = copy_to_tensor_model_parallel_region(x)
+        out = F.linear(x, self.weight, self.bias)
+        if self.gather_output:
+            out = gather_from_tensor_model_parallel_region(out)
+        return out
+
+
+class RowParallelLinear(nn.Module):
+    """Y = X A^T with A column-sharded over TP (input dim split).
+
+    Forward all-reduces the partial sums (or reduce-scatters under SP);
+    backward has no collective.
+    """
+
+    def __init__(
+        self,
+        input_size: int,
+        output_size: int,
+        bias: bool = False,
+        input_is_parallel: bool = True,
+        sequence_parallel: bool = False,
+        init_method: Callable = _default_init,
+        dtype: torch.dtype = torch.float32,
+        init_seed: Optional[int] = None,
+    ):
+        super().__init__()
+        world = ps.get_tensor_model_parallel_world_size()
+        assert input_size % world == 0
+        self.input_size = input_size
+        self.output_size = output_size
+        self.input_size_per_partition = input_size // world
+        self.input_is_parallel = input_is_parallel
+        self.sequence_parallel = sequence_parallel and world > 1
+        self.weight = nn.Parameter(
+            _shard_master((output_size, input_size), 1, 1, init_method, dtype, init_seed)
+        )
+        self.weight.tensor_model_parallel = True
+        self.weight.partition_dim = 1
+        if bias:
+            # bias is replicated; applied after the reduce
+            self.bias = nn.Parameter(torch.zeros(output_size, dtype=dtype))
+        else:
+            self.register_parameter("bias", None)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if not self.input_is_parallel:
+            x = scatter_to_tensor_model_parallel_region(x)
+        out = F.linear(x, self.weight)
+        if self.sequence_parallel:
+            out = reduce_scatter_to_sequence_parallel_region(out)
+        else:
+            out = reduce_from_tensor_model_parallel_region(out)
+        if self.bias is not None:
+            out = out + self.bias
+        return out
+
+
+class ParallelEmbedding(nn.Module):
+    """Vocab-sharded embedding: each rank holds vocab/tp rows; out-of-range
+    tokens produce zeros and the partial embeddings are all-reduced."""
+
+    def __init__(
+        self,
+        num_embeddings: int,
+        embedding_dim: int,
+        init_method: Callable = lambda w: nn.init.normal_(w, std=0.02),
+        dtype: torch.dtype = torch.float32,
+        init_seed: Optional[int] = None,
+    ):
+        super().__init__()
+        world = ps.get_tensor_model_parallel_world_size()
+        rank = ps.get_tensor_model_parallel_rank()
+        assert num_embeddings % world == 0
+        self.num_embeddings = num_embeddings
+        self.embedding_dim = embedding_dim
+        self.vocab_per_partition = num_embeddings // world
+        self.vocab_start = rank * self.vocab_per_partition
+        self.vocab_end = self.vocab_start + self.vocab_per_partition
+        self.weight = nn.Parameter(
+            _shard_master((num_embeddings, embedding_dim), 0, 1, init_method, dtype, init_seed)
+        )
+        self.weight.tensor_model_parallel = True
+        self.weight.partition_dim = 0
+
+    def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
+        world = ps.get_tensor_model_parallel_world_size()
+        if world == 1:
+            return F.embedding(input_ids, self.weight)
+        mask = (input_ids < self.vocab_start) | (input_ids >= self.vocab_end)
+        local = input_ids.clamp(self.vocab_start, self.vocab_end - 1) - self.vocab_start
+        out = F.embedding(local, self.weight)
+        out = out.masked_fill(mask.unsqueeze(-1), 0.0)
+        return reduce_from_tensor_model_parallel_region(out)
+
+
+class GQAQKVColumnParallelLinear(nn.Module):
+    """Fused QKV projection with KV-head replication for GQA.
+
+    Lets TP exceed the number of KV heads: K/V weights are replicated
+    ``kv_size_multiplier`` times so each TP rank gets
+    ``num_kv_heads * kv_size_multiplier / tp`` KV heads (capability parity
+    with reference modeling_llama.py:310-320, head math :358-362).
+
+    Output: one tensor [..., (q_heads/tp + 2*kv_heads*mult/tp) * head_dim]
+    laid out as [Q | K | V] per rank.
+    """
+
+    def __init__(
+        self,
+        hidden_size: int,
+        num_heads: int,
+        num_kv_heads: int,
+        head_dim: int,
+        kv_size_multiplier: int = 1,
+        bias: bool = False,
+        sequence_parallel: bool = False,
+        init_method: Callable = _default_init,
+        dtype: torch.dtype = torch.float32,
+        init_seed: Optional[int] = None,
+    ):
+        super().__init__()
+        world = ps.get_tensor_model_parallel_world_size()
+        rank = ps.get_tensor_model_parallel_rank()
+        assert num_heads % world == 0
+        assert (num_kv_heads * kv_size_multiplier) % world == 0
+        self.hidden_size = hidden_size
+        self.num_heads = num_heads
+        self.num_kv_heads = num_kv_heads
+        self.head_dim = head_dim
+        self.kv_size_multiplier = kv_size_multiplier
+        self.sequence_parallel = sequence_parallel and world > 1
+        self.num_heads_per_partition = num_heads // world
+        self.num_kv_heads_per_partition = num_kv_heads * kv_size_multiplier // world
+
+        q_shard = _shard_master(
+            (num_heads * head_dim, hidden_size), 0, 1, init_method, dtype,
+            init_seed,
+        )
+        # master K/V: init the true (unreplicated) KV weight, then replicate
+        # each head `kv_size_multiplier` times ADJACENTLY (repeat_interleave)
+        # so rank r's local KV heads are exactly the ones its local Q-head
+        # group attends to (reference kv_size_multiplier semantics,
+        # modeling_llama.py:310-320, head math :358-362).
+        def _kv_shard(seed_off):
+            seed = None if init_seed is None else init_seed + seed_off
+            if seed is not None:
+                st = torch.random.get_rng_state()
+                torch.manual_seed(seed)
+            master = torch.empty(num_kv_heads * head_dim, hidden_size, dtype=torch.float32)
+            init_method(master)
+            if seed is not None:
+                torch.random.set_rng_state(st)
+            rep = (
+                master.view(num_kv_heads, head_dim, hidden_size)
+                .repeat_interleave(kv_size_multiplier, dim=0)
+                .reshape(num_kv_heads * kv_size_multiplier * head_dim, hidden_size)
+            )
+            per = num_kv_heads * kv_size_multiplier * head_dim // world
+            return rep[rank * per : (rank + 1) * per].contiguous().to(dtype)
+
+        k_shard = _kv_shard(1)
+        v_shard = _kv_shard(2)
+        self.weight_q = nn.Parameter(q_shard)
+        self.weight_k = nn.Parameter(k_shard)
+        self.weight_v = nn.Parameter(v_shard)
+        for w in (self.weight_q, self.weight_k, self.weight_v):
+            w.tensor_model_parallel = True
+            w.partition_dim = 0
+
+        # Keep replicated KV copies in lockstep: every replica of a source
+        # head must apply the SUM of all replicas' grads (then each copy
+        # steps identically — training matches the unreplicated model).
+        mult = kv_size_multiplier
+        if mult > 1:
+            per_heads = num_kv_heads * mult // world  # local kv heads
+            within = min(per_heads, mult)
+            assert mult % within == 0 and per_heads % within == 0, (
+                f"kv replication layout unsupported: per_rank={per_heads}, "
+                f"mult={mult}"
+            )
+            cross = mult // within  # ranks sharing each source head
+            hd = head_dim
+
+            def _sync(grad):
+                g = grad.view(per_heads // within, within, hd, grad.size(-1))
+                g = g.sum(dim=1, keepdim=True).expand_as(g).contiguous()
+                g = g.view_as(grad)
+                if cross > 1:
+                    import torch.distributed as dist
+                    grp = ps.get_tensor_model_parallel_replica_group(cross)
+                    dist.all_reduce(g, group=grp)
+                return g
+
+            self.weight_k.register_hook(_sync)
+            self.weight_v.register_hook(_sync)
+        if bias:
+            self.bias_q = nn.Parameter(torch.zeros(self.num_heads_per_partition * head_dim, dtype=dtype))
+            self.bias_k = nn.Parameter(torch.zeros(self.num_kv_heads_per_partition * head_dim, dtype=dtype))
+            self.bias_v = nn.Parameter(torch.zeros(self.num_kv_heads_per_partition * head_dim, dtype=dtype))
+        else:
+            self.bias_q = self.bias_k = self.bias_v = None
+
+    def forward(self, x: torch.Tensor):
+        if self.sequence_parallel:
+            x = gather_from_sequence_parallel_region(x)
+        else:
+            x = copy_to_tensor_model_parallel_region(x)
+        q = F.linear(x, self.weight_q, self.bias_q)
+        k = F.linear(x, self.weight_k, self.bias_k)
+        v = F.linear(x, self.weight_v, self.bias_v)
+        return q, k, v

@@ -1,0 +1,182 @@
+"""Sharded checkpoint engine (reference nxd.save_checkpoint /
+load_checkpoint contract, nlp_overrides.py:535-639 + dir layout from
+nnm_model_ckpt_to_nxdt_model_ckpt_converter.py:77-113).
+
+On-disk layout::
+
+    <dir>/<tag>.ckpt/
+        model/dp_rank_00_tp_rank_XX_pp_rank_XX.pt    # model shard per rank
+        optim/dp_rank_XX_tp_rank_XX_pp_rank_XX.pt    # ZeRO-1 shard per rank
+        user_content.pt                               # loop state, config
+        done                                          # commit marker
+
+Model shards are written once per (tp, pp) coordinate (dp rank 0);
+optimizer shards are per-DP-rank (ZeRO-1 state is DP-sharded). Async save
+runs in a background thread over CPU copies (reference async_checkpointing
+semantics); keep-top-k prunes oldest step tags.
+"""
+
+from __future__ import annotations
+
+import os
+import re
+import shutil
+import threading
+from typing import Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+
+from ..parallel import state as ps
+
+
+def _rank_tags():
+    return (
+        ps.get_data_parallel_rank(),
+        ps.get_tensor_model_parallel_rank(),
+        ps.get_pipeline_model_parallel_rank(),
+    )
+
+
+def _model_shard_name() -> str:
+    _, tp, pp = _rank_tags()
+    return f"dp_rank_00_tp_rank_{tp:02d}_pp_rank_{pp:02d}.pt"
+
+
+def _optim_shard_name() -> str:
+    dp, tp, pp = _rank_tags()
+    return f"dp_rank_{dp:02d}_tp_rank_{tp:02d}_pp_rank_{pp:02d}.pt"
+
+
+def _cpu_copy(obj):
+    if torch.is_tensor(obj):
+        return obj.detach().to("cpu", non_blocking=False)
+    if isinstance(obj, dict):
+        return {k: _cpu_copy(v) for k, v in obj.items()}
+    if isinstance(obj, (list, tuple)):
+        t = type(obj)
+        return t(_cpu_copy(v) for v in obj)
+    return obj
+
+
+class CheckpointIO:
+    def __init__(self, async_save: bool = False):
+        self.async_save = async_save
+        self._pending: List[threading.Thread] = []
+
+    # ---- save ----
+    def save(
+        self,
+        ckpt_dir: str,
+        tag: str,
+        module,
+        user_content: Dict,
+        keep_top_k: int = 0,
+    ):
+        root = os.path.join(ckpt_dir, f"{tag}.ckpt")
+        os.makedirs(os.path.join(root, "model"), exist_ok=True)
+        os.makedirs(os.path.join(root, "optim"), exist_ok=True)
+
+        dp, tp, pp = _rank_tags()
+        work = []
+        if dp == 0 and ps.get_context_model_parallel_rank() == 0:
+            work.append(
+                (os.path.join(root, "model", _model_shard_name()),
+                 _cpu_copy(module.model.state_dict()))
+            )
+        if module.optimizer is not None and ps.get_context_model_parallel_rank() == 0:
+            work.append(
+                (os.path.join(root, "optim", _optim_shard_name()),
+                 _cpu_copy(module.optimizer.state_dict()))
+            )
+        is_global_zero = (not dist.is_initialized()) or dist.get_rank() == 0
+        if is_global_zero:
+            uc = dict(user_content)
+            if module.scheduler is not None:
+                uc["scheduler"] = module.scheduler.state_dict()
+            work.append((os.path.join(root, "user_content.pt"), uc))
+
+        def _write():
+            for path, obj in work:
+                tmp = path + ".tmp"
+                torch.save(obj, tmp)
+                os.replace(tmp, path)
+
+        if self.async_save:
+            t = threading.Thread(target=_write, daemon=False)
+            t.start()
+            self._pending.append(t)
+        else:
+            _write()
+
+        if dist.is_initialized():
+            dist.barrier()
+        if is_global_zero and not self.async_save:
+            open(os.path.join(root, "done"), "w").close()
+            if keep_top_k:
+                self._prune(ckpt_dir, keep_top_k)
+        elif is_global_zero and self.async_save:
+            def _commit(threads=list(self._pending), root=root, k=keep_top_k, d=ckpt_dir):
+                for t in threads:
+                    t.join()
+                open(os.path.join(root, "done"), "w").close()
+                if k:
+                    self._prune(d, k)
+            tc = threading.Thread(target=_commit, daemon=False)
+            tc.start()
+            self._pending = [tc]
+
+    def finalize(self):
+        for t in self._pending:
+            t.join()
+        self._pending = []
+
+    def _prune(self, ckpt_dir: str, keep: int):
+        tags = []
+        for name in os.listdir(ckpt_dir):
+            m = re.match(r"step=(\d+).*\.ckpt$", name)
+            if m and os.path.exists(os.path.join(ckpt_dir, name, "done")):
+                tags.append((int(m.group(1)), name))
+        tags.sort()
+        for _, name in tags[:-keep] if keep > 0 else []:
+            shutil.rmtree(os.path.join(ckpt_dir, name), ignore_errors=True)
+
+    # ---- load ----
+    def load(self, path: str, module, weight_init_only: bool = False) -> Dict:
+        """path: .../<tag>.ckpt directory. Returns user_content."""
+        model_path = os.path.join(path, "model", _model_shard_name())
+        sd = torch.load(model_path, map_location="cpu", weights_only=False)
+        module.model.load_state_dict(sd)
+        if not weight_init_only and module.optimizer is not None:
+            opath = os.path.join(path, "optim", _optim_shard_name())
+            if os.path.exists(opath):
+                osd = torch.load(opath, map_location="cpu", weights_only=False)
+                module.optimizer.load_state_dict(osd)
+        uc_path = os.path.join(path, "user_content.pt")
+        uc = {}
+        if os.path.exists(uc_path):
+            uc = torch.load(uc_path, map_location="cpu", weights_only=False)
+            if uc.get("scheduler") and module.scheduler is not None and not weight_init_only:
+                module.scheduler.load_state_dict(uc["scheduler"])
+        return uc
+
+
+def find_latest_checkpoint(ckpt_dir: str) -> Optional[str]:
+    """Resume discovery: newest complete `*.ckpt` dir (reference
+    exp_manager.py:370-385 semantics, by step number then mtime)."""
+    if not ckpt_dir or not os.path.isdir(ckpt_dir):
+        return None
+    best = None
+    best_key = (-1, -1.0)
+    for name in os.listdir(ckpt_dir):
+        p = os.path.join(ckpt_dir, name)
+        if not name.endswith(".ckpt") or not os.path.isdir(p):
+            continue
+        if not os.path.exists(os.path.join(p, "done")):
+            continue
+        m = re.match(r"step=(\d+)", name)
+        step = int(m.group(1)) if m else 0
+        key = (step, os.path.getmtime(p))
+        if key > best_key:
+            best_key, best = key, p
+    return best

@@ -1,0 +1,208 @@
+"""Model modules: the PTL-shaped wrapper layer (no Lightning dependency).
+
+Mirrors the capability of the reference's ``BaseModelModule``
+(lightning_modules/model/base.py): num_microbatches math, microbatch
+grad-accumulation loop, DP/CP loss all-reduce, throughput + param/grad-norm
+bookkeeping, optimizer construction (ZeRO-1 AdamW + LR schedule).
+XLA machinery (mark_step, step closures) is gone — eager HIP streams.
+"""
+
+from __future__ import annotations
+
+import time
+from typing import Dict, Iterator, Optional
+
+import torch
+import torch.distributed as dist
+
+from ..parallel import state as ps
+from ..models.llama import LlamaConfig, LlamaForCausalLM
+from ..optim.zero1 import ZeRO1AdamW
+from ..optim.lr_scheduler import build_scheduler
+from ..utils.throughput import Throughput
+
+
+class BaseModelModule:
+    """Owns the model + optimizer and runs one global-batch training step."""
+
+    def __init__(self, cfg: Dict):
+        self.cfg = cfg
+        self.global_batch_size = int(cfg["data"]["global_batch_size"])
+        self.micro_batch_size = int(cfg["data"]["micro_batch_size"])
+        self.seq_length = int(cfg["data"]["seq_length"])
+        dp = ps.get_data_parallel_world_size()
+        assert self.global_batch_size % (self.micro_batch_size * dp) == 0, (
+            f"GBS {self.global_batch_size} must divide by MBS*DP = "
+            f"{self.micro_batch_size}*{dp}"
+        )
+        # grad accumulation implied by GBS/MBS/DP (reference model/base.py:57)
+        self.num_microbatches = self.global_batch_size // (self.micro_batch_size * dp)
+        self.model: Optional[torch.nn.Module] = None
+        self.optimizer: Optional[ZeRO1AdamW] = None
+        self.scheduler = None
+        self.device = torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
+        self.throughput = Throughput(window=10)
+        self.log_param_norm = bool(cfg.get("exp_manager", {}).get("log_parameter_norm", False))
+        self.log_grad_norm = bool(cfg.get("exp_manager", {}).get("log_gradient_norm", True))
+
+    # -- to override --
+    def build_model(self) -> torch.nn.Module:
+        raise NotImplementedError
+
+    def setup(self):
+        self.model = self.build_model().to(self.device)
+        self.model.train()
+
+    def configure_optimizers(self, max_steps: int):
+        ocfg = self.cfg["model"].get("optim", {})
+        lr = float(ocfg.get("lr", 3e-4))
+        self.optimizer = ZeRO1AdamW(
+            list(self.model.named_parameters()),
+            lr=lr,
+            betas=tuple(ocfg.get("betas", (0.9, 0.95))),
+            eps=float(ocfg.get("eps", 1e-8)),
+            weight_decay=float(ocfg.get("weight_decay", 0.01)),
+            grad_clip=float(self.cfg["model"].get("grad_clip", 1.0)),
+        )
+        sched_cfg = ocfg.get("sched", {})
+        self.scheduler = build_scheduler(
+            sched_cfg.get("name", "linear"),
+            self.optimizer,
+            max_lr=lr,
+            warmup_steps=int(sched_cfg.get("warmup_steps", 100)),
+            total_steps=int(sched_cfg.get("max_steps", max_steps)),
+            min_lr=float(sched_cfg.get("min_lr", 0.0)),
+        )
+
+    def model_fwd_calc_loss(self, batch: Dict[str, torch.Tensor]) -> torch.Tensor:
+        return self.model(
+            batch["input_ids"],
+            labels=batch.get("labels", batch["input_ids"]),
+            loss_mask=batch.get("loss_mask"),
+        )
+
+    def get_batch_on_this_context_parallel_rank(self, batch):
+        """Split the sequence across the CP group (reference model/base.py:199)."""
+        cp = ps.get_context_model_parallel_world_size()
+        if cp == 1:
+            return batch
+        r = ps.get_context_model_parallel_rank()
+        out = {}
+        for k, v in batch.items():
+            if torch.is_tensor(v) and v.dim() >= 2 and v.size(1) == self.seq_length:
+                out[k] = v.chunk(cp, dim=1)[r].contiguous()
+            else:
+                out[k] = v
+        return out
+
+    def forward_backward_step(self, microbatches) -> torch.Tensor:
+        """Grad-accumulation loop; returns DP/CP-reduced mean loss."""
+        running = torch.zeros((), dtype=torch.float32, device=self.device)
+        n = self.num_microbatches
+        for batch in microbatches:
+            batch = {
+                k: (v.to(self.device, non_blocking=True) if torch.is_tensor(v) else v)
+                for k, v in batch.items()
+            }
+            batch = self.get_batch_on_this_context_parallel_rank(batch)
+            loss = self.model_fwd_calc_loss(batch)
+            (loss / n).backward()
+            running += loss.detach().float()
+        running /= n
+        if ps.get_data_parallel_world_size() > 1:
+            dist.all_reduce(running, group=ps.get_data_parallel_group())
+            running /= ps.get_data_parallel_world_size()
+        if ps.get_context_model_parallel_world_size() > 1:
+            dist.all_reduce(running, group=ps.get_context_model_parallel_group())
+            running /= ps.get_context_model_parallel_world_size()
+        return running
+
+    def training_step(self, microbatches) -> Dict[str, float]:
+        self.optimizer.zero_grad()
+        t0 = time.perf_counter()
+        loss = self.forward_backward_step(microbatches)
+        gnorm = self.optimizer.step()
+        self.scheduler.step()
+        if self.device.type == "cuda":
+            torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        seqs = self.micro_batch_size * self.num_microbatches * ps.get_data_parallel_world_size()
+        self.throughput.update(seqs, dt)
+        metrics = {
+            "reduced_train_loss": float(loss),
+            "lr": self.optimizer.lr,
+            "throughput_seq_s": self.throughput.value,
+            "step_time_s": dt,
+        }
+        if self.log_grad_norm and gnorm is not None:
+            metrics["gradient_norm"] = float(gnorm)
+        if self.log_param_norm:
+            metrics["parameter_norm"] = float(self.calculate_parameter_norm())
+        return metrics
+
+    @torch.no_grad()
+    def validation_step(self, batch) -> torch.Tensor:
+        batch = {
+            k: (v.to(self.device) if torch.is_tensor(v) else v) for k, v in batch.items()
+        }
+        batch = self.get_batch_on_this_context_parallel_rank(batch)
+        return self.model_fwd_calc_loss(batch).detach().float()
+
+    @torch.no_grad()
+    def calculate_parameter_norm(self) -> torch.Tensor:
+        """Global param norm over TP/PP groups (reference model/base.py:397-452)."""
+        tp_rank = ps.get_tensor_model_parallel_rank()
+        sq = torch.zeros((), dtype=torch.float32, device=self.device)
+        for p in self.model.parameters():
+            if getattr(p, "tensor_model_parallel", False) or tp_rank == 0:
+                sq += p.float().pow(2).sum()
+        if ps.get_tensor_model_parallel_world_size() > 1:
+            dist.all_reduce(sq, group=ps.get_tensor_model_parallel_group())
+        if ps.get_pipeline_model_parallel_world_size() > 1:
+            dist.all_reduce(sq, group=ps.get_pipeline_model_parallel_group())
+        return sq.sqrt()
+
+    # -- checkpoint content --
+    def state_dict(self):
+        return {
+            "model": self.model.state_dict(),
+            "optimizer": self.optimizer.state_dict(),
+            "scheduler": self.scheduler.state_dict() if self.scheduler else None,
+        }
+
+    def load_state_dict(self, sd):
+        self.model.load_state_dict(sd["model"])
+        if sd.get("optimizer") and self.optimizer:
+            self.optimizer.load_state_dict(sd["optimizer"])
+        if sd.get("scheduler") and self.scheduler:
+            self.scheduler.load_state_dict(sd["scheduler"])
+
+
+class LlamaModule(BaseModelModule):
+    """HF-style Llama model module (reference HFLLamaModule parity)."""
+
+    def build_model(self) -> torch.nn.Module:
+        mcfg = self.cfg["model"]
+        precision = str(self.cfg.get("precision", {}).get("type", "bf16"))
+        want_bf16 = ("bf16" in precision) or ("mixed" in precision)
+        # CPU runs stay fp32 (bf16 matmul is unusably slow off-GPU)
+        dtype = "bfloat16" if (want_bf16 and torch.cuda.is_available()) else "float32"
+        dstr = self.cfg.get("distributed_strategy", {})
+        cfg = LlamaConfig(
+            vocab_size=int(mcfg.get("vocab_size", 128256)),
+            hidden_size=int(mcfg.get("hidden_size", 4096)),
+            intermediate_size=int(mcfg.get("intermediate_size", 14336)),
+            num_hidden_layers=int(mcfg.get("num_layers", 32)),
+            num_attention_heads=int(mcfg.get("num_attention_heads", 32)),
+            num_key_value_heads=int(mcfg.get("num_kv_heads", 8)),
+            max_position_embeddings=int(self.seq_length),
+            rms_norm_eps=float(mcfg.get("rms_norm_eps", 1e-5)),
+            rope_theta=float(mcfg.get("rope_theta", 500000.0)),
+            sequence_parallel=bool(dstr.get("sequence_parallel", False)),
+            qkv_linear=bool(mcfg.get("qkv_linear", False)),
+            kv_replicator=int(mcfg.get("kv_replicator", 1)),
+            fuse_qkv=bool(mcfg.get("fuse_qkv", True)),
+            activation_checkpoint=mcfg.get("activation_checkpoint"),
+            dtype=dtype,
+        )
+        return LlamaForCausalLM(cfg)

@@ -1,0 +1,115 @@
+"""Experiment manager (reference utils/exp_manager.py parity).
+
+Log-dir versioning, resume discovery, run archival, TensorBoard logger,
+per-step timing callback.
+"""
+
+from __future__ import annotations
+
+import datetime
+import json
+import os
+import shutil
+import sys
+import time
+from typing import Dict, List, Optional, Tuple
+
+import torch.distributed as dist
+
+
+def _is_global_zero() -> bool:
+    return (not dist.is_initialized()) or dist.get_rank() == 0
+
+
+class JsonlLogger:
+    """Always-on lightweight metrics logger (one JSON line per log call)."""
+
+    def __init__(self, path: str):
+        self.path = path
+        self._fh = open(path, "a") if _is_global_zero() else None
+
+    def log_metrics(self, metrics: Dict, step: int):
+        if self._fh:
+            rec = {"step": step, **{k: v for k, v in metrics.items()}}
+            self._fh.write(json.dumps(rec) + "\n")
+            self._fh.flush()
+
+
+class TensorBoardLogger:
+    def __init__(self, log_dir: str):
+        self.writer = None
+        if _is_global_zero():
+            try:
+                from torch.utils.tensorboard import SummaryWriter
+                self.writer = SummaryWriter(log_dir)
+            except Exception:
+                self.writer = None
+
+    def log_metrics(self, metrics: Dict, step: int):
+        if self.writer:
+            for k, v in metrics.items():
+                if isinstance(v, (int, float)):
+                    self.writer.add_scalar(k, v, step)
+
+
+class TimingCallback:
+    """Per-step wall time into metrics (reference TimingCallback,
+    exp_manager.py:64-78, without the XLA step-closure)."""
+
+    def __init__(self):
+        self._t = None
+
+    def on_train_batch_end(self, trainer, module, metrics):
+        now = time.perf_counter()
+        if self._t is not None:
+            metrics.setdefault("train_step_timing", now - self._t)
+        self._t = now
+
+
+def exp_manager(trainer, em_cfg: Dict) -> Tuple[List, Optional[str]]:
+    """Create log dir (versioned), loggers, and return (loggers, ckpt_dir)."""
+    if not em_cfg:
+        return [], None
+    exp_dir = em_cfg.get("explicit_log_dir") or em_cfg.get("exp_dir")
+    if not exp_dir:
+        return [], None
+    name = em_cfg.get("name", "default")
+    log_dir = os.path.join(exp_dir, name) if name != "default" else exp_dir
+    ckpt_dir = os.path.join(log_dir, "checkpoints")
+    if _is_global_zero():
+        os.makedirs(ckpt_dir, exist_ok=True)
+        # archive previous run logs into run_N (reference exp_manager.py:387-404)
+        prev = [d for d in os.listdir(log_dir) if d.startswith("run_")]
+        run_idx = len(prev)
+        if em_cfg.get("resume_if_exists") and os.path.exists(
+            os.path.join(log_dir, "metrics.jsonl")
+        ):
+            arch = os.path.join(log_dir, f"run_{run_idx}")
+            os.makedirs(arch, exist_ok=True)
+            for f in ("metrics.jsonl", "cmd-args.log"):
+                src = os.path.join(log_dir, f)
+                if os.path.exists(src):
+                    shutil.move(src, os.path.join(arch, f))
+        with open(os.path.join(log_dir, "cmd-args.log"), "w") as f:
+            f.write(" ".join(sys.argv) + "\n" + datetime.datetime.now().isoformat() + "\n")
+    if dist.is_initialized():
+        dist.barrier()
+    loggers: List = [JsonlLogger(os.path.join(log_dir, "metrics.jsonl"))]
+    if em_cfg.get("create_tensorboard_logger"):
+        loggers.append(TensorBoardLogger(os.path.join(log_dir, "tb")))
+    if em_cfg.get("create_wandb_logger"):
+        try:
+            import wandb  # not installed in this image; guarded
+
+            class _W:
+                def __init__(self):
+                    wandb.init(project=em_cfg.get("wandb_project", name), dir=log_dir)
+
+                def log_metrics(self, m, s):
+                    wandb.log(m, step=s)
+
+            loggers.append(_W())
+        except Exception:
+            pass
+    trainer.callbacks.append(TimingCallback())
+    return loggers, ckpt_dir

@@ -1,0 +1,47 @@
+"""In-tree build of the gfx950 HIP extension.
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+The built _C*.so lands next to neuronx_distributed_training_amd/ops/ and
+travels to the GPU box with the repo snapshot.
+"""
+
+import os
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+CSRC = os.path.join(HERE, "neuronx_distributed_training_amd", "ops", "csrc")
+
+sources = [
+    os.path.join(CSRC, f)
+    for f in (
+        "bindings.cpp",
+        "rmsnorm.hip",
+        "swiglu.hip",
+        "rope.hip",
+        "adamw.hip",
+        "flash_attn_fwd.hip",
+        "flash_attn_bwd.hip",
+        "probe.hip",
+    )
+]
+
+setup(
+    name="neuronx_distributed_training_amd_ext",
+    ext_modules=[
+        CUDAExtension(
+            name="neuronx_distributed_training_amd.ops._C",
+            sources=sources,
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
+)

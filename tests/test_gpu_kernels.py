@@ -1,0 +1,167 @@
+"""HIP kernel numerics on MI355X vs plain PyTorch fp32 references.
+All tests here are @gpu (run via gpurun / the driver's GPU tier)."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def ext():
+    from neuronx_distributed_training_amd import ops
+
+    k = ops.require_extension()  # fail loudly if the .so is missing
+    return k
+
+
+def test_mfma_layout_probe(ext):
+    """A[16,32] @ B[32,16] through the fragment maps vs torch.matmul.
+    Asymmetric operands catch any transpose (guide §5.4 rule 16)."""
+    torch.manual_seed(0)
+    a = (torch.randn(16, 32) * 0.5).bfloat16().cuda()
+    b = (torch.arange(32 * 16).float().reshape(32, 16) % 7 - 3).bfloat16().cuda()
+    c = ext.mfma_probe(a.contiguous(), b.contiguous())
+    ref = a.float() @ b.float()
+    torch.cuda.synchronize()
+    assert torch.allclose(c, ref, atol=1e-2, rtol=1e-2), (c - ref).abs().max()
+
+
+def test_rmsnorm_gpu(ext):
+    from neuronx_distributed_training_amd.ops import rmsnorm
+
+    torch.manual_seed(1)
+    x = torch.randn(512, 4096, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    w = torch.randn(4096, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    y = rmsnorm(x, w, 1e-5)
+    xr = x.detach().float().clone().requires_grad_(True)
+    wr = w.detach().float().clone().requires_grad_(True)
+    ref = xr * torch.rsqrt(xr.pow(2).mean(-1, keepdim=True) + 1e-5) * wr
+    assert torch.allclose(y.float(), ref, atol=0.05, rtol=0.05)
+    g = torch.randn_like(y)
+    y.backward(g)
+    ref.backward(g.float())
+    assert torch.allclose(x.grad.float(), xr.grad, atol=0.05, rtol=0.05)
+    # dw accumulates over 512 rows — looser tol for bf16 products
+    assert torch.allclose(w.grad.float(), wr.grad, atol=0.8, rtol=0.05), (
+        (w.grad.float() - wr.grad).abs().max()
+    )
+
+
+def test_swiglu_gpu(ext):
+    from neuronx_distributed_training_amd.ops import swiglu
+
+    torch.manual_seed(2)
+    gu = torch.randn(1024, 2048, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    y = swiglu(gu)
+    gur = gu.detach().float().clone().requires_grad_(True)
+    gate, up = gur.chunk(2, dim=-1)
+    ref = torch.nn.functional.silu(gate) * up
+    assert torch.allclose(y.float(), ref, atol=0.05, rtol=0.05)
+    g = torch.randn_like(y)
+    y.backward(g)
+    ref.backward(g.float())
+    assert torch.allclose(gu.grad.float(), gur.grad, atol=0.05, rtol=0.05)
+
+
+def test_rope_gpu(ext):
+    from neuronx_distributed_training_amd.ops.rope import (
+        apply_rotary_pos_emb,
+        build_rope_cache,
+    )
+
+    torch.manual_seed(3)
+    cos, sin = build_rope_cache(256, 128, base=500000.0, device="cuda")
+    x = torch.randn(2, 8, 256, 128, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    y = apply_rotary_pos_emb(x, cos, sin)
+
+    def rotate_half(t):
+        t1, t2 = t.chunk(2, dim=-1)
+        return torch.cat((-t2, t1), dim=-1)
+
+    xf = x.detach().float()
+    ref = xf * cos[:256] + rotate_half(xf) * sin[:256]
+    assert torch.allclose(y.float(), ref, atol=0.05, rtol=0.05)
+    y.sum().backward()
+    assert torch.isfinite(x.grad.float()).all()
+
+
+@pytest.mark.parametrize("s,hq,hkv", [(256, 4, 4), (512, 8, 2), (333, 4, 1)])
+def test_flash_attn_fwd_gpu(ext, s, hq, hkv):
+    from neuronx_distributed_training_amd.ops import flash_attn_func
+
+    torch.manual_seed(4)
+    b, d = 2, 128
+    q = torch.randn(b, hq, s, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(b, hkv, s, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(b, hkv, s, d, device="cuda", dtype=torch.bfloat16)
+    o = flash_attn_func(q, k, v, causal=True)
+    kx = k.repeat_interleave(hq // hkv, 1).float()
+    vx = v.repeat_interleave(hq // hkv, 1).float()
+    ref = torch.nn.functional.scaled_dot_product_attention(
+        q.float(), kx, vx, is_causal=True
+    )
+    err = (o.float() - ref).abs().max()
+    assert err < 0.02, f"max err {err}"
+
+
+def test_flash_attn_bwd_gpu(ext):
+    from neuronx_distributed_training_amd.ops import flash_attn_func
+
+    torch.manual_seed(5)
+    b, hq, hkv, s, d = 2, 4, 2, 512, 128
+    q = torch.randn(b, hq, s, d, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    k = torch.randn(b, hkv, s, d, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    v = torch.randn(b, hkv, s, d, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    o = flash_attn_func(q, k, v, causal=True)
+    g = torch.randn_like(o)
+    o.backward(g)
+
+    qr = q.detach().float().clone().requires_grad_(True)
+    kr = k.detach().float().clone().requires_grad_(True)
+    vr = v.detach().float().clone().requires_grad_(True)
+    ref = torch.nn.functional.scaled_dot_product_attention(
+        qr,
+        kr.repeat_interleave(hq // hkv, 1),
+        vr.repeat_interleave(hq // hkv, 1),
+        is_causal=True,
+    )
+    ref.backward(g.float())
+    for got, want, name in (
+        (q.grad, qr.grad, "dq"),
+        (k.grad, kr.grad, "dk"),
+        (v.grad, vr.grad, "dv"),
+    ):
+        err = (got.float() - want).abs().max()
+        scale = want.abs().max().clamp(min=1)
+        assert err / scale < 0.05, f"{name} rel err {err/scale}"
+
+
+def test_adamw_gpu(ext):
+    torch.manual_seed(6)
+    n = 10000
+    p = torch.randn(n, device="cuda")
+    g = torch.randn(n, device="cuda")
+    m = torch.zeros(n, device="cuda")
+    v = torch.zeros(n, device="cuda")
+    wd_mask = (torch.rand(n, device="cuda") > 0.5)
+    pr, mr, vr = p.clone(), m.clone(), v.clone()
+    lr, b1, b2, eps, wd, t = 1e-3, 0.9, 0.95, 1e-8, 0.01, 3
+    ext.adamw_step(p, g, m, v, wd_mask, lr, b1, b2, eps, wd, t)
+    # reference
+    mr = b1 * mr + (1 - b1) * g
+    vr = b2 * vr + (1 - b2) * g * g
+    bc1, bc2 = 1 - b1 ** t, 1 - b2 ** t
+    denom = (vr / bc2).sqrt() + eps
+    decay = torch.where(wd_mask, torch.tensor(1 - lr * wd, device="cuda"), torch.tensor(1.0, device="cuda"))
+    pr = pr * decay - (lr / bc1) * mr / denom
+    assert torch.allclose(p, pr, atol=1e-6)
+    assert torch.allclose(m, mr, atol=1e-6)
+    assert torch.allclose(v, vr, atol=1e-6)
+
+
+def test_extension_loaded_is_intree(ext):
+    """The loaded .so must live in the repo tree (native-code check)."""
+    assert "/neuronx_distributed_training_amd/ops/" in ext.__file__, ext.__file__

@@ -1,0 +1,83 @@
+"""CPU-path numerics of the fused ops vs plain torch references."""
+
+import math
+
+import torch
+import torch.nn.functional as F
+
+from neuronx_distributed_training_amd.ops import rmsnorm, swiglu, flash_attn_func
+from neuronx_distributed_training_amd.ops.rope import (
+    apply_rotary_pos_emb,
+    build_rope_cache,
+)
+
+
+def test_rmsnorm_matches_manual():
+    torch.manual_seed(0)
+    x = torch.randn(10, 64, requires_grad=True)
+    w = torch.randn(64, requires_grad=True)
+    y = rmsnorm(x, w, eps=1e-5)
+    xr = x.detach().clone().requires_grad_(True)
+    wr = w.detach().clone().requires_grad_(True)
+    ref = xr * torch.rsqrt(xr.pow(2).mean(-1, keepdim=True) + 1e-5) * wr
+    assert torch.allclose(y, ref, atol=1e-5)
+    g = torch.randn_like(y)
+    y.backward(g)
+    ref.backward(g)
+    assert torch.allclose(x.grad, xr.grad, atol=1e-4)
+    assert torch.allclose(w.grad, wr.grad, atol=1e-4)
+
+
+def test_swiglu_matches_manual():
+    torch.manual_seed(1)
+    gu = torch.randn(6, 32, requires_grad=True)
+    y = swiglu(gu)
+    gur = gu.detach().clone().requires_grad_(True)
+    gate, up = gur.chunk(2, dim=-1)
+    ref = F.silu(gate) * up
+    assert torch.allclose(y, ref, atol=1e-5)
+    g = torch.randn_like(y)
+    y.backward(g)
+    ref.backward(g)
+    assert torch.allclose(gu.grad, gur.grad, atol=1e-4)
+
+
+def test_rope_orthogonality_and_ref():
+    cos, sin = build_rope_cache(32, 16, base=10000.0)
+    x = torch.randn(2, 3, 32, 16, requires_grad=True)
+    y = apply_rotary_pos_emb(x, cos, sin)
+    # norm-preserving per pair
+    assert torch.allclose(
+        y.pow(2).sum(-1), x.pow(2).sum(-1), atol=1e-4
+    )
+    # matches HF-style reference
+    def rotate_half(t):
+        t1, t2 = t.chunk(2, dim=-1)
+        return torch.cat((-t2, t1), dim=-1)
+    ref = x * cos[:32] + rotate_half(x) * sin[:32]
+    assert torch.allclose(y, ref, atol=1e-5)
+    y.sum().backward()
+    assert x.grad is not None
+
+
+def test_flash_attn_cpu_matches_sdpa():
+    torch.manual_seed(2)
+    b, hq, hkv, s, d = 2, 4, 2, 33, 16
+    q = torch.randn(b, hq, s, d, requires_grad=True)
+    k = torch.randn(b, hkv, s, d, requires_grad=True)
+    v = torch.randn(b, hkv, s, d, requires_grad=True)
+    o = flash_attn_func(q, k, v, causal=True)
+    qr = q.detach().clone().requires_grad_(True)
+    kr = k.detach().clone().requires_grad_(True)
+    vr = v.detach().clone().requires_grad_(True)
+    ref = F.scaled_dot_product_attention(
+        qr, kr.repeat_interleave(hq // hkv, 1), vr.repeat_interleave(hq // hkv, 1),
+        is_causal=True,
+    )
+    assert torch.allclose(o, ref, atol=1e-4), (o - ref).abs().max()
+    g = torch.randn_like(o)
+    o.backward(g)
+    ref.backward(g)
+    assert torch.allclose(q.grad, qr.grad, atol=1e-4)
+    assert torch.allclose(k.grad, kr.grad, atol=1e-4)
+    assert torch.allclose(v.grad, vr.grad, atol=1e-4)
