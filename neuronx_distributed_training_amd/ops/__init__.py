@@ -28,7 +28,10 @@ def _try_load():
     if _C is not None:
         return _C
     try:
-        from . import _C as mod  # built in-tree by setup.py build_ext --inplace
+        import importlib
+        mod = importlib.import_module(
+            "neuronx_distributed_training_amd.ops._C"
+        )  # built in-tree by setup.py build_ext --inplace
         _C = mod
     except ImportError:
         # in-tree .so next to this file (hipcc -shared direct build)
