@@ -13,15 +13,17 @@ void launch_rmsnorm_bwd(const void*, const void*, const void*, const void*,
 void launch_swiglu_fwd(const void*, void*, long, int, hipStream_t);
 void launch_swiglu_bwd(const void*, const void*, void*, long, int,
                        hipStream_t);
-void launch_rope_fwd(const void*, const void*, const void*, void*, long, int,
-                     int, int, float, hipStream_t);
+void launch_rope_fwd(const void*, const void*, const void*, void*, int, int,
+                     int, int, const long*, int, float, hipStream_t);
 void launch_adamw(void*, const void*, void*, void*, const void*, long, float,
                   float, float, float, float, int, hipStream_t);
 void launch_flash_fwd(const void*, const void*, const void*, void*, void*,
-                      int, int, int, int, int, bool, float, hipStream_t);
+                      int, int, int, int, int, bool, float, const long*,
+                      const long*, const long*, hipStream_t);
 void launch_flash_bwd(const void*, const void*, const void*, const void*,
                       const void*, const void*, void*, void*, void*, int, int,
-                      int, int, int, bool, float, hipStream_t);
+                      int, int, int, bool, float, const long*, const long*,
+                      const long*, const long*, hipStream_t);
 void launch_mfma_probe(const void*, const void*, void*, hipStream_t);
 }
 
@@ -87,20 +89,22 @@ torch::Tensor swiglu_bwd(torch::Tensor dy, torch::Tensor gu) {
 
 torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cost, torch::Tensor sint,
                        long pos_offset) {
-  CHECK_IN(x);
-  TORCH_CHECK(x.dim() == 4, "rope: x must be [b, h, s, d]");
+  // x: logical [b, h, s, d], any strides with d contiguous (e.g. a view of
+  // the fused QKV output). Returns a [s, b, h, d]-contiguous tensor exposed
+  // as a [b, h, s, d] view — zero permute copies in the attention path.
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.stride(3) == 1,
+              "rope: x must be [b, h, s, d] on GPU with d contiguous");
   auto cos_c = cost.contiguous();
   auto sin_c = sint.contiguous();
-  // table may arrive fp32 on CPU buffers moved to GPU; require f32
   TORCH_CHECK(cos_c.scalar_type() == torch::kFloat32, "rope table must be f32");
-  long BH = x.size(0) * x.size(1);
-  int S = (int)x.size(2);
-  int D = (int)x.size(3);
-  auto y = torch::empty_like(x);
-  float sign = 1.0f;
+  int B = (int)x.size(0), H = (int)x.size(1), S = (int)x.size(2),
+      D = (int)x.size(3);
+  auto y_mem = torch::empty({S, B, H, D}, x.options());
+  long xstr[3] = {x.stride(2), x.stride(0), x.stride(1)};  // s, b, h
   launch_rope_fwd(x.data_ptr(), cos_c.data_ptr(), sin_c.data_ptr(),
-                  y.data_ptr(), BH, S, D, (int)pos_offset, sign, cur_stream());
-  return y;
+                  y_mem.data_ptr(), S, B, H, D, xstr, (int)pos_offset, 1.0f,
+                  cur_stream());
+  return y_mem.permute({1, 2, 0, 3});
 }
 
 void adamw_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
@@ -113,46 +117,62 @@ void adamw_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                (float)eps, (float)wd, (int)step, cur_stream());
 }
 
+static void check_bhsd(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.dim() == 4 && t.stride(3) == 1,
+              name, " must be [b, h, s, d] on GPU with d contiguous");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+}
+
 std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
                                           torch::Tensor v, bool causal,
                                           double scale) {
-  CHECK_IN(q);
-  CHECK_IN(k);
-  CHECK_IN(v);
-  TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "flash_attn: bf16 only");
+  // q/k/v: logical [b, h, s, d], arbitrary strides (views of the QKV GEMM
+  // output). Returns O as a [b, h, s, d] view of [s, b, h, d] storage.
+  check_bhsd(q, "q");
+  check_bhsd(k, "k");
+  check_bhsd(v, "v");
   int B = (int)q.size(0), HQ = (int)q.size(1), S = (int)q.size(2),
       D = (int)q.size(3);
   int HKV = (int)k.size(1);
   TORCH_CHECK(D == 128 || D == 64, "head_dim must be 64 or 128");
   TORCH_CHECK(HQ % HKV == 0, "GQA head mismatch");
-  auto o = torch::empty_like(q);
+  auto o_mem = torch::empty({S, B, HQ, D}, q.options());
   auto lse = torch::empty({B, HQ, S}, q.options().dtype(torch::kFloat32));
-  launch_flash_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
-                   lse.data_ptr(), B, HQ, HKV, S, D, causal, (float)scale,
-                   cur_stream());
-  return {o, lse};
+  long qs[3] = {q.stride(2), q.stride(0), q.stride(1)};
+  long ks[3] = {k.stride(2), k.stride(0), k.stride(1)};
+  long vs[3] = {v.stride(2), v.stride(0), v.stride(1)};
+  launch_flash_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o_mem.data_ptr(),
+                   lse.data_ptr(), B, HQ, HKV, S, D, causal, (float)scale, qs,
+                   ks, vs, cur_stream());
+  return {o_mem.permute({1, 2, 0, 3}), lse};
 }
 
 std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                                           torch::Tensor k, torch::Tensor v,
                                           torch::Tensor o, torch::Tensor lse,
                                           bool causal, double scale) {
-  CHECK_IN(dout);
-  CHECK_IN(q);
+  if (dout.stride(3) != 1) dout = dout.contiguous();
+  check_bhsd(dout, "dout");
+  check_bhsd(q, "q");
   int B = (int)q.size(0), HQ = (int)q.size(1), S = (int)q.size(2),
       D = (int)q.size(3);
   int HKV = (int)k.size(1);
   auto delta = (dout.to(torch::kFloat32) * o.to(torch::kFloat32))
                    .sum(-1)
                    .contiguous();  // [B, HQ, S] f32
-  auto dq = torch::empty_like(q);
-  auto dk = torch::empty_like(k);
-  auto dv = torch::empty_like(v);
+  auto dq_mem = torch::empty({S, B, HQ, D}, q.options());
+  auto dk_mem = torch::empty({S, B, HKV, D}, q.options());
+  auto dv_mem = torch::empty({S, B, HKV, D}, q.options());
+  long qs[3] = {q.stride(2), q.stride(0), q.stride(1)};
+  long ks[3] = {k.stride(2), k.stride(0), k.stride(1)};
+  long vs[3] = {v.stride(2), v.stride(0), v.stride(1)};
+  long ds[3] = {dout.stride(2), dout.stride(0), dout.stride(1)};
   launch_flash_bwd(dout.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
-                   lse.data_ptr(), delta.data_ptr(), dq.data_ptr(),
-                   dk.data_ptr(), dv.data_ptr(), B, HQ, HKV, S, D, causal,
-                   (float)scale, cur_stream());
-  return {dq, dk, dv};
+                   lse.data_ptr(), delta.data_ptr(), dq_mem.data_ptr(),
+                   dk_mem.data_ptr(), dv_mem.data_ptr(), B, HQ, HKV, S, D,
+                   causal, (float)scale, qs, ks, vs, ds, cur_stream());
+  return {dq_mem.permute({1, 2, 0, 3}), dk_mem.permute({1, 2, 0, 3}),
+          dv_mem.permute({1, 2, 0, 3})};
 }
 
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b) {

@@ -1,35 +1,37 @@
 #include "hip/hip_runtime.h"
-// Causal flash attention BACKWARD for MI355X (gfx950).
-// FlashAttention-2 style recompute, two kernels, no atomics:
-//   dkv: grid over key blocks  — recompute P^T, accumulate dV, dK
-//   dq : grid over query blocks — recompute P,  accumulate dQ
+// Causal flash attention BACKWARD for MI355X (gfx950) — v2.
+// FlashAttention-2 recompute, two kernels, no atomics:
+//   dkv: grid over 256-key blocks — recompute P^T, accumulate dV, dK
+//   dq : grid over 256-query blocks — recompute P, accumulate dQ
 // GQA: dkv sums over the q-head group of each kv head in-kernel.
-// delta = rowsum(dO ∘ O) is computed by the caller (eager, tiny).
+// delta = rowsum(dO ∘ O) is computed by the caller.
+//
+// v2: strided [s, b, h, d] IO (no permutes), 8 waves with 2 sub-blocks per
+// wave (every staged LDS B-fragment feeds 2 MFMAs), paired transpose
+// writes. dkv: BN=256 keys (32/wave), q-tile 64. dq: BM=256 (32/wave),
+// key tile 64.
 #include "attn_common.h"
 
 // ============================ dK / dV ============================
-// Block: 4 waves, BN=64 keys (16/wave). Loops q heads in the GQA group ×
-// q blocks ≥ key block (causal).
 template <int D, bool CAUSAL>
-__global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
-    const bf16* __restrict__ dO,  // [B, HQ, S, D]
-    const bf16* __restrict__ Q,   // [B, HQ, S, D]
-    const bf16* __restrict__ K,   // [B, HKV, S, D]
-    const bf16* __restrict__ V,   // [B, HKV, S, D]
-    const float* __restrict__ LSE,    // [B, HQ, S]
-    const float* __restrict__ DELTA,  // [B, HQ, S]
-    bf16* __restrict__ dK,  // [B, HKV, S, D]
-    bf16* __restrict__ dV,  // [B, HKV, S, D]
-    int S, int HQ, int HKV, float scale) {
-  constexpr int BM = 64, BN = 64;
+__global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
+    const bf16* __restrict__ dO, const bf16* __restrict__ Q,
+    const bf16* __restrict__ K, const bf16* __restrict__ V,
+    const float* __restrict__ LSE, const float* __restrict__ DELTA,
+    bf16* __restrict__ dK,  // [s, b, hkv, d] contiguous
+    bf16* __restrict__ dV, int S, int Bb, int HQ, int HKV, float scale,
+    long sQs, long sQb, long sQh, long sKs, long sKb, long sKh, long sVs,
+    long sVb, long sVh, long sDs, long sDb, long sDh) {
+  constexpr int BN = 256;  // keys per block
+  constexpr int BM = 64;   // q tile
   constexpr int KP = D + 8;
   constexpr int VP = BM + 8;
-  __shared__ __bf16 q_lds[BM * KP];    // Q rows      (B for S^T)
-  __shared__ __bf16 do_lds[BM * KP];   // dO rows     (B for dP^T)
-  __shared__ __bf16 dot_lds[D * VP];   // dO^T        (B for dV)
-  __shared__ __bf16 qt_lds[D * VP];    // Q^T         (B for dK)
-  __shared__ __bf16 pT_lds[4 * 16 * VP];   // per-wave P^T  (A for dV)
-  __shared__ __bf16 dsT_lds[4 * 16 * VP];  // per-wave dS^T (A for dK)
+  __shared__ __bf16 q_lds[BM * KP];
+  __shared__ __bf16 do_lds[BM * KP];
+  __shared__ __bf16 dot_lds[D * VP];
+  __shared__ __bf16 qt_lds[D * VP];
+  __shared__ __bf16 pT_lds[8 * 32 * VP];
+  __shared__ __bf16 dsT_lds[8 * 32 * VP];
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -39,139 +41,167 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
   const int b = bh / HKV;
   const int group = HQ / HKV;
 
-  const long kvoff = ((long)b * HKV + hkv) * S * D;
-  const bf16* Kp = K + kvoff;
-  const bf16* Vp = V + kvoff;
+  const bf16* Kp = K + b * sKb + hkv * sKh;
+  const bf16* Vp = V + b * sVb + hkv * sVh;
   const int kbase = kblock * BN;
-  const int krow_w = kbase + wid * 16;
+  const int krow_w = kbase + wid * 32;  // 2 sub-blocks of 16 keys
 
-  // K/V fragments for this wave's 16 keys (A-layout rows)
   constexpr int DK = D / 32;
-  bf16x8_t kfrag[DK], vfrag[DK];
-  {
-    const int r = krow_w + (lane & 15);
-    const int row = (r < S) ? r : (S - 1);
+  bf16x8_t kfrag[2][DK], vfrag[2][DK];
+#pragma unroll
+  for (int sb = 0; sb < 2; ++sb) {
+    const int r = krow_w + sb * 16 + (lane & 15);
+    const long row = (r < S) ? r : (S - 1);
 #pragma unroll
     for (int kk = 0; kk < DK; ++kk) {
-      *(int4*)&kfrag[kk] =
-          *(const int4*)(Kp + (long)row * D + kk * 32 + (lane >> 4) * 8);
-      *(int4*)&vfrag[kk] =
-          *(const int4*)(Vp + (long)row * D + kk * 32 + (lane >> 4) * 8);
+      *(int4*)&kfrag[sb][kk] =
+          *(const int4*)(Kp + row * sKs + kk * 32 + (lane >> 4) * 8);
+      *(int4*)&vfrag[sb][kk] =
+          *(const int4*)(Vp + row * sVs + kk * 32 + (lane >> 4) * 8);
     }
   }
 
   constexpr int DN = D / 16;
-  f32x4_t dvacc[DN], dkacc[DN];
+  f32x4_t dvacc[2][DN], dkacc[2][DN];
 #pragma unroll
-  for (int nj = 0; nj < DN; ++nj) {
-    dvacc[nj] = f32x4_t{0.f, 0.f, 0.f, 0.f};
-    dkacc[nj] = f32x4_t{0.f, 0.f, 0.f, 0.f};
-  }
+  for (int sb = 0; sb < 2; ++sb)
+#pragma unroll
+    for (int nj = 0; nj < DN; ++nj) {
+      dvacc[sb][nj] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+      dkacc[sb][nj] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+    }
+
+  const int wkey_min = krow_w;  // first key of this wave
 
   for (int g = 0; g < group; ++g) {
     const int hq = hkv * group + g;
-    const long qoff = ((long)b * HQ + hq) * S * D;
-    const bf16* Qp = Q + qoff;
-    const bf16* dOp = dO + qoff;
+    const bf16* Qp = Q + b * sQb + hq * sQh;
+    const bf16* dOp = dO + b * sDb + hq * sDh;
     const float* Lp = LSE + ((long)b * HQ + hq) * S;
     const float* Dp = DELTA + ((long)b * HQ + hq) * S;
 
-    const int ib0 = CAUSAL ? kblock : 0;  // BM == BN
+    const int ib0 = CAUSAL ? (kbase / BM) : 0;
     const int nqb = (S + BM - 1) / BM;
     for (int ib = ib0; ib < nqb; ++ib) {
       const int qbase = ib * BM;
-      // ---- stage Q, dO (row-major + transposed) ----
-      {
-        constexpr int VECS = BM * D / 8;
-        for (int t = threadIdx.x; t < VECS; t += 256) {
+      {  // stage Q, dO row-major (16B) + transposed (paired b32)
+        constexpr int KVECS = BM * D / 8;
+        for (int t = threadIdx.x; t < KVECS; t += 512) {
           const int row = t / (D / 8);
           const int col8 = (t % (D / 8)) * 8;
           const int gr = qbase + row;
-          int4 qv, dv;
-          if (gr < S) {
-            qv = *(const int4*)(Qp + (long)gr * D + col8);
-            dv = *(const int4*)(dOp + (long)gr * D + col8);
-          } else {
-            qv = int4{0, 0, 0, 0};
-            dv = int4{0, 0, 0, 0};
-          }
+          int4 qv = (gr < S) ? *(const int4*)(Qp + (long)gr * sQs + col8)
+                             : int4{0, 0, 0, 0};
+          int4 dv = (gr < S) ? *(const int4*)(dOp + (long)gr * sDs + col8)
+                             : int4{0, 0, 0, 0};
           *(int4*)&q_lds[row * KP + col8] = qv;
           *(int4*)&do_lds[row * KP + col8] = dv;
-          const __bf16* qe = (const __bf16*)&qv;
-          const __bf16* de = (const __bf16*)&dv;
+        }
+        constexpr int TVECS = BM * D / 16;  // row pairs
+        for (int t = threadIdx.x; t < TVECS; t += 512) {
+          const int row = (t / (D / 8)) * 2;
+          const int col8 = (t % (D / 8)) * 8;
+          const int g0 = qbase + row, g1 = g0 + 1;
+          int4 q0 = (g0 < S) ? *(const int4*)(Qp + (long)g0 * sQs + col8)
+                             : int4{0, 0, 0, 0};
+          int4 q1 = (g1 < S) ? *(const int4*)(Qp + (long)g1 * sQs + col8)
+                             : int4{0, 0, 0, 0};
+          int4 d0 = (g0 < S) ? *(const int4*)(dOp + (long)g0 * sDs + col8)
+                             : int4{0, 0, 0, 0};
+          int4 d1 = (g1 < S) ? *(const int4*)(dOp + (long)g1 * sDs + col8)
+                             : int4{0, 0, 0, 0};
+          const __bf16 *qe0 = (const __bf16*)&q0, *qe1 = (const __bf16*)&q1;
+          const __bf16 *de0 = (const __bf16*)&d0, *de1 = (const __bf16*)&d1;
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
-            qt_lds[(col8 + j) * VP + row] = qe[j];
-            dot_lds[(col8 + j) * VP + row] = de[j];
+            __bf16 qp[2] = {qe0[j], qe1[j]};
+            __bf16 dp[2] = {de0[j], de1[j]};
+            *(uint*)&qt_lds[(col8 + j) * VP + row] = *(uint*)qp;
+            *(uint*)&dot_lds[(col8 + j) * VP + row] = *(uint*)dp;
           }
         }
       }
       __syncthreads();
 
-      // ---- per q-chunk: S^T, P^T, dP^T, dS^T ----
-      __bf16* pw = &pT_lds[wid * 16 * VP];
-      __bf16* dw = &dsT_lds[wid * 16 * VP];
+      // causal: skip waves whose keys are all above this q tile
+      if (!CAUSAL || wkey_min <= qbase + BM - 1) {
+        __bf16* pw = &pT_lds[wid * 32 * VP];
+        __bf16* dw = &dsT_lds[wid * 32 * VP];
 #pragma unroll
-      for (int nq = 0; nq < 4; ++nq) {
-        f32x4_t st = f32x4_t{0.f, 0.f, 0.f, 0.f};
-        f32x4_t dpt = f32x4_t{0.f, 0.f, 0.f, 0.f};
+        for (int nq = 0; nq < 4; ++nq) {
+          f32x4_t st[2], dpt[2];
+          st[0] = st[1] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+          dpt[0] = dpt[1] = f32x4_t{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-        for (int kk = 0; kk < DK; ++kk) {
-          bf16x8_t qb =
-              load_frag_b_rowmajorT(&q_lds[nq * 16 * KP], KP, kk * 32, lane);
-          st = MFMA_16x16x32(kfrag[kk], qb, st);
-          bf16x8_t db =
-              load_frag_b_rowmajorT(&do_lds[nq * 16 * KP], KP, kk * 32, lane);
-          dpt = MFMA_16x16x32(vfrag[kk], db, dpt);
+          for (int kk = 0; kk < DK; ++kk) {
+            bf16x8_t qb =
+                load_frag_b_rowmajorT(&q_lds[nq * 16 * KP], KP, kk * 32, lane);
+            st[0] = MFMA_16x16x32(kfrag[0][kk], qb, st[0]);
+            st[1] = MFMA_16x16x32(kfrag[1][kk], qb, st[1]);
+            bf16x8_t db =
+                load_frag_b_rowmajorT(&do_lds[nq * 16 * KP], KP, kk * 32, lane);
+            dpt[0] = MFMA_16x16x32(vfrag[0][kk], db, dpt[0]);
+            dpt[1] = MFMA_16x16x32(vfrag[1][kk], db, dpt[1]);
+          }
+          const int qcol = qbase + nq * 16 + (lane & 15);
+          const float lse = (qcol < S) ? Lp[qcol] : 1e30f;
+          const float delta = (qcol < S) ? Dp[qcol] : 0.f;
+#pragma unroll
+          for (int sb = 0; sb < 2; ++sb) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              const int krow = krow_w + sb * 16 + (lane >> 4) * 4 + r;
+              const bool dead =
+                  (qcol >= S) || (krow >= S) || (CAUSAL && qcol < krow);
+              const float p = dead ? 0.f : __expf(st[sb][r] * scale - lse);
+              const float ds = p * (dpt[sb][r] - delta) * scale;
+              const int lrow = sb * 16 + (lane >> 4) * 4 + r;
+              pw[lrow * VP + nq * 16 + (lane & 15)] = (__bf16)p;
+              dw[lrow * VP + nq * 16 + (lane & 15)] = (__bf16)ds;
+            }
+          }
         }
-        const int qcol = qbase + nq * 16 + (lane & 15);
-        const float lse = (qcol < S) ? Lp[qcol] : 1e30f;
-        const float delta = (qcol < S) ? Dp[qcol] : 0.f;
+        // dV += P^T dO ; dK += dS^T Q (B-frags shared by both sub-blocks)
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int krow = krow_w + (lane >> 4) * 4 + r;
-          float s = st[r] * scale;
-          const bool dead =
-              (qcol >= S) || (krow >= S) || (CAUSAL && qcol < krow);
-          const float p = dead ? 0.f : __expf(s - lse);
-          const float ds = p * (dpt[r] - delta) * scale;
-          pw[((lane >> 4) * 4 + r) * VP + nq * 16 + (lane & 15)] = (__bf16)p;
-          dw[((lane >> 4) * 4 + r) * VP + nq * 16 + (lane & 15)] = (__bf16)ds;
-        }
-      }
-      // own-wave LDS writes are ordered before own-wave reads; tiles
-      // (q_lds etc.) are stable until the barrier at loop end.
-
-      // ---- dV += P^T dO ; dK += dS^T Q ----
+        for (int nj = 0; nj < DN; ++nj) {
 #pragma unroll
-      for (int nj = 0; nj < DN; ++nj) {
-#pragma unroll
-        for (int kk = 0; kk < BM / 32; ++kk) {
-          bf16x8_t pa = load_frag_a(pw, VP, kk * 32, lane);
-          bf16x8_t dob =
-              load_frag_b_rowmajorT(&dot_lds[nj * 16 * VP], VP, kk * 32, lane);
-          dvacc[nj] = MFMA_16x16x32(pa, dob, dvacc[nj]);
-          bf16x8_t da = load_frag_a(dw, VP, kk * 32, lane);
-          bf16x8_t qb2 =
-              load_frag_b_rowmajorT(&qt_lds[nj * 16 * VP], VP, kk * 32, lane);
-          dkacc[nj] = MFMA_16x16x32(da, qb2, dkacc[nj]);
+          for (int kk = 0; kk < BM / 32; ++kk) {
+            bf16x8_t dob = load_frag_b_rowmajorT(&dot_lds[nj * 16 * VP], VP,
+                                                 kk * 32, lane);
+            bf16x8_t pa0 = load_frag_a(pw, VP, kk * 32, lane);
+            bf16x8_t pa1 = load_frag_a(pw + 16 * VP, VP, kk * 32, lane);
+            dvacc[0][nj] = MFMA_16x16x32(pa0, dob, dvacc[0][nj]);
+            dvacc[1][nj] = MFMA_16x16x32(pa1, dob, dvacc[1][nj]);
+            bf16x8_t qb2 = load_frag_b_rowmajorT(&qt_lds[nj * 16 * VP], VP,
+                                                 kk * 32, lane);
+            bf16x8_t da0 = load_frag_a(dw, VP, kk * 32, lane);
+            bf16x8_t da1 = load_frag_a(dw + 16 * VP, VP, kk * 32, lane);
+            dkacc[0][nj] = MFMA_16x16x32(da0, qb2, dkacc[0][nj]);
+            dkacc[1][nj] = MFMA_16x16x32(da1, qb2, dkacc[1][nj]);
+          }
         }
       }
       __syncthreads();
     }
   }
 
-  // ---- store dK, dV ----
-  bf16* dKp = dK + kvoff;
-  bf16* dVp = dV + kvoff;
+  // ---- store dK, dV as [s, b, hkv, d] contiguous ----
+  const long sOs = (long)Bb * HKV * D;
+  bf16* dKp = dK + ((long)b * HKV + hkv) * D;
+  bf16* dVp = dV + ((long)b * HKV + hkv) * D;
 #pragma unroll
-  for (int nj = 0; nj < DN; ++nj) {
+  for (int sb = 0; sb < 2; ++sb) {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int krow = krow_w + (lane >> 4) * 4 + r;
-      if (krow < S) {
-        dKp[(long)krow * D + nj * 16 + (lane & 15)] = f2bf(dkacc[nj][r]);
-        dVp[(long)krow * D + nj * 16 + (lane & 15)] = f2bf(dvacc[nj][r]);
+    for (int nj = 0; nj < DN; ++nj) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int krow = krow_w + sb * 16 + (lane >> 4) * 4 + r;
+        if (krow < S) {
+          dKp[(long)krow * sOs + nj * 16 + (lane & 15)] =
+              f2bf(dkacc[sb][nj][r]);
+          dVp[(long)krow * sOs + nj * 16 + (lane & 15)] =
+              f2bf(dvacc[sb][nj][r]);
+        }
       }
     }
   }
@@ -179,18 +209,21 @@ __global__ __launch_bounds__(256) void flash_bwd_dkv_kernel(
 
 // ============================ dQ ============================
 template <int D, bool CAUSAL>
-__global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
+__global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
     const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     const bf16* __restrict__ K, const bf16* __restrict__ V,
     const float* __restrict__ LSE, const float* __restrict__ DELTA,
-    bf16* __restrict__ dQ, int S, int HQ, int HKV, float scale) {
-  constexpr int BM = 64, BN = 64;
+    bf16* __restrict__ dQ,  // [s, b, hq, d] contiguous
+    int S, int Bb, int HQ, int HKV, float scale, long sQs, long sQb, long sQh,
+    long sKs, long sKb, long sKh, long sVs, long sVb, long sVh, long sDs,
+    long sDb, long sDh) {
+  constexpr int BM = 256, BN = 64;
   constexpr int KP = D + 8;
   constexpr int VP = BN + 8;
-  __shared__ __bf16 k_lds[BN * KP];       // K rows   (B for S)
-  __shared__ __bf16 v_lds[BN * KP];       // V rows   (B for dP)
-  __shared__ __bf16 kt_lds[D * VP];       // K^T      (B for dQ)
-  __shared__ __bf16 ds_lds[4 * 16 * VP];  // per-wave dS (A for dQ)
+  __shared__ __bf16 k_lds[BN * KP];
+  __shared__ __bf16 v_lds[BN * KP];
+  __shared__ __bf16 kt_lds[D * VP];
+  __shared__ __bf16 ds_lds[8 * 32 * VP];
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -200,141 +233,169 @@ __global__ __launch_bounds__(256) void flash_bwd_dq_kernel(
   const int b = bh / HQ;
   const int hkv = hq / (HQ / HKV);
 
-  const long qoff = ((long)b * HQ + hq) * S * D;
-  const long kvoff = ((long)b * HKV + hkv) * S * D;
-  const bf16* Qp = Q + qoff;
-  const bf16* dOp = dO + qoff;
-  const bf16* Kp = K + kvoff;
-  const bf16* Vp = V + kvoff;
+  const bf16* Qp = Q + b * sQb + hq * sQh;
+  const bf16* dOp = dO + b * sDb + hq * sDh;
+  const bf16* Kp = K + b * sKb + hkv * sKh;
+  const bf16* Vp = V + b * sVb + hkv * sVh;
   const float* Lp = LSE + ((long)b * HQ + hq) * S;
   const float* Dp = DELTA + ((long)b * HQ + hq) * S;
 
   const int q0 = qblock * BM;
-  const int qrow_w = q0 + wid * 16;
+  const int qrow_w = q0 + wid * 32;
 
   constexpr int DK = D / 32;
-  bf16x8_t qfrag[DK], dofrag[DK];
-  {
-    const int r = qrow_w + (lane & 15);
-    const int row = (r < S) ? r : (S - 1);
+  bf16x8_t qfrag[2][DK], dofrag[2][DK];
+  float lse[2][4], delta[2][4];
+#pragma unroll
+  for (int sb = 0; sb < 2; ++sb) {
+    const int r = qrow_w + sb * 16 + (lane & 15);
+    const long row = (r < S) ? r : (S - 1);
 #pragma unroll
     for (int kk = 0; kk < DK; ++kk) {
-      *(int4*)&qfrag[kk] =
-          *(const int4*)(Qp + (long)row * D + kk * 32 + (lane >> 4) * 8);
-      *(int4*)&dofrag[kk] =
-          *(const int4*)(dOp + (long)row * D + kk * 32 + (lane >> 4) * 8);
+      *(int4*)&qfrag[sb][kk] =
+          *(const int4*)(Qp + row * sQs + kk * 32 + (lane >> 4) * 8);
+      *(int4*)&dofrag[sb][kk] =
+          *(const int4*)(dOp + row * sDs + kk * 32 + (lane >> 4) * 8);
     }
-  }
-  float lse[4], delta[4];
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int qrow = qrow_w + (lane >> 4) * 4 + r;
-    lse[r] = (qrow < S) ? Lp[qrow] : 1e30f;
-    delta[r] = (qrow < S) ? Dp[qrow] : 0.f;
+    for (int rr = 0; rr < 4; ++rr) {
+      const int qrow = qrow_w + sb * 16 + (lane >> 4) * 4 + rr;
+      lse[sb][rr] = (qrow < S) ? Lp[qrow] : 1e30f;
+      delta[sb][rr] = (qrow < S) ? Dp[qrow] : 0.f;
+    }
   }
 
   constexpr int DN = D / 16;
-  f32x4_t dqacc[DN];
+  f32x4_t dqacc[2][DN];
 #pragma unroll
-  for (int nj = 0; nj < DN; ++nj) dqacc[nj] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+  for (int sb = 0; sb < 2; ++sb)
+#pragma unroll
+    for (int nj = 0; nj < DN; ++nj) dqacc[sb][nj] = f32x4_t{0.f, 0.f, 0.f, 0.f};
 
   const int kend = CAUSAL ? min(S, q0 + BM) : S;
   const int nkb = (kend + BN - 1) / BN;
+  const int wrow_max = qrow_w + 31;
+
   for (int jb = 0; jb < nkb; ++jb) {
     const int kbase = jb * BN;
-    {
-      constexpr int VECS = BN * D / 8;
-      for (int t = threadIdx.x; t < VECS; t += 256) {
+    {  // stage K, V row-major + K^T paired
+      constexpr int KVECS = BN * D / 8;
+      for (int t = threadIdx.x; t < KVECS; t += 512) {
         const int row = t / (D / 8);
         const int col8 = (t % (D / 8)) * 8;
         const int gr = kbase + row;
-        int4 kv, vv;
-        if (gr < S) {
-          kv = *(const int4*)(Kp + (long)gr * D + col8);
-          vv = *(const int4*)(Vp + (long)gr * D + col8);
-        } else {
-          kv = int4{0, 0, 0, 0};
-          vv = int4{0, 0, 0, 0};
-        }
+        int4 kv = (gr < S) ? *(const int4*)(Kp + (long)gr * sKs + col8)
+                           : int4{0, 0, 0, 0};
+        int4 vv = (gr < S) ? *(const int4*)(Vp + (long)gr * sVs + col8)
+                           : int4{0, 0, 0, 0};
         *(int4*)&k_lds[row * KP + col8] = kv;
         *(int4*)&v_lds[row * KP + col8] = vv;
-        const __bf16* ke = (const __bf16*)&kv;
+      }
+      constexpr int TVECS = BN * D / 16;
+      for (int t = threadIdx.x; t < TVECS; t += 512) {
+        const int row = (t / (D / 8)) * 2;
+        const int col8 = (t % (D / 8)) * 8;
+        const int g0 = kbase + row, g1 = g0 + 1;
+        int4 k0 = (g0 < S) ? *(const int4*)(Kp + (long)g0 * sKs + col8)
+                           : int4{0, 0, 0, 0};
+        int4 k1 = (g1 < S) ? *(const int4*)(Kp + (long)g1 * sKs + col8)
+                           : int4{0, 0, 0, 0};
+        const __bf16 *e0 = (const __bf16*)&k0, *e1 = (const __bf16*)&k1;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) kt_lds[(col8 + j) * VP + row] = ke[j];
+        for (int j = 0; j < 8; ++j) {
+          __bf16 pr[2] = {e0[j], e1[j]};
+          *(uint*)&kt_lds[(col8 + j) * VP + row] = *(uint*)pr;
+        }
       }
     }
     __syncthreads();
 
-    __bf16* dsw = &ds_lds[wid * 16 * VP];
+    if (!CAUSAL || kbase <= wrow_max) {
+      __bf16* dsw = &ds_lds[wid * 32 * VP];
 #pragma unroll
-    for (int nk = 0; nk < 4; ++nk) {
-      f32x4_t st = f32x4_t{0.f, 0.f, 0.f, 0.f};
-      f32x4_t dpt = f32x4_t{0.f, 0.f, 0.f, 0.f};
+      for (int nk = 0; nk < 4; ++nk) {
+        f32x4_t st[2], dpt[2];
+        st[0] = st[1] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+        dpt[0] = dpt[1] = f32x4_t{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int kk = 0; kk < DK; ++kk) {
-        bf16x8_t kb =
-            load_frag_b_rowmajorT(&k_lds[nk * 16 * KP], KP, kk * 32, lane);
-        st = MFMA_16x16x32(qfrag[kk], kb, st);
-        bf16x8_t vb =
-            load_frag_b_rowmajorT(&v_lds[nk * 16 * KP], KP, kk * 32, lane);
-        dpt = MFMA_16x16x32(dofrag[kk], vb, dpt);
+        for (int kk = 0; kk < DK; ++kk) {
+          bf16x8_t kb =
+              load_frag_b_rowmajorT(&k_lds[nk * 16 * KP], KP, kk * 32, lane);
+          st[0] = MFMA_16x16x32(qfrag[0][kk], kb, st[0]);
+          st[1] = MFMA_16x16x32(qfrag[1][kk], kb, st[1]);
+          bf16x8_t vb =
+              load_frag_b_rowmajorT(&v_lds[nk * 16 * KP], KP, kk * 32, lane);
+          dpt[0] = MFMA_16x16x32(dofrag[0][kk], vb, dpt[0]);
+          dpt[1] = MFMA_16x16x32(dofrag[1][kk], vb, dpt[1]);
+        }
+        const int kcol = kbase + nk * 16 + (lane & 15);
+#pragma unroll
+        for (int sb = 0; sb < 2; ++sb) {
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int qrow = qrow_w + sb * 16 + (lane >> 4) * 4 + r;
+            const bool dead = (kcol >= S) || (CAUSAL && kcol > qrow);
+            const float p =
+                dead ? 0.f : __expf(st[sb][r] * scale - lse[sb][r]);
+            const float ds = p * (dpt[sb][r] - delta[sb][r]) * scale;
+            dsw[(sb * 16 + (lane >> 4) * 4 + r) * VP + nk * 16 +
+                (lane & 15)] = (__bf16)ds;
+          }
+        }
       }
-      const int kcol = kbase + nk * 16 + (lane & 15);
+#pragma unroll
+      for (int nj = 0; nj < DN; ++nj) {
+#pragma unroll
+        for (int kk = 0; kk < BN / 32; ++kk) {
+          bf16x8_t kb2 =
+              load_frag_b_rowmajorT(&kt_lds[nj * 16 * VP], VP, kk * 32, lane);
+          bf16x8_t da0 = load_frag_a(dsw, VP, kk * 32, lane);
+          bf16x8_t da1 = load_frag_a(dsw + 16 * VP, VP, kk * 32, lane);
+          dqacc[0][nj] = MFMA_16x16x32(da0, kb2, dqacc[0][nj]);
+          dqacc[1][nj] = MFMA_16x16x32(da1, kb2, dqacc[1][nj]);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  const long sOs = (long)Bb * HQ * D;
+  bf16* dQp = dQ + ((long)b * HQ + hq) * D;
+#pragma unroll
+  for (int sb = 0; sb < 2; ++sb)
+#pragma unroll
+    for (int nj = 0; nj < DN; ++nj)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int qrow = qrow_w + (lane >> 4) * 4 + r;
-        const bool dead = (kcol >= S) || (CAUSAL && kcol > qrow);
-        const float p = dead ? 0.f : __expf(st[r] * scale - lse[r]);
-        const float ds = p * (dpt[r] - delta[r]) * scale;
-        dsw[((lane >> 4) * 4 + r) * VP + nk * 16 + (lane & 15)] = (__bf16)ds;
+        const int qrow = qrow_w + sb * 16 + (lane >> 4) * 4 + r;
+        if (qrow < S)
+          dQp[(long)qrow * sOs + nj * 16 + (lane & 15)] =
+              f2bf(dqacc[sb][nj][r]);
       }
-    }
-
-    // dQ += dS K  (A = dS 16×BN, B = K BN×D via K^T image)
-#pragma unroll
-    for (int nj = 0; nj < DN; ++nj) {
-#pragma unroll
-      for (int kk = 0; kk < BN / 32; ++kk) {
-        bf16x8_t da = load_frag_a(dsw, VP, kk * 32, lane);
-        bf16x8_t kb2 =
-            load_frag_b_rowmajorT(&kt_lds[nj * 16 * VP], VP, kk * 32, lane);
-        dqacc[nj] = MFMA_16x16x32(da, kb2, dqacc[nj]);
-      }
-    }
-    __syncthreads();
-  }
-
-  bf16* dQp = dQ + qoff;
-#pragma unroll
-  for (int nj = 0; nj < DN; ++nj) {
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int qrow = qrow_w + (lane >> 4) * 4 + r;
-      if (qrow < S)
-        dQp[(long)qrow * D + nj * 16 + (lane & 15)] = f2bf(dqacc[nj][r]);
-    }
-  }
 }
 
 extern "C" {
 void launch_flash_bwd(const void* dout, const void* q, const void* k,
                       const void* v, const void* lse, const void* delta,
                       void* dq, void* dk, void* dv, int B, int HQ, int HKV,
-                      int S, int D, bool causal, float scale,
+                      int S, int D, bool causal, float scale, const long* qstr,
+                      const long* kstr, const long* vstr, const long* dostr,
                       hipStream_t stream) {
-  dim3 blk(256);
-  dim3 gkv((S + 63) / 64, B * HKV);
-  dim3 gq((S + 63) / 64, B * HQ);
+  dim3 blk(512);
+  dim3 gkv((S + 255) / 256, B * HKV);
+  dim3 gq((S + 255) / 256, B * HQ);
 #define CASE(DD, CC)                                                          \
   do {                                                                        \
    hipLaunchKernelGGL(( flash_bwd_dkv_kernel<DD, CC>), dim3(gkv), dim3(blk), 0, stream,                     \
         (const bf16*)dout, (const bf16*)q, (const bf16*)k, (const bf16*)v,    \
-        (const float*)lse, (const float*)delta, (bf16*)dk, (bf16*)dv, S, HQ,  \
-        HKV, scale);                                                          \
+        (const float*)lse, (const float*)delta, (bf16*)dk, (bf16*)dv, S, B,   \
+        HQ, HKV, scale, qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2], \
+        vstr[0], vstr[1], vstr[2], dostr[0], dostr[1], dostr[2]);             \
    hipLaunchKernelGGL(( flash_bwd_dq_kernel<DD, CC>), dim3(gq), dim3(blk), 0, stream,                       \
         (const bf16*)dout, (const bf16*)q, (const bf16*)k, (const bf16*)v,    \
-        (const float*)lse, (const float*)delta, (bf16*)dq, S, HQ, HKV,        \
-        scale);                                                               \
+        (const float*)lse, (const float*)delta, (bf16*)dq, S, B, HQ, HKV,     \
+        scale, qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2], vstr[0], \
+        vstr[1], vstr[2], dostr[0], dostr[1], dostr[2]);                      \
   } while (0)
   if (D == 128) {
     if (causal) CASE(128, true); else CASE(128, false);

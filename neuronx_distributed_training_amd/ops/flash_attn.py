@@ -44,9 +44,8 @@ class _FlashAttnFn(torch.autograd.Function):
     def forward(ctx, q, k, v, causal: bool, scale: float):
         kern = kernels_for(q)
         if kern is not None:
-            o, lse = kern.flash_attn_fwd(
-                q.contiguous(), k.contiguous(), v.contiguous(), causal, scale
-            )
+            # kernel takes arbitrary-strided [b, h, s, d] views (d contig)
+            o, lse = kern.flash_attn_fwd(q, k, v, causal, scale)
         else:
             o, lse = _cpu_ref_fwd(q, k, v, causal, scale)
         ctx.save_for_backward(q, k, v, o, lse)
@@ -60,7 +59,7 @@ class _FlashAttnFn(torch.autograd.Function):
         kern = kernels_for(q)
         if kern is not None:
             dq, dk, dv = kern.flash_attn_bwd(
-                do.contiguous(), q, k, v, o, lse, ctx.causal, ctx.scale
+                do, q, k, v, o, lse, ctx.causal, ctx.scale
             )
             return dq, dk, dv, None, None
         # CPU reference backward (fp32, explicit)

@@ -61,7 +61,7 @@ class _RopeFn(torch.autograd.Function):
         ctx.pos_offset = pos_offset
         k = kernels_for(x)
         if k is not None:
-            return k.rope_fwd(x.contiguous(), cos, sin, pos_offset)
+            return k.rope_fwd(x, cos, sin, pos_offset)
         s = x.size(-2)
         c = cos[pos_offset : pos_offset + s].to(x.dtype)
         sn = sin[pos_offset : pos_offset + s].to(x.dtype)
@@ -72,7 +72,7 @@ class _RopeFn(torch.autograd.Function):
         cos, sin = ctx.saved_tensors
         k = kernels_for(dy)
         if k is not None:
-            dx = k.rope_fwd(dy.contiguous(), cos, -sin, ctx.pos_offset)
+            dx = k.rope_fwd(dy if dy.stride(-1) == 1 else dy.contiguous(), cos, -sin, ctx.pos_offset)
         else:
             s = dy.size(-2)
             c = cos[ctx.pos_offset : ctx.pos_offset + s].to(dy.dtype)
