@@ -1,0 +1,266 @@
+"""Mixture-of-experts stack for MI355X.
+
+Capability parity with the reference's NxD MoE
+(``modules.moe.{model.MoE, expert_mlps.ExpertMLPs, routing.RouterTopK/
+RouterSinkhorn, loss_function.load_balancing_loss_func}``; call sites
+transformer.py:423-464, modeling_mixtral.py:342-374 — SURVEY.md §2.3):
+
+- RouterTopK (softmax→top-k) and RouterSinkhorn (sinkhorn iterations);
+- dropless expert compute (sort tokens by expert, one hipBLASLt GEMM per
+  local expert) and capacity-factor mode (drop overflow tokens);
+- expert parallelism: all-to-all token dispatch over the EP group
+  (RCCL over xGMI), experts sharded across EP ranks;
+- load-balancing auxiliary loss.
+
+Expert MLPs are SwiGLU (gate_up fused + down), matching the dense path.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..parallel import state as ps
+from ..ops import swiglu
+
+__all__ = ["RouterTopK", "RouterSinkhorn", "ExpertMLPs", "MoE",
+           "load_balancing_loss_func"]
+
+
+class _AllToAll(torch.autograd.Function):
+    """Autograd-aware variable-split all-to-all over the EP group."""
+
+    @staticmethod
+    def forward(ctx, x, out_splits, in_splits, group):
+        ctx.out_splits = out_splits
+        ctx.in_splits = in_splits
+        ctx.group = group
+        out = x.new_empty(int(sum(out_splits)), *x.shape[1:])
+        dist.all_to_all_single(out, x, out_splits, in_splits, group=group)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad):
+        out = grad.new_empty(int(sum(ctx.in_splits)), *grad.shape[1:])
+        dist.all_to_all_single(
+            out, grad.contiguous(), ctx.in_splits, ctx.out_splits, group=ctx.group
+        )
+        return out, None, None, None
+
+
+def load_balancing_loss_func(router_logits: torch.Tensor, num_experts: int,
+                             top_k: int) -> torch.Tensor:
+    """Switch-style aux loss: num_experts * sum_e f_e * P_e
+    (reference modeling_mixtral.py:872-888 contract)."""
+    probs = torch.softmax(router_logits.float(), dim=-1)  # [T, E]
+    _, sel = probs.topk(top_k, dim=-1)
+    onehot = torch.zeros_like(probs).scatter_(1, sel, 1.0)
+    tokens_per_expert = onehot.mean(0)  # f_e * top_k
+    prob_per_expert = probs.mean(0)     # P_e
+    return num_experts * (tokens_per_expert * prob_per_expert).sum() / top_k
+
+
+class RouterTopK(nn.Module):
+    def __init__(self, hidden_size: int, num_experts: int, top_k: int,
+                 dtype: torch.dtype = torch.float32, init_seed: Optional[int] = None):
+        super().__init__()
+        self.num_experts = num_experts
+        self.top_k = top_k
+        if init_seed is not None:
+            st = torch.random.get_rng_state()
+            torch.manual_seed(init_seed)
+        w = torch.empty(num_experts, hidden_size, dtype=torch.float32)
+        nn.init.normal_(w, std=0.02)
+        if init_seed is not None:
+            torch.random.set_rng_state(st)
+        self.weight = nn.Parameter(w.to(dtype))
+
+    def forward(self, x: torch.Tensor):
+        """x: [T, H] → (weights [T, k], indices [T, k], logits [T, E])."""
+        logits = F.linear(x, self.weight).float()
+        probs = torch.softmax(logits, dim=-1)
+        topw, topi = probs.topk(self.top_k, dim=-1)
+        topw = topw / topw.sum(dim=-1, keepdim=True)  # renormalize
+        return topw.to(x.dtype), topi, logits
+
+
+class RouterSinkhorn(RouterTopK):
+    """Sinkhorn-balanced routing (reference RouterSinkhorn contract)."""
+
+    def __init__(self, *args, n_iter: int = 3, **kw):
+        super().__init__(*args, **kw)
+        self.n_iter = n_iter
+
+    def forward(self, x: torch.Tensor):
+        logits = F.linear(x, self.weight).float()
+        with torch.no_grad():
+            cost = torch.exp(logits)
+            d0 = torch.ones(cost.size(0), device=cost.device)
+            d1 = torch.ones(cost.size(1), device=cost.device)
+            eps = 1e-8
+            for _ in range(self.n_iter):
+                d0 = 1.0 / (cost @ d1 + eps)
+                d1 = cost.size(0) / cost.size(1) / (cost.t() @ d0 + eps)
+            balanced = cost * d0.unsqueeze(1) * d1.unsqueeze(0)
+            _, topi = balanced.topk(self.top_k, dim=-1)
+        # gradients flow through softmax of raw logits at chosen experts
+        probs = torch.softmax(logits, dim=-1)
+        topw = probs.gather(-1, topi)
+        topw = topw / topw.sum(dim=-1, keepdim=True)
+        return topw.to(x.dtype), topi, logits
+
+
+class ExpertMLPs(nn.Module):
+    """Local shard of the expert set: num_experts/ep SwiGLU MLPs.
+
+    Weights are stacked [E_local, ...] so the dropless path runs one
+    GEMM per local expert over its sorted token slab.
+    """
+
+    def __init__(self, num_experts: int, hidden_size: int,
+                 intermediate_size: int, dtype: torch.dtype = torch.float32,
+                 init_seed: Optional[int] = None):
+        super().__init__()
+        ep = ps.get_expert_model_parallel_world_size()
+        ep_rank = ps.get_expert_model_parallel_rank()
+        assert num_experts % ep == 0
+        self.num_experts = num_experts
+        self.num_local = num_experts // ep
+        self.local_offset = ep_rank * self.num_local
+        self.hidden_size = hidden_size
+        self.intermediate_size = intermediate_size
+
+        if init_seed is not None:
+            st = torch.random.get_rng_state()
+            torch.manual_seed(init_seed)
+        gu = torch.empty(num_experts, 2 * intermediate_size, hidden_size)
+        dn = torch.empty(num_experts, hidden_size, intermediate_size)
+        nn.init.normal_(gu, std=0.02)
+        nn.init.normal_(dn, std=0.02)
+        if init_seed is not None:
+            torch.random.set_rng_state(st)
+        sl = slice(self.local_offset, self.local_offset + self.num_local)
+        self.gate_up = nn.Parameter(gu[sl].to(dtype))
+        self.down = nn.Parameter(dn[sl].to(dtype))
+        self.gate_up.expert_model_parallel = True
+        self.down.expert_model_parallel = True
+
+    def forward(self, x: torch.Tensor, counts: torch.Tensor) -> torch.Tensor:
+        """x: [T, H] tokens sorted by local expert; counts: [E_local] token
+        counts per local expert. Returns same-order outputs."""
+        outs = []
+        start = 0
+        cl = counts.tolist()
+        for e in range(self.num_local):
+            n = cl[e]
+            if n == 0:
+                continue
+            xe = x[start : start + n]
+            h = F.linear(xe, self.gate_up[e])
+            h = swiglu(h)
+            outs.append(F.linear(h, self.down[e]))
+            start += n
+        if not outs:
+            return x.new_zeros(0, self.hidden_size)
+        return torch.cat(outs, dim=0)
+
+
+class MoE(nn.Module):
+    """Router + (optional EP all-to-all) + expert MLPs + combine.
+
+    Dropless by default; with ``capacity_factor`` set, tokens beyond
+    capacity per expert are dropped (their MoE output is 0 — residual
+    carries them, matching capacity-factor semantics).
+    """
+
+    def __init__(self, router: RouterTopK, experts: ExpertMLPs,
+                 capacity_factor: Optional[float] = None):
+        super().__init__()
+        self.router = router
+        self.experts = experts
+        self.capacity_factor = capacity_factor
+        self.ep = ps.get_expert_model_parallel_world_size()
+
+    def forward(self, x: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+        """x: [T, H] (caller flattens [s,b,h]). Returns (y [T, H],
+        router_logits [T, E])."""
+        T, H = x.shape
+        k = self.router.top_k
+        E = self.router.num_experts
+        topw, topi, logits = self.router(x)
+
+        # flatten (token, choice) pairs
+        flat_exp = topi.reshape(-1)                   # [T*k]
+        flat_tok = (
+            torch.arange(T, device=x.device).repeat_interleave(k)
+        )
+        flat_w = topw.reshape(-1)
+
+        if self.capacity_factor:
+            cap = int(math.ceil(self.capacity_factor * T * k / E))
+            # rank within each expert queue; drop beyond capacity
+            sort_by_e = torch.argsort(flat_exp, stable=True)
+            sorted_e = flat_exp[sort_by_e]
+            ones = torch.ones_like(sorted_e)
+            pos_in_e = torch.cumsum(ones, 0) - 1
+            seg_start = torch.searchsorted(sorted_e, torch.arange(E, device=x.device))
+            pos_in_e = pos_in_e - seg_start[sorted_e]
+            keep_sorted = pos_in_e < cap
+            keep = torch.zeros_like(keep_sorted)
+            keep[sort_by_e] = keep_sorted
+            flat_exp = flat_exp[keep]
+            flat_tok = flat_tok[keep]
+            flat_w = flat_w[keep]
+
+        order = torch.argsort(flat_exp, stable=True)
+        sorted_exp = flat_exp[order]
+        sorted_tok = flat_tok[order]
+        sorted_w = flat_w[order]
+        counts = torch.bincount(sorted_exp, minlength=E)  # [E]
+
+        xin = x[sorted_tok]  # gathered inputs, expert-sorted
+
+        if self.ep > 1:
+            group = ps.get_expert_model_parallel_group()
+            nl = self.experts.num_local
+            # tokens destined for rank r = those routed to experts [r*nl,(r+1)*nl)
+            send_counts = counts.view(self.ep, nl).sum(-1)
+            recv_counts = torch.empty_like(send_counts)
+            dist.all_to_all_single(recv_counts, send_counts, group=group)
+            in_splits = send_counts.tolist()
+            out_splits = recv_counts.tolist()
+            # per-(src, local_expert) counts
+            counts_mat = torch.empty(self.ep * nl, dtype=counts.dtype,
+                                     device=counts.device)
+            dist.all_to_all_single(counts_mat, counts.contiguous(), group=group)
+            counts_mat = counts_mat.view(self.ep, nl)
+
+            xrecv = _AllToAll.apply(xin.contiguous(), out_splits, in_splits, group)
+            # received rows arrive [src][expert]-ordered; resort expert-major
+            eids = torch.cat(
+                [
+                    torch.repeat_interleave(
+                        torch.arange(nl, device=x.device), counts_mat[r]
+                    )
+                    for r in range(self.ep)
+                ]
+            ) if xrecv.size(0) else torch.zeros(0, dtype=torch.long, device=x.device)
+            order2 = torch.argsort(eids, stable=True)
+            local_counts = counts_mat.sum(0)
+            yloc = self.experts(xrecv[order2], local_counts)
+            inv = torch.empty_like(order2)
+            inv[order2] = torch.arange(order2.numel(), device=x.device)
+            back = yloc[inv]
+            yexp = _AllToAll.apply(back.contiguous(), in_splits, out_splits, group)
+        else:
+            yexp = self.experts(xin, counts)
+
+        # combine: scatter-add weighted expert outputs back to tokens
+        y = torch.zeros_like(x)
+        y.index_add_(0, sorted_tok, yexp * sorted_w.unsqueeze(-1))
+        return y, logits

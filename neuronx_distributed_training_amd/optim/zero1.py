@@ -56,8 +56,19 @@ class ZeRO1AdamW:
             named_params = list(named_params)
         if named_params and not isinstance(named_params[0], tuple):
             named_params = [(f"param_{i}", p) for i, p in enumerate(named_params)]
-        self.named_params = [(n, p) for n, p in named_params if p.requires_grad]
-        assert self.named_params, "no trainable parameters"
+        all_named = [(n, p) for n, p in named_params if p.requires_grad]
+        assert all_named, "no trainable parameters"
+        # expert-parallel params are EP-sharded: their grads sync over the
+        # expert-DP group (ranks holding the same experts), NOT the full DP
+        # group — handled by a separate unsharded state buffer.
+        self.named_params = [
+            (n, p) for n, p in all_named
+            if not getattr(p, "expert_model_parallel", False)
+        ]
+        self.expert_named_params = [
+            (n, p) for n, p in all_named
+            if getattr(p, "expert_model_parallel", False)
+        ]
 
         self.dp_group = ps.get_data_parallel_group()
         self.dp_world = ps.get_data_parallel_world_size()
@@ -122,9 +133,32 @@ class ZeRO1AdamW:
             for (n, p), o in zip(self.named_params, offsets)
         ]
 
+        # ---- expert-parallel state (unsharded, synced over expert-DP) ----
+        self.expert_state = []
+        for n, p in self.expert_named_params:
+            m32 = p.detach().float().clone()
+            self.expert_state.append(
+                {
+                    "param": p,
+                    "master": m32,
+                    "exp_avg": torch.zeros_like(m32),
+                    "exp_avg_sq": torch.zeros_like(m32),
+                    "decay": not (p.ndim <= 1 or any(k in n.lower() for k in no_decay_keys)),
+                }
+            )
+        self.expert_dp_group = ps.get_expert_data_parallel_group()
+        self.expert_dp_world = max(
+            1, self.dp_world // ps.get_expert_model_parallel_world_size()
+        )
+
     # -- hooks the trainer uses --
     def zero_grad(self, set_to_none: bool = True):
         for _, p in self.named_params:
+            if set_to_none:
+                p.grad = None
+            elif p.grad is not None:
+                p.grad.zero_()
+        for _, p in self.expert_named_params:
             if set_to_none:
                 p.grad = None
             elif p.grad is not None:
@@ -159,9 +193,26 @@ class ZeRO1AdamW:
             dist.all_reduce(shard, group=cp_group)
             shard.div_(ps.get_context_model_parallel_world_size())
 
+        # 1b) expert grads: average over the expert-DP group
+        expert_grads = []
+        for st in self.expert_state:
+            p = st["param"]
+            g = (
+                p.grad.float()
+                if p.grad is not None
+                else torch.zeros_like(st["master"])
+            )
+            if self.expert_dp_world > 1 and self.expert_dp_group is not None:
+                dist.all_reduce(g, group=self.expert_dp_group)
+                g.div_(self.expert_dp_world)
+            expert_grads.append(g)
+
         # 2) global grad norm: count TP-sharded params on all ranks,
         #    replicated params only on tp rank 0; reduce over DP then TP/PP.
         sq = (shard * self.normmask_shard).pow(2).sum()
+        if expert_grads and ps.get_tensor_model_parallel_rank() == 0:
+            # each expert set appears expert_dp_world times across DP
+            sq = sq + sum(g.pow(2).sum() for g in expert_grads) / self.expert_dp_world
         if self.dp_world > 1:
             dist.all_reduce(sq, group=self.dp_group)
         if ps.get_tensor_model_parallel_world_size() > 1:
@@ -174,6 +225,8 @@ class ZeRO1AdamW:
             scale = self.grad_clip / (gnorm + 1e-6)
             scale = torch.clamp(scale, max=1.0)
             shard.mul_(scale)
+            for g in expert_grads:
+                g.mul_(scale)
 
         # 3) AdamW on the fp32 shard
         self.step_count += 1
@@ -204,6 +257,16 @@ class ZeRO1AdamW:
             )
             self.master_shard.addcdiv_(self.exp_avg, denom, value=-step_size)
 
+        # 3b) AdamW on expert state (identical on every expert-DP replica)
+        for st, g in zip(self.expert_state, expert_grads):
+            st["exp_avg"].mul_(b1).add_(g, alpha=1 - b1)
+            st["exp_avg_sq"].mul_(b2).addcmul_(g, g, value=1 - b2)
+            denom = (st["exp_avg_sq"] / (1 - b2 ** t)).sqrt_().add_(self.eps)
+            if st["decay"] and self.weight_decay:
+                st["master"].mul_(1.0 - self.lr * self.weight_decay)
+            st["master"].addcdiv_(st["exp_avg"], denom, value=-(self.lr / (1 - b1 ** t)))
+            st["param"].data.copy_(st["master"].to(st["param"].dtype))
+
         # 4) all-gather updated params in model dtype
         upd = self.master_shard.to(self.model_dtype)
         if self.dp_world > 1:
@@ -223,7 +286,7 @@ class ZeRO1AdamW:
 
     # -- checkpointing: per-DP-rank shard state --
     def state_dict(self) -> dict:
-        return {
+        sd = {
             "step_count": self.step_count,
             "lr": self.lr,
             "master_shard": self.master_shard,
@@ -233,6 +296,12 @@ class ZeRO1AdamW:
             "shard_size": self.shard_size,
             "total": self.total,
         }
+        if self.expert_state:
+            sd["expert"] = [
+                {k: v for k, v in st.items() if k != "param"}
+                for st in self.expert_state
+            ]
+        return sd
 
     def load_state_dict(self, sd: dict):
         assert sd["total"] == self.total, "optimizer layout mismatch"
@@ -242,6 +311,11 @@ class ZeRO1AdamW:
         self.master_shard.copy_(sd["master_shard"].to(self.device))
         self.exp_avg.copy_(sd["exp_avg"].to(self.device))
         self.exp_avg_sq.copy_(sd["exp_avg_sq"].to(self.device))
+        if sd.get("expert"):
+            for st, saved in zip(self.expert_state, sd["expert"]):
+                for k in ("master", "exp_avg", "exp_avg_sq"):
+                    st[k].copy_(saved[k].to(self.device))
+                st["param"].data.copy_(st["master"].to(st["param"].dtype))
         # restore params from masters so resume is exact
         upd = self.master_shard.to(self.model_dtype)
         if self.dp_world > 1:
