@@ -52,6 +52,11 @@ class BaseModelModule:
     def setup(self):
         self.model = self.build_model().to(self.device)
         self.model.train()
+        self.pp_engine = None
+        if ps.get_pipeline_model_parallel_world_size() > 1:
+            from .pipeline import PipelineEngine
+
+            self.pp_engine = PipelineEngine(self.model)
 
     def configure_optimizers(self, max_steps: int):
         ocfg = self.cfg["model"].get("optim", {})
@@ -97,6 +102,22 @@ class BaseModelModule:
 
     def forward_backward_step(self, microbatches) -> torch.Tensor:
         """Grad-accumulation loop; returns DP/CP-reduced mean loss."""
+        if self.pp_engine is not None:
+            # 1F1B over the stage module; batches stay on CPU until their
+            # microbatch runs (reference data/base.py:58-64 semantics)
+            mbs = [self.get_batch_on_this_context_parallel_rank(
+                       {k: v for k, v in b.items()}) for b in microbatches]
+            loss = self.pp_engine.run_train(mbs).float()
+            # loss lives on the last stage only; SUM over PP broadcasts it
+            dist.all_reduce(loss, group=ps.get_pipeline_model_parallel_group())
+            running = loss
+            if ps.get_data_parallel_world_size() > 1:
+                dist.all_reduce(running, group=ps.get_data_parallel_group())
+                running /= ps.get_data_parallel_world_size()
+            if ps.get_context_model_parallel_world_size() > 1:
+                dist.all_reduce(running, group=ps.get_context_model_parallel_group())
+                running /= ps.get_context_model_parallel_world_size()
+            return running
         running = torch.zeros((), dtype=torch.float32, device=self.device)
         n = self.num_microbatches
         for batch in microbatches:
@@ -146,6 +167,10 @@ class BaseModelModule:
             k: (v.to(self.device) if torch.is_tensor(v) else v) for k, v in batch.items()
         }
         batch = self.get_batch_on_this_context_parallel_rank(batch)
+        if self.pp_engine is not None:
+            loss = self.pp_engine.run_eval([batch]).float()
+            dist.all_reduce(loss, group=ps.get_pipeline_model_parallel_group())
+            return loss
         return self.model_fwd_calc_loss(batch).detach().float()
 
     @torch.no_grad()
@@ -205,4 +230,8 @@ class LlamaModule(BaseModelModule):
             activation_checkpoint=mcfg.get("activation_checkpoint"),
             dtype=dtype,
         )
+        if ps.get_pipeline_model_parallel_world_size() > 1:
+            from ..models.llama_pipeline import LlamaStage
+
+            return LlamaStage(cfg, pipeline_cuts=dstr.get("pipeline_cuts"))
         return LlamaForCausalLM(cfg)

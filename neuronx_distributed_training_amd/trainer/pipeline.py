@@ -1,0 +1,166 @@
+"""Pipeline-parallel runtime: 1F1B schedule over RCCL/xGMI P2P.
+
+Replaces the reference's NxD pipeline engine contract
+(``nxd.initialize_parallel_model`` + ``model.run_train`` — SURVEY.md §2.3
+"Pipeline engine") the MI355X-native way: no graph tracing — the model is
+partitioned by layer index into explicit stage modules
+(models/llama_pipeline.py), activations/grads move with
+``dist.batch_isend_irecv``, and the standard 1F1B schedule runs in eager
+Python (microbatch streaming, loss on the last stage, tied-embedding grad
+all-reduce across the embedding group, reference module.py:80-120).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+
+from ..parallel import state as ps
+
+
+class _SendPool:
+    """Non-blocking sends (a blocking send deadlocks 1F1B: neighbor ranks
+    can both be in send — e.g. rank r sending the next activation while
+    rank r+1 sends the previous grad back). Holds (work, tensor) so the
+    buffer outlives the transfer; completed sends are pruned lazily."""
+
+    def __init__(self):
+        self.pending = []
+
+    def send(self, t: torch.Tensor, dst: int):
+        t = t.contiguous()
+        w = dist.isend(t, dst)
+        self.pending.append((w, t))
+        self.pending = [(w, t) for (w, t) in self.pending if not w.is_completed()]
+
+    def drain(self):
+        for w, _ in self.pending:
+            w.wait()
+        self.pending = []
+
+
+def _recv(shape, dtype, device, src: int) -> torch.Tensor:
+    t = torch.empty(shape, dtype=dtype, device=device)
+    dist.recv(t, src)
+    return t
+
+
+class PipelineEngine:
+    """Drives one stage module through the 1F1B schedule.
+
+    The stage module contract (models/llama_pipeline.py LlamaStage):
+      - ``forward(x)``: stage input → stage output (first stage: input_ids
+        [b, s] → hidden [s, b, h]; last stage: hidden → scalar loss)
+      - ``set_batch(batch)``: give the last stage labels/loss-mask (and the
+        first stage input_ids) for the CURRENT microbatch
+      - attributes ``hidden_shape_for(batch)`` and ``dtype``
+    """
+
+    def __init__(self, stage_module, grad_scale: float = 1.0):
+        self.stage = stage_module
+        self.rank = ps.get_pipeline_model_parallel_rank()
+        self.world = ps.get_pipeline_model_parallel_world_size()
+        self.prev = ps.get_pipeline_model_parallel_prev_rank()
+        self.next = ps.get_pipeline_model_parallel_next_rank()
+        self.is_first = self.rank == 0
+        self.is_last = self.rank == self.world - 1
+        self.device = next(stage_module.parameters()).device
+        self._sends = _SendPool()
+
+    def run_train(self, microbatches: List[Dict[str, torch.Tensor]]) -> torch.Tensor:
+        """Full 1F1B over the microbatch list; returns mean loss (last
+        stage; zeros elsewhere — caller broadcasts for logging)."""
+        M = len(microbatches)
+        num_warmup = min(self.world - self.rank - 1, M)
+        num_steady = M - num_warmup
+
+        in_store: List[Optional[torch.Tensor]] = []
+        out_store: List[Optional[torch.Tensor]] = []
+        losses = []
+        fwd_idx = 0
+        bwd_idx = 0
+
+        def fwd_one():
+            nonlocal fwd_idx
+            batch = microbatches[fwd_idx]
+            self.stage.set_batch(batch)
+            if self.is_first:
+                x = None
+                inp = None
+            else:
+                shape = self.stage.hidden_shape_for(batch)
+                inp = _recv(shape, self.stage.dtype, self.device, self.prev)
+                inp.requires_grad_(True)
+            out = self.stage(inp)
+            if self.is_last:
+                losses.append(out.detach())
+                out_scaled = out / M
+                out_store.append(out_scaled)
+            else:
+                self._sends.send(out.detach(), self.next)
+                out_store.append(out)
+            in_store.append(inp)
+            fwd_idx += 1
+
+        def bwd_one():
+            nonlocal bwd_idx
+            out = out_store[bwd_idx]
+            inp = in_store[bwd_idx]
+            if self.is_last:
+                torch.autograd.backward(out)
+            else:
+                dout = _recv(tuple(out.shape), out.dtype, self.device, self.next)
+                torch.autograd.backward(out, grad_tensors=dout)
+            if not self.is_first:
+                self._sends.send(inp.grad, self.prev)
+            out_store[bwd_idx] = None
+            in_store[bwd_idx] = None
+            bwd_idx += 1
+
+        for _ in range(num_warmup):
+            fwd_one()
+        for i in range(num_steady):
+            fwd_one()
+            bwd_one()
+        while bwd_idx < M:
+            bwd_one()
+        self._sends.drain()
+
+        self._sync_tied_embeddings()
+
+        if losses:
+            return torch.stack(losses).mean()
+        return torch.zeros((), device=self.device)
+
+    @torch.no_grad()
+    def run_eval(self, microbatches) -> torch.Tensor:
+        losses = []
+        for batch in microbatches:
+            self.stage.set_batch(batch)
+            if self.is_first:
+                inp = None
+            else:
+                shape = self.stage.hidden_shape_for(batch)
+                inp = _recv(shape, self.stage.dtype, self.device, self.prev)
+            out = self.stage(inp)
+            if self.is_last:
+                losses.append(out)
+            else:
+                self._sends.send(out, self.next)
+        self._sends.drain()
+        if losses:
+            return torch.stack(losses).mean()
+        return torch.zeros((), device=self.device)
+
+    def _sync_tied_embeddings(self):
+        """All-reduce tied word-embedding grads across first/last stage
+        (reference models/megatron/module.py:80-120)."""
+        w = getattr(self.stage, "tied_embedding_weight", None)
+        if w is None or self.world == 1:
+            return
+        grp = ps.get_embedding_group()
+        if grp is None or w.grad is None:
+            return
+        dist.all_reduce(w.grad, group=grp)

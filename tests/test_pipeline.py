@@ -1,0 +1,107 @@
+"""Pipeline parallelism: PP=2 loss parity with PP=1, 1F1B training step,
+tied embeddings grad sync (gloo, CPU)."""
+
+import torch
+import pytest
+
+from tests.distutils import run_distributed
+
+TINY = dict(
+    vocab_size=128,
+    hidden_size=64,
+    intermediate_size=128,
+    num_hidden_layers=4,
+    num_attention_heads=4,
+    num_key_value_heads=2,
+    max_position_embeddings=32,
+)
+
+
+def _pp_loss(rank, world, tie):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.llama import LlamaConfig
+    from neuronx_distributed_training_amd.models.llama_pipeline import LlamaStage
+    from neuronx_distributed_training_amd.trainer.pipeline import PipelineEngine
+
+    ps.initialize_model_parallel(pipeline_model_parallel_size=world)
+    torch.manual_seed(7)
+    cfg = LlamaConfig(**TINY, tie_word_embeddings=tie)
+    stage = LlamaStage(cfg)
+    eng = PipelineEngine(stage)
+    g = torch.Generator().manual_seed(99)
+    micro = []
+    for _ in range(4):
+        ids = torch.randint(0, 128, (1, 32), generator=g)
+        micro.append({"input_ids": ids, "labels": ids.clone()})
+    loss = eng.run_train(micro)
+    import torch.distributed as dist
+    loss = loss.float()
+    dist.all_reduce(loss, group=ps.get_pipeline_model_parallel_group())
+    gsum = sum(
+        float(p.grad.abs().sum()) for p in stage.parameters() if p.grad is not None
+    )
+    assert gsum > 0
+    return float(loss)
+
+
+def _ref_loss(rank, world, tie):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.llama import (
+        LlamaConfig, LlamaForCausalLM,
+    )
+
+    ps.initialize_model_parallel()
+    torch.manual_seed(7)
+    cfg = LlamaConfig(**TINY, tie_word_embeddings=tie)
+    model = LlamaForCausalLM(cfg)
+    g = torch.Generator().manual_seed(99)
+    total = 0.0
+    for _ in range(4):
+        ids = torch.randint(0, 128, (1, 32), generator=g)
+        total += float(model(ids, labels=ids))
+    return total / 4
+
+
+@pytest.mark.parametrize("tie", [False, True])
+def test_pp2_loss_matches_pp1(tie):
+    ref = run_distributed(_ref_loss, 1, tie)[0]
+    pp = run_distributed(_pp_loss, 2, tie)
+    assert abs(pp[0] - pp[1]) < 1e-6  # broadcast consistent
+    assert abs(pp[0] - ref) < 5e-3, (pp[0], ref)
+
+
+def _pp_train(rank, world):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+    from neuronx_distributed_training_amd.optim.zero1 import ZeRO1AdamW
+
+    ps.initialize_model_parallel(pipeline_model_parallel_size=world)
+    cfg = {
+        "data": {"global_batch_size": 4, "micro_batch_size": 1, "seq_length": 32},
+        "distributed_strategy": {"pipeline_model_parallel_size": world},
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 4, "num_attention_heads": 4, "num_kv_heads": 2,
+            "grad_clip": 1.0, "optim": {"lr": 5e-3, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(1)
+    mod = LlamaModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=10)
+    g = torch.Generator().manual_seed(5)
+    ids = torch.randint(0, 128, (1, 32), generator=g)
+    micro = [{"input_ids": ids, "labels": ids.clone()} for _ in range(4)]
+    losses = []
+    for _ in range(6):
+        m = mod.training_step(micro)
+        losses.append(m["reduced_train_loss"])
+    assert losses[-1] < losses[0], losses
+    return losses
+
+
+def test_pp2_training_decreases():
+    res = run_distributed(_pp_train, 2)
+    assert res[0] == res[1]
