@@ -1,0 +1,171 @@
+"""SFT packing/padding datasets, DPO/ORPO modules, LoRA."""
+
+import json
+import os
+
+import torch
+import pytest
+
+from tests.distutils import run_distributed
+
+
+def _write_sft(path, n=12):
+    with open(path, "w") as f:
+        for i in range(n):
+            f.write(json.dumps({"prompt": f"q{i} " * 5, "completion": f"a{i} " * 8}) + "\n")
+
+
+def _write_dpo(path, n=8):
+    with open(path, "w") as f:
+        for i in range(n):
+            f.write(
+                json.dumps(
+                    {"prompt": f"q{i} " * 4, "chosen": f"good{i} " * 6,
+                     "rejected": f"bad{i} " * 6}
+                )
+                + "\n"
+            )
+
+
+def test_packing_dataset(tmp_path):
+    from neuronx_distributed_training_amd.data.packing import ConcatDataset, PaddedDataset
+
+    samples = [
+        {"input_ids": list(range(2, 12)), "labels": list(range(2, 12))}
+        for _ in range(5)
+    ]
+    packed = ConcatDataset(samples, chunk_size=16, eos_token_id=1)
+    assert len(packed) >= 3
+    item = packed[0]
+    assert item["input_ids"].numel() == 16
+    assert item["loss_mask"].sum() > 0
+    padded = PaddedDataset(samples, max_length=16)
+    it = padded[0]
+    assert it["input_ids"].numel() == 16
+    assert it["loss_mask"][:10].all() and not it["loss_mask"][10:].any()
+
+
+def _dpo_cfg(tmpdir, path, mode="dpo"):
+    align = {"dpo": {"kl_beta": 0.1, "max_prompt_length": 16}} if mode == "dpo" else {
+        "orpo": {"beta": 0.1, "max_prompt_length": 16}
+    }
+    return {
+        "trainer": {"max_steps": 2},
+        "data": {
+            "kind": "alignment",
+            "dataset_path": path,
+            "global_batch_size": 4,
+            "micro_batch_size": 2,
+            "seq_length": 32,
+            "tokenizer": "bytes",
+        },
+        "distributed_strategy": {},
+        "model": {
+            "vocab_size": 256, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 2, "num_attention_heads": 4, "num_kv_heads": 2,
+            "grad_clip": 1.0, "optim": {"lr": 1e-3, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+        "model_alignment_strategy": align,
+    }
+
+
+def _dpo_run(rank, world, tmpdir, mode):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.alignment import DPOModule, ORPOModule
+    from neuronx_distributed_training_amd.data.datamodule import build_datamodule
+
+    ps.initialize_model_parallel()
+    torch.manual_seed(0)
+    path = os.path.join(tmpdir, "dpo.jsonl")
+    cfg = _dpo_cfg(tmpdir, path, mode)
+    module = (DPOModule if mode == "dpo" else ORPOModule)(cfg)
+    module.setup()
+    module.configure_optimizers(max_steps=4)
+    dm = build_datamodule(cfg)
+    dm.setup()
+    module.on_train_start(dm)
+    loader = iter(dm.train_dataloader())
+    micro = list(dm.microbatch_iterator(loader))
+    metrics = module.training_step(micro)
+    assert "reduced_train_loss" in metrics
+    if mode == "dpo":
+        assert "reward_accuracy" in metrics
+    return metrics["reduced_train_loss"]
+
+
+@pytest.mark.parametrize("mode", ["dpo", "orpo"])
+def test_dpo_orpo_train_step(tmp_path, mode):
+    path = os.path.join(str(tmp_path), "dpo.jsonl")
+    _write_dpo(path)
+    loss = run_distributed(_dpo_run, 1, str(tmp_path), mode)[0]
+    assert loss == loss  # finite
+
+
+def test_sft_datamodule(tmp_path):
+    from neuronx_distributed_training_amd.data.alignment import ModelAlignmentDataModule
+
+    path = os.path.join(str(tmp_path), "sft.jsonl")
+    _write_sft(path)
+    cfg = {
+        "data": {
+            "kind": "alignment", "dataset_path": path,
+            "global_batch_size": 2, "micro_batch_size": 1, "seq_length": 64,
+            "tokenizer": "bytes",
+        },
+        "model_alignment_strategy": {"sft": {"packing": True}},
+    }
+    dm = ModelAlignmentDataModule(cfg)
+    dm.setup()
+    item = dm.train_ds[0]
+    assert item["input_ids"].numel() == 64
+    # prompt tokens masked out of the loss
+    assert (item["loss_mask"] == 0).any()
+
+
+def _lora_run(rank, world):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.llama import LlamaConfig, LlamaForCausalLM
+    from neuronx_distributed_training_amd.modules.lora import (
+        LoraConfig, apply_lora, merge_lora,
+    )
+    from neuronx_distributed_training_amd.optim.zero1 import ZeRO1AdamW
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(3)
+    model = LlamaForCausalLM(
+        LlamaConfig(vocab_size=128, hidden_size=64, intermediate_size=128,
+                    num_hidden_layers=2, num_attention_heads=4,
+                    num_key_value_heads=2, max_position_embeddings=32)
+    )
+    n = apply_lora(model, LoraConfig(lora_rank=4, target_modules=["q_proj", "kv_proj", "o_proj", "gate_up_proj"]))
+    assert n > 0
+    trainable = [p for p in model.parameters() if p.requires_grad]
+    total = list(model.parameters())
+    assert len(trainable) < len(total)
+    ids = torch.randint(0, 128, (2, 32), generator=torch.Generator().manual_seed(1))
+    loss0 = model(ids, labels=ids)
+    opt = ZeRO1AdamW(
+        [(f"l{i}", p) for i, p in enumerate(trainable)], lr=1e-2, grad_clip=1.0
+    )
+    for _ in range(5):
+        opt.zero_grad()
+        loss = model(ids, labels=ids)
+        loss.backward()
+        opt.step()
+    assert float(loss) < float(loss0)
+    # merged model gives same output as adapter model (eval: dropout off)
+    model.eval()
+    with torch.no_grad():
+        before = float(model(ids, labels=ids))
+    merge_lora(model)
+    with torch.no_grad():
+        after = float(model(ids, labels=ids))
+    assert abs(before - after) < 1e-3, (before, after)
+    return after
+
+
+@pytest.mark.parametrize("world", [1, 2])
+def test_lora(world):
+    run_distributed(_lora_run, world)
