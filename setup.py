@@ -31,6 +31,9 @@ sources = [
     )
 ]
 
+import pybind11
+from setuptools import Extension
+
 setup(
     name="neuronx_distributed_training_amd_ext",
     ext_modules=[
@@ -41,7 +44,19 @@ setup(
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
             },
-        )
+        ),
+        Extension(
+            name="neuronx_distributed_training_amd.data._helpers_cpp",
+            sources=[
+                os.path.join(
+                    HERE, "neuronx_distributed_training_amd", "data",
+                    "csrc_cpu", "helpers.cpp",
+                )
+            ],
+            include_dirs=[pybind11.get_include()],
+            extra_compile_args=["-O3", "-std=c++17"],
+            language="c++",
+        ),
     ],
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
 )

@@ -1,0 +1,179 @@
+"""Megatron model family + data pipeline tests (CPU)."""
+
+import os
+
+import numpy as np
+import torch
+import pytest
+
+from tests.distutils import run_distributed
+
+
+def test_indexed_dataset_roundtrip(tmp_path):
+    from neuronx_distributed_training_amd.data.indexed_dataset import (
+        MMapIndexedDataset, MMapIndexedDatasetBuilder,
+    )
+
+    prefix = os.path.join(str(tmp_path), "corpus")
+    b = MMapIndexedDatasetBuilder(prefix, dtype=np.int32)
+    docs = [list(range(5, 25)), list(range(100, 140)), [7, 8, 9]]
+    for d in docs:
+        b.add_document(d)
+    b.finalize()
+    ds = MMapIndexedDataset(prefix)
+    assert len(ds) == 3
+    for i, d in enumerate(docs):
+        assert list(ds[i]) == d
+    assert list(ds.get(1, offset=5, length=4)) == [105, 106, 107, 108]
+
+
+def test_sample_idx_builders_agree(tmp_path):
+    from neuronx_distributed_training_amd.data.gpt_dataset import (
+        build_sample_idx_py, _helpers,
+    )
+
+    sizes = np.array([20, 40, 3, 17, 33], dtype=np.int32)
+    doc_idx = np.array([3, 1, 0, 4, 2, 0, 1, 3, 2, 4], dtype=np.int64)
+    tokens = int(sizes.sum())  # tokens per single epoch
+    py = build_sample_idx_py(sizes, doc_idx, 16, 2, tokens)
+    h = _helpers()
+    if h is not None:
+        cpp = h.build_sample_idx(sizes, doc_idx, 16, 2, tokens)
+        assert np.array_equal(py, np.asarray(cpp))
+    # each consecutive pair spans exactly seq_length+1 tokens
+    cum = np.concatenate([[0], np.cumsum(sizes[doc_idx])])
+    for i in range(len(py) - 1):
+        t0 = cum[py[i][0]] + py[i][1]
+        t1 = cum[py[i + 1][0]] + py[i + 1][1]
+        assert t1 - t0 == 16, (i, t1 - t0)
+
+
+def test_gpt_dataset_samples(tmp_path):
+    from neuronx_distributed_training_amd.data.indexed_dataset import (
+        MMapIndexedDatasetBuilder,
+    )
+    from neuronx_distributed_training_amd.data.gpt_dataset import (
+        build_train_valid_test_datasets,
+    )
+
+    prefix = os.path.join(str(tmp_path), "c2")
+    b = MMapIndexedDatasetBuilder(prefix)
+    rng = np.random.RandomState(0)
+    for _ in range(50):
+        b.add_document(rng.randint(0, 1000, size=rng.randint(10, 60)))
+    b.finalize()
+    tr, va, te = build_train_valid_test_datasets(
+        prefix, "80,10,10", seq_length=32, train_samples=20, valid_samples=4,
+        test_samples=4, cache_dir=os.path.join(str(tmp_path), "cache"),
+    )
+    assert len(tr) == 20
+    item = tr[0]
+    assert item["input_ids"].numel() == 32
+    # labels are inputs shifted by one within the token stream
+    i2 = tr[5]
+    assert (i2["labels"][:-1] == i2["input_ids"][1:]).all()
+
+
+def test_samplers_resume():
+    from neuronx_distributed_training_amd.data.samplers import (
+        MegatronPretrainingBatchSampler,
+        MegatronPretrainingRandomBatchSampler,
+    )
+
+    s = MegatronPretrainingBatchSampler(
+        total_samples=32, consumed_samples=0, micro_batch_size=2,
+        data_parallel_rank=0, data_parallel_size=2, global_batch_size=8,
+    )
+    batches = list(s)
+    assert batches[0] == [0, 1]
+    s1 = MegatronPretrainingBatchSampler(
+        total_samples=32, consumed_samples=8, micro_batch_size=2,
+        data_parallel_rank=1, data_parallel_size=2, global_batch_size=8,
+    )
+    assert list(s1)[0] == [10, 11]
+    r = MegatronPretrainingRandomBatchSampler(
+        total_samples=32, consumed_samples=0, micro_batch_size=2,
+        data_parallel_rank=0, data_parallel_size=2, global_batch_size=8,
+        seed=7,
+    )
+    r2 = MegatronPretrainingRandomBatchSampler(
+        total_samples=32, consumed_samples=4, micro_batch_size=2,
+        data_parallel_rank=0, data_parallel_size=2, global_batch_size=8,
+        seed=7,
+    )
+    assert list(r)[1:] == list(r2)  # resume skips consumed
+
+
+def _gpt_train(rank, world, tp, block_type, pos_type, activation):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.megatron_gpt import (
+        GPTConfig, GPTModel,
+    )
+    from neuronx_distributed_training_amd.optim.zero1 import ZeRO1AdamW
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=tp)
+    torch.manual_seed(0)
+    cfg = GPTConfig(
+        vocab_size=128, hidden_size=64, num_layers=2, num_attention_heads=4,
+        num_kv_heads=2, max_position_embeddings=32,
+        transformer_block_type=block_type, position_embedding_type=pos_type,
+        activation=activation,
+        normalization="rmsnorm" if activation == "swiglu" else "layernorm",
+    )
+    model = GPTModel(cfg)
+    opt = ZeRO1AdamW(list(model.named_parameters()), lr=5e-3, grad_clip=1.0)
+    g = torch.Generator().manual_seed(5)
+    ids = torch.randint(0, 128, (2, 32), generator=g)
+    labels = torch.roll(ids, -1, dims=1)
+    losses = []
+    for _ in range(6):
+        opt.zero_grad()
+        loss = model(ids, labels=labels)
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0], losses
+    return losses[0]
+
+
+@pytest.mark.parametrize(
+    "block,pos,act",
+    [
+        ("pre_ln", "rope", "swiglu"),
+        ("post_ln", "learned_absolute", "gelu"),
+        ("pre_ln", "rope", "geglu"),
+    ],
+)
+def test_gpt_variants_train(block, pos, act):
+    run_distributed(_gpt_train, 1, 1, block, pos, act)
+
+
+def test_gpt_tp2_matches_tp1():
+    l1 = run_distributed(_gpt_train, 1, 1, "pre_ln", "rope", "swiglu")[0]
+    l2 = run_distributed(_gpt_train, 2, 2, "pre_ln", "rope", "swiglu")[0]
+    assert abs(l1 - l2) < 5e-3, (l1, l2)
+
+
+def _gpt_moe(rank, world):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.megatron_gpt import (
+        GPTConfig, GPTModel,
+    )
+
+    ps.initialize_model_parallel()
+    torch.manual_seed(1)
+    cfg = GPTConfig(
+        vocab_size=128, hidden_size=64, num_layers=2, num_attention_heads=4,
+        max_position_embeddings=32, num_moe_experts=4, moe_top_k=2,
+        moe_frequency=2,
+    )
+    model = GPTModel(cfg)
+    ids = torch.randint(0, 128, (2, 32))
+    loss = model(ids, labels=torch.roll(ids, -1, 1))
+    loss.backward()
+    assert torch.isfinite(loss)
+    return float(loss)
+
+
+def test_gpt_moe_layer():
+    run_distributed(_gpt_moe, 1)
