@@ -194,7 +194,13 @@ class LlamaAttention(nn.Module):
 
     def core_attention(self, q, k, v):
         """q/k/v: [b, h, s, d] bf16 → o [b, h, s, d]. The recompute unit for
-        selective activation checkpointing (reference CoreAttention)."""
+        selective activation checkpointing (reference CoreAttention).
+        Under context parallelism the flash kernel runs inside the CP ring
+        (reference ring-attention dispatch, modeling_llama.py:482-489)."""
+        if ps.get_context_model_parallel_world_size() > 1:
+            from ..ops.ring_attn import ring_flash_attn
+
+            return ring_flash_attn(q, k, v, scale=self.scale)
         return flash_attn_func(q, k, v, causal=True, scale=self.scale)
 
     def forward(self, x, cos, sin, pos_offset: int = 0):

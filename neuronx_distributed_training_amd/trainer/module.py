@@ -87,11 +87,27 @@ class BaseModelModule:
         )
 
     def get_batch_on_this_context_parallel_rank(self, batch):
-        """Split the sequence across the CP group (reference model/base.py:199)."""
+        """Split the sequence across the CP group (reference model/base.py:199).
+
+        Labels are shifted to next-token form BEFORE the split (the model
+        skips its own logit shift under CP, reference
+        modeling_llama.py:816-819), so chunk boundaries stay correct.
+        """
         cp = ps.get_context_model_parallel_world_size()
         if cp == 1:
             return batch
         r = ps.get_context_model_parallel_rank()
+        batch = dict(batch)
+        if "labels" in batch and torch.is_tensor(batch["labels"]):
+            lab = batch["labels"]
+            shifted = torch.roll(lab, -1, dims=1).clone()
+            mask = batch.get("loss_mask")
+            if mask is None:
+                mask = torch.ones_like(lab, dtype=torch.float32)
+            mask = torch.roll(mask, -1, dims=1).clone()
+            mask[:, -1] = 0.0  # no label for the final position
+            batch["labels"] = shifted
+            batch["loss_mask"] = mask
         out = {}
         for k, v in batch.items():
             if torch.is_tensor(v) and v.dim() >= 2 and v.size(1) == self.seq_length:

@@ -1,0 +1,84 @@
+"""Context parallelism: ring attention parity (fwd+bwd) and CP=2 training
+loss vs CP=1 (gloo, CPU)."""
+
+import torch
+import pytest
+
+from tests.distutils import run_distributed
+
+
+def _ring_attn(rank, world):
+    import torch.nn.functional as F
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.ops.ring_attn import ring_flash_attn
+
+    ps.initialize_model_parallel(context_parallel_size=world)
+    torch.manual_seed(0)
+    b, h, hkv, s, d = 2, 4, 2, 32, 16
+    qf = torch.randn(b, h, s, d)
+    kf = torch.randn(b, hkv, s, d)
+    vf = torch.randn(b, hkv, s, d)
+    g = torch.randn(b, h, s, d)
+    r = ps.get_context_model_parallel_rank()
+    sl = s // world
+    q = qf[:, :, r * sl : (r + 1) * sl].clone().requires_grad_(True)
+    k = kf[:, :, r * sl : (r + 1) * sl].clone().requires_grad_(True)
+    v = vf[:, :, r * sl : (r + 1) * sl].clone().requires_grad_(True)
+    o = ring_flash_attn(q, k, v)
+    o.backward(g[:, :, r * sl : (r + 1) * sl])
+
+    # full-sequence reference
+    qr = qf.clone().requires_grad_(True)
+    kr = kf.clone().requires_grad_(True)
+    vr = vf.clone().requires_grad_(True)
+    ref = F.scaled_dot_product_attention(
+        qr, kr.repeat_interleave(h // hkv, 1), vr.repeat_interleave(h // hkv, 1),
+        is_causal=True,
+    )
+    ref.backward(g)
+    assert torch.allclose(o, ref[:, :, r * sl : (r + 1) * sl], atol=1e-4), (
+        (o - ref[:, :, r * sl : (r + 1) * sl]).abs().max()
+    )
+    assert torch.allclose(q.grad, qr.grad[:, :, r * sl : (r + 1) * sl], atol=1e-4)
+    assert torch.allclose(k.grad, kr.grad[:, :, r * sl : (r + 1) * sl], atol=1e-4)
+    assert torch.allclose(v.grad, vr.grad[:, :, r * sl : (r + 1) * sl], atol=1e-4)
+    return 0.0
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_ring_attention_matches_full(world):
+    run_distributed(_ring_attn, world)
+
+
+def _cp_train_loss(rank, world):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+
+    ps.initialize_model_parallel(context_parallel_size=world)
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 2, "seq_length": 32},
+        "distributed_strategy": {"context_parallel_size": world},
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 2, "num_attention_heads": 4, "num_kv_heads": 2,
+            "grad_clip": 1.0, "optim": {"lr": 1e-3, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(3)
+    mod = LlamaModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=10)
+    g = torch.Generator().manual_seed(5)
+    ids = torch.randint(0, 128, (2, 32), generator=g)
+    micro = [{"input_ids": ids, "labels": ids.clone()}]
+    m = mod.training_step(micro)
+    return m["reduced_train_loss"]
+
+
+def test_cp2_loss_matches_cp1():
+    l1 = run_distributed(_cp_train_loss, 1)[0]
+    l2 = run_distributed(_cp_train_loss, 2)
+    assert abs(l2[0] - l2[1]) < 1e-6
+    assert abs(l1 - l2[0]) < 0.05, (l1, l2[0])
