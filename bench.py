@@ -14,7 +14,24 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import shutil
+import tempfile
 import time
+
+# hipBLASLt/rocBLAS algorithm selections pre-tuned on MI355X (TunableOp):
+# must be configured BEFORE torch import. Each rank copies the canonical
+# CSV to its device-ordinal filename.
+_REPO = os.path.dirname(os.path.abspath(__file__))
+_TUNE = os.path.join(_REPO, "tunableop", "tunableop_gfx950.csv")
+if os.path.exists(_TUNE) and os.environ.get("NXDT_DISABLE_TUNABLEOP") != "1":
+    _lr = os.environ.get("LOCAL_RANK", "0")
+    _td = tempfile.mkdtemp(prefix="tunableop_")
+    shutil.copy(_TUNE, os.path.join(_td, f"tunableop_gfx950{_lr}.csv"))
+    os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
+    os.environ.setdefault(
+        "PYTORCH_TUNABLEOP_FILENAME", os.path.join(_td, "tunableop_gfx950.csv")
+    )
 
 import torch
 import torch.distributed as dist

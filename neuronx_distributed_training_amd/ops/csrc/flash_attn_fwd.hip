@@ -75,6 +75,11 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
   const int nkb = (kend + BN - 1) / BN;
   const int wrow_max = qrow_w + 31;
 
+  // T5 static form: the younger dispatch half gets priority so it is not
+  // starved of VALU issue at segment starts (guide §5.5 T5).
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
   for (int jb = 0; jb < nkb; ++jb) {
     const int kbase = jb * BN;
     // ---- stage K [BN][D] (16B writes) + V^T [D][BN] (paired b32 writes) ----
@@ -127,19 +132,35 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
       __bf16* pw = &p_lds[wid * 32 * VP];
 #pragma unroll
       for (int sb = 0; sb < 2; ++sb) {
+        // interior tiles (every key visible to every row) skip the mask
+        const bool full_tile =
+            (kbase + BN <= S) &&
+            (!CAUSAL || (kbase + BN - 1 <= qrow_w + sb * 16));
         float tile_max[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
         float sv[4][4];
+        if (full_tile) {
 #pragma unroll
-        for (int nk = 0; nk < 4; ++nk) {
-          const int kcol = kbase + nk * 16 + (lane & 15);
+          for (int nk = 0; nk < 4; ++nk) {
 #pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            const int qrow = qrow_w + sb * 16 + (lane >> 4) * 4 + r;
-            float s = sacc[sb][nk][r] * scale;
-            const bool dead = (kcol >= S) || (CAUSAL && kcol > qrow);
-            s = dead ? -1e30f : s;
-            sv[nk][r] = s;
-            tile_max[r] = fmaxf(tile_max[r], s);
+            for (int r = 0; r < 4; ++r) {
+              const float s = sacc[sb][nk][r] * scale;
+              sv[nk][r] = s;
+              tile_max[r] = fmaxf(tile_max[r], s);
+            }
+          }
+        } else {
+#pragma unroll
+          for (int nk = 0; nk < 4; ++nk) {
+            const int kcol = kbase + nk * 16 + (lane & 15);
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              const int qrow = qrow_w + sb * 16 + (lane >> 4) * 4 + r;
+              float s = sacc[sb][nk][r] * scale;
+              const bool dead = (kcol >= S) || (CAUSAL && kcol > qrow);
+              s = dead ? -1e30f : s;
+              sv[nk][r] = s;
+              tile_max[r] = fmaxf(tile_max[r], s);
+            }
           }
         }
 #pragma unroll
