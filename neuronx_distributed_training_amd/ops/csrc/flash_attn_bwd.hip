@@ -21,10 +21,7 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
     bf16* __restrict__ dV, int S, int Bb, int HQ, int HKV, float scale,
     long sQs, long sQb, long sQh, long sKs, long sKb, long sKh, long sVs,
     long sVb, long sVh, long sDs, long sDb, long sDh) {
-  // 8 waves × 16 keys (no per-wave sub-blocking: the 2-sub-block variant
-  // allocates 256 VGPR + 72 B/lane scratch — the spill costs more than
-  // the halved B-fragment reads save).
-  constexpr int BN = 128;  // keys per block
+  constexpr int BN = 256;  // keys per block
   constexpr int BM = 64;   // q tile
   constexpr int KP = D + 8;
   constexpr int VP = BM + 8;
@@ -32,8 +29,8 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
   __shared__ __bf16 do_lds[BM * KP];
   __shared__ __bf16 dot_lds[D * VP];
   __shared__ __bf16 qt_lds[D * VP];
-  __shared__ __bf16 pT_lds[8 * 16 * VP];
-  __shared__ __bf16 dsT_lds[8 * 16 * VP];
+  __shared__ __bf16 pT_lds[8 * 32 * VP];
+  __shared__ __bf16 dsT_lds[8 * 32 * VP];
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -46,29 +43,32 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
   const bf16* Kp = K + b * sKb + hkv * sKh;
   const bf16* Vp = V + b * sVb + hkv * sVh;
   const int kbase = kblock * BN;
-  const int krow_w = kbase + wid * 16;
+  const int krow_w = kbase + wid * 32;  // 2 sub-blocks of 16 keys
 
   constexpr int DK = D / 32;
-  bf16x8_t kfrag[DK], vfrag[DK];
-  {
-    const int r = krow_w + (lane & 15);
+  bf16x8_t kfrag[2][DK], vfrag[2][DK];
+#pragma unroll
+  for (int sb = 0; sb < 2; ++sb) {
+    const int r = krow_w + sb * 16 + (lane & 15);
     const long row = (r < S) ? r : (S - 1);
 #pragma unroll
     for (int kk = 0; kk < DK; ++kk) {
-      *(int4*)&kfrag[kk] =
+      *(int4*)&kfrag[sb][kk] =
           *(const int4*)(Kp + row * sKs + kk * 32 + (lane >> 4) * 8);
-      *(int4*)&vfrag[kk] =
+      *(int4*)&vfrag[sb][kk] =
           *(const int4*)(Vp + row * sVs + kk * 32 + (lane >> 4) * 8);
     }
   }
 
   constexpr int DN = D / 16;
-  f32x4_t dvacc[DN], dkacc[DN];
+  f32x4_t dvacc[2][DN], dkacc[2][DN];
 #pragma unroll
-  for (int nj = 0; nj < DN; ++nj) {
-    dvacc[nj] = f32x4_t{0.f, 0.f, 0.f, 0.f};
-    dkacc[nj] = f32x4_t{0.f, 0.f, 0.f, 0.f};
-  }
+  for (int sb = 0; sb < 2; ++sb)
+#pragma unroll
+    for (int nj = 0; nj < DN; ++nj) {
+      dvacc[sb][nj] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+      dkacc[sb][nj] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+    }
 
   const int wkey_min = krow_w;  // first key of this wave
 
@@ -124,49 +124,59 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
 
       // causal: skip waves whose keys are all above this q tile
       if (!CAUSAL || wkey_min <= qbase + BM - 1) {
-        __bf16* pw = &pT_lds[wid * 16 * VP];
-        __bf16* dw = &dsT_lds[wid * 16 * VP];
+        __bf16* pw = &pT_lds[wid * 32 * VP];
+        __bf16* dw = &dsT_lds[wid * 32 * VP];
 #pragma unroll
         for (int nq = 0; nq < 4; ++nq) {
-          f32x4_t st = f32x4_t{0.f, 0.f, 0.f, 0.f};
-          f32x4_t dpt = f32x4_t{0.f, 0.f, 0.f, 0.f};
+          f32x4_t st[2], dpt[2];
+          st[0] = st[1] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+          dpt[0] = dpt[1] = f32x4_t{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
           for (int kk = 0; kk < DK; ++kk) {
             bf16x8_t qb =
                 load_frag_b_rowmajorT(&q_lds[nq * 16 * KP], KP, kk * 32, lane);
-            st = MFMA_16x16x32(kfrag[kk], qb, st);
+            st[0] = MFMA_16x16x32(kfrag[0][kk], qb, st[0]);
+            st[1] = MFMA_16x16x32(kfrag[1][kk], qb, st[1]);
             bf16x8_t db =
                 load_frag_b_rowmajorT(&do_lds[nq * 16 * KP], KP, kk * 32, lane);
-            dpt = MFMA_16x16x32(vfrag[kk], db, dpt);
+            dpt[0] = MFMA_16x16x32(vfrag[0][kk], db, dpt[0]);
+            dpt[1] = MFMA_16x16x32(vfrag[1][kk], db, dpt[1]);
           }
           const int qcol = qbase + nq * 16 + (lane & 15);
           const float lse = (qcol < S) ? Lp[qcol] : 1e30f;
           const float delta = (qcol < S) ? Dp[qcol] : 0.f;
 #pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            const int krow = krow_w + (lane >> 4) * 4 + r;
-            const bool dead =
-                (qcol >= S) || (krow >= S) || (CAUSAL && qcol < krow);
-            const float p = dead ? 0.f : __expf(st[r] * scale - lse);
-            const float ds = p * (dpt[r] - delta) * scale;
-            const int lrow = (lane >> 4) * 4 + r;
-            pw[lrow * VP + nq * 16 + (lane & 15)] = (__bf16)p;
-            dw[lrow * VP + nq * 16 + (lane & 15)] = (__bf16)ds;
+          for (int sb = 0; sb < 2; ++sb) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              const int krow = krow_w + sb * 16 + (lane >> 4) * 4 + r;
+              const bool dead =
+                  (qcol >= S) || (krow >= S) || (CAUSAL && qcol < krow);
+              const float p = dead ? 0.f : __expf(st[sb][r] * scale - lse);
+              const float ds = p * (dpt[sb][r] - delta) * scale;
+              const int lrow = sb * 16 + (lane >> 4) * 4 + r;
+              pw[lrow * VP + nq * 16 + (lane & 15)] = (__bf16)p;
+              dw[lrow * VP + nq * 16 + (lane & 15)] = (__bf16)ds;
+            }
           }
         }
-        // dV += P^T dO ; dK += dS^T Q
+        // dV += P^T dO ; dK += dS^T Q (B-frags shared by both sub-blocks)
 #pragma unroll
         for (int nj = 0; nj < DN; ++nj) {
 #pragma unroll
           for (int kk = 0; kk < BM / 32; ++kk) {
             bf16x8_t dob = load_frag_b_rowmajorT(&dot_lds[nj * 16 * VP], VP,
                                                  kk * 32, lane);
-            bf16x8_t pa = load_frag_a(pw, VP, kk * 32, lane);
-            dvacc[nj] = MFMA_16x16x32(pa, dob, dvacc[nj]);
+            bf16x8_t pa0 = load_frag_a(pw, VP, kk * 32, lane);
+            bf16x8_t pa1 = load_frag_a(pw + 16 * VP, VP, kk * 32, lane);
+            dvacc[0][nj] = MFMA_16x16x32(pa0, dob, dvacc[0][nj]);
+            dvacc[1][nj] = MFMA_16x16x32(pa1, dob, dvacc[1][nj]);
             bf16x8_t qb2 = load_frag_b_rowmajorT(&qt_lds[nj * 16 * VP], VP,
                                                  kk * 32, lane);
-            bf16x8_t da = load_frag_a(dw, VP, kk * 32, lane);
-            dkacc[nj] = MFMA_16x16x32(da, qb2, dkacc[nj]);
+            bf16x8_t da0 = load_frag_a(dw, VP, kk * 32, lane);
+            bf16x8_t da1 = load_frag_a(dw + 16 * VP, VP, kk * 32, lane);
+            dkacc[0][nj] = MFMA_16x16x32(da0, qb2, dkacc[0][nj]);
+            dkacc[1][nj] = MFMA_16x16x32(da1, qb2, dkacc[1][nj]);
           }
         }
       }
@@ -179,13 +189,18 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
   bf16* dKp = dK + ((long)b * HKV + hkv) * D;
   bf16* dVp = dV + ((long)b * HKV + hkv) * D;
 #pragma unroll
-  for (int nj = 0; nj < DN; ++nj) {
+  for (int sb = 0; sb < 2; ++sb) {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int krow = krow_w + (lane >> 4) * 4 + r;
-      if (krow < S) {
-        dKp[(long)krow * sOs + nj * 16 + (lane & 15)] = f2bf(dkacc[nj][r]);
-        dVp[(long)krow * sOs + nj * 16 + (lane & 15)] = f2bf(dvacc[nj][r]);
+    for (int nj = 0; nj < DN; ++nj) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int krow = krow_w + sb * 16 + (lane >> 4) * 4 + r;
+        if (krow < S) {
+          dKp[(long)krow * sOs + nj * 16 + (lane & 15)] =
+              f2bf(dkacc[sb][nj][r]);
+          dVp[(long)krow * sOs + nj * 16 + (lane & 15)] =
+              f2bf(dvacc[sb][nj][r]);
+        }
       }
     }
   }
@@ -366,7 +381,7 @@ void launch_flash_bwd(const void* dout, const void* q, const void* k,
                       const long* kstr, const long* vstr, const long* dostr,
                       hipStream_t stream) {
   dim3 blk(512);
-  dim3 gkv((S + 127) / 128, B * HKV);
+  dim3 gkv((S + 255) / 256, B * HKV);
   dim3 gq((S + 255) / 256, B * HQ);
 #define CASE(DD, CC)                                                          \
   do {                                                                        \
