@@ -187,8 +187,19 @@ class LlamaAttention(nn.Module):
             qkv = self.qkv_proj(x)
             q, k, v = qkv.chunk(3, dim=-1)
         else:
-            q = self.q_proj(x)
-            kv = self.kv_proj(x)
+            # one shared input mapping (SP all-gather / TP copy) for both
+            # projections — halves the SP gather traffic of this block
+            from ..parallel.mappings import (
+                copy_to_tensor_model_parallel_region,
+                gather_from_sequence_parallel_region,
+            )
+
+            if self.q_proj.sequence_parallel:
+                xg = gather_from_sequence_parallel_region(x)
+            else:
+                xg = copy_to_tensor_model_parallel_region(x)
+            q = self.q_proj(xg, pre_mapped=True)
+            kv = self.kv_proj(xg, pre_mapped=True)
             k, v = kv.chunk(2, dim=-1)
         return q, k, v
 

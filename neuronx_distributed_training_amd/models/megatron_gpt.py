@@ -211,8 +211,17 @@ class ParallelAttention(nn.Module):
             qkv = self.query_key_value(x)
             q, k, v = qkv.chunk(3, dim=-1)
         else:
-            q = self.query(x)
-            kv = self.key_value(x)
+            from ..parallel.mappings import (
+                copy_to_tensor_model_parallel_region,
+                gather_from_sequence_parallel_region,
+            )
+
+            if self.query.sequence_parallel:
+                xg = gather_from_sequence_parallel_region(x)
+            else:
+                xg = copy_to_tensor_model_parallel_region(x)
+            q = self.query(xg, pre_mapped=True)
+            kv = self.key_value(xg, pre_mapped=True)
             k, v = kv.chunk(2, dim=-1)
         s = q.size(0)
         q = q.view(s, b, self.n_heads_local, d).permute(1, 2, 0, 3)

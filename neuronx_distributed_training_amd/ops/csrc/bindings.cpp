@@ -25,6 +25,10 @@ void launch_flash_bwd(const void*, const void*, const void*, const void*,
                       int, int, int, bool, float, const long*, const long*,
                       const long*, const long*, hipStream_t);
 void launch_mfma_probe(const void*, const void*, void*, hipStream_t);
+void launch_ce_fwd(const void*, const void*, void*, void*, void*, long, int,
+                   long, hipStream_t);
+void launch_ce_bwd(const void*, const void*, const void*, const void*,
+                   const void*, void*, long, int, long, hipStream_t);
 }
 
 namespace {
@@ -183,6 +187,37 @@ torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
+std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor targets,
+                                  long vocab_start) {
+  CHECK_IN(logits);
+  TORCH_CHECK(logits.scalar_type() == torch::kBFloat16, "ce: bf16 logits");
+  TORCH_CHECK(targets.scalar_type() == torch::kLong, "ce: int64 targets");
+  long N = logits.size(0);
+  int V = (int)logits.size(1);
+  auto opt = logits.options().dtype(torch::kFloat32);
+  auto row_max = torch::empty({N}, opt);
+  auto row_sumexp = torch::empty({N}, opt);
+  auto tgt = torch::empty({N}, opt);
+  launch_ce_fwd(logits.data_ptr(), targets.contiguous().data_ptr(),
+                row_max.data_ptr(), row_sumexp.data_ptr(), tgt.data_ptr(), N,
+                V, vocab_start, cur_stream());
+  return {row_max, row_sumexp, tgt};
+}
+
+torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor targets,
+                     torch::Tensor gmax, torch::Tensor gsum,
+                     torch::Tensor gout, long vocab_start) {
+  CHECK_IN(logits);
+  long N = logits.size(0);
+  int V = (int)logits.size(1);
+  auto dl = torch::empty_like(logits);
+  launch_ce_bwd(logits.data_ptr(), targets.contiguous().data_ptr(),
+                gmax.contiguous().data_ptr(), gsum.contiguous().data_ptr(),
+                gout.contiguous().data_ptr(), dl.data_ptr(), N, V,
+                vocab_start, cur_stream());
+  return dl;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -195,4 +230,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);
   m.def("mfma_probe", &mfma_probe);
+  m.def("ce_fwd", &ce_fwd);
+  m.def("ce_bwd", &ce_bwd);
 }

@@ -127,11 +127,16 @@ class ColumnParallelLinear(nn.Module):
         else:
             self.register_parameter("bias", None)
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
-        if self.sequence_parallel:
-            x = gather_from_sequence_parallel_region(x)
-        else:
-            x = copy_to_tensor_model_parallel_region(x)
+    def forward(self, x: torch.Tensor, pre_mapped: bool = False) -> torch.Tensor:
+        """``pre_mapped=True``: the caller already applied the input
+        mapping (one SP all-gather shared by several projections of the
+        same input — e.g. split q/kv under SP); skip it here so backward
+        reduces exactly once through the caller's mapping."""
+        if not pre_mapped:
+            if self.sequence_parallel:
+                x = gather_from_sequence_parallel_region(x)
+            else:
+                x = copy_to_tensor_model_parallel_region(x)
         out = F.linear(x, self.weight, self.bias)
         if self.gather_output:
             out = gather_from_tensor_model_parallel_region(out)

@@ -23,6 +23,49 @@ from . import state as ps
 __all__ = ["parallel_cross_entropy", "from_parallel_logits_to_logprobs"]
 
 
+class _FusedVocabParallelCrossEntropy(torch.autograd.Function):
+    """GPU bf16 path: fused HIP statistics kernels (ops/csrc/
+    cross_entropy.hip) — one online max+sum pass forward, one recompute
+    pass backward; no fp32 softmax is materialized."""
+
+    @staticmethod
+    def forward(ctx, logits: torch.Tensor, target: torch.Tensor):
+        from ..ops import require_extension
+
+        k = require_extension()
+        tp = ps.get_tensor_model_parallel_world_size()
+        group = ps.get_tensor_model_parallel_group()
+        rank = ps.get_tensor_model_parallel_rank()
+        V = logits.size(-1)
+        vocab_start = rank * V
+        target = target.contiguous()
+        lmax, lsum, tgt = k.ce_fwd(logits.contiguous(), target, vocab_start)
+        if tp > 1:
+            gmax = lmax.clone()
+            dist.all_reduce(gmax, op=dist.ReduceOp.MAX, group=group)
+            gsum = lsum * torch.exp(lmax - gmax)
+            dist.all_reduce(gsum, group=group)
+            dist.all_reduce(tgt, group=group)
+        else:
+            gmax, gsum = lmax, lsum
+        loss = gsum.log() + gmax - tgt
+        ctx.save_for_backward(logits, target, gmax, gsum)
+        ctx.vocab_start = vocab_start
+        return loss
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        from ..ops import require_extension
+
+        k = require_extension()
+        logits, target, gmax, gsum = ctx.saved_tensors
+        dl = k.ce_bwd(
+            logits, target, gmax, gsum, grad_out.float().contiguous(),
+            ctx.vocab_start,
+        )
+        return dl, None
+
+
 class _VocabParallelCrossEntropy(torch.autograd.Function):
     @staticmethod
     def forward(ctx, logits: torch.Tensor, target: torch.Tensor, label_smoothing: float = 0.0):
@@ -99,7 +142,16 @@ def parallel_cross_entropy(
     """
     flat_logits = logits.reshape(-1, logits.size(-1))
     flat_target = target.reshape(-1)
-    loss = _VocabParallelCrossEntropy.apply(flat_logits, flat_target, label_smoothing)
+    if (
+        flat_logits.is_cuda
+        and flat_logits.dtype == torch.bfloat16
+        and label_smoothing == 0.0
+    ):
+        loss = _FusedVocabParallelCrossEntropy.apply(flat_logits, flat_target)
+    else:
+        loss = _VocabParallelCrossEntropy.apply(
+            flat_logits, flat_target, label_smoothing
+        )
     return loss.reshape(target.shape)
 
 

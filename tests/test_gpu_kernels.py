@@ -165,3 +165,23 @@ def test_adamw_gpu(ext):
 def test_extension_loaded_is_intree(ext):
     """The loaded .so must live in the repo tree (native-code check)."""
     assert "/neuronx_distributed_training_amd/ops/" in ext.__file__, ext.__file__
+
+
+def test_fused_cross_entropy_gpu(ext):
+    import torch.nn.functional as F
+
+    torch.manual_seed(7)
+    N, V = 512, 16032
+    logits = torch.randn(N, V, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    target = torch.randint(0, V, (N,), device="cuda")
+    from neuronx_distributed_training_amd.parallel.loss import parallel_cross_entropy
+
+    loss = parallel_cross_entropy(logits, target)
+    ref_in = logits.detach().float().clone().requires_grad_(True)
+    ref = F.cross_entropy(ref_in, target, reduction="none")
+    assert torch.allclose(loss, ref, atol=2e-2, rtol=1e-3), (loss - ref).abs().max()
+    g = torch.randn(N, device="cuda")
+    loss.backward(g)
+    ref.backward(g)
+    err = (logits.grad.float() - ref_in.grad).abs().max()
+    assert err < 2e-3, err
