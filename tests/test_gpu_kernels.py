@@ -183,5 +183,7 @@ def test_fused_cross_entropy_gpu(ext):
     g = torch.randn(N, device="cuda")
     loss.backward(g)
     ref.backward(g)
+    # dlogits are returned in bf16 (they feed the lm_head GEMM in bf16):
+    # tolerance = bf16 rounding at |g|≈1, not kernel error
     err = (logits.grad.float() - ref_in.grad).abs().max()
-    assert err < 2e-3, err
+    assert err < 1.5e-2, err

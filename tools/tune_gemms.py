@@ -70,6 +70,10 @@ def main():
               flush=True)
         del x, w, g
         torch.cuda.empty_cache()
+        # persist incrementally (a timeout must not lose finished tunings)
+        src = work + "0.csv"
+        if os.path.exists(src):
+            shutil.copy(src, canonical)
 
     # persist: torch writes on exit; also copy the per-device file back
     import atexit
