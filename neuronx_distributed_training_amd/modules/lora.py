@@ -72,16 +72,22 @@ class LoRAColumnParallelLinear(nn.Module):
             self.lora_A.tensor_model_parallel = True
             self.lora_A.partition_dim = 1
 
-    def forward(self, x):
+    @property
+    def sequence_parallel(self):
+        return self.base.sequence_parallel
+
+    def forward(self, x, pre_mapped: bool = False):
         b = self.base
         delta_in = self.dropout(x)
         if isinstance(b, ColumnParallelLinear):
-            if b.sequence_parallel:
+            if pre_mapped:
+                xg = x
+            elif b.sequence_parallel:
                 xg = gather_from_sequence_parallel_region(x)
             else:
                 xg = copy_to_tensor_model_parallel_region(x)
             out = F.linear(xg, b.weight, b.bias)
-            if b.sequence_parallel:
+            if not pre_mapped and b.sequence_parallel:
                 delta_in = gather_from_sequence_parallel_region(delta_in)
             out = out + F.linear(F.linear(delta_in, self.lora_A), self.lora_B) * self.scaling
             if b.gather_output:

@@ -249,5 +249,23 @@ class LlamaModule(BaseModelModule):
         if ps.get_pipeline_model_parallel_world_size() > 1:
             from ..models.llama_pipeline import LlamaStage
 
-            return LlamaStage(cfg, pipeline_cuts=dstr.get("pipeline_cuts"))
-        return LlamaForCausalLM(cfg)
+            model = LlamaStage(cfg, pipeline_cuts=dstr.get("pipeline_cuts"))
+        else:
+            model = LlamaForCausalLM(cfg)
+        peft = mcfg.get("peft")
+        if peft:
+            from ..modules.lora import LoraConfig, apply_lora
+
+            n = apply_lora(
+                model,
+                LoraConfig(
+                    lora_rank=int(peft.get("lora_rank", 16)),
+                    lora_alpha=float(peft.get("lora_alpha", 32)),
+                    lora_dropout=float(peft.get("lora_dropout", 0.05)),
+                    target_modules=list(
+                        peft.get("target_modules", ["q_proj", "kv_proj", "o_proj"])
+                    ),
+                ),
+            )
+            assert n > 0, "peft enabled but no target modules matched"
+        return model

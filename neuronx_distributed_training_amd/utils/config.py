@@ -29,7 +29,14 @@ def load_config(path: str, overrides: Optional[List[str]] = None) -> Dict:
 
 
 def _parse_scalar(v: str) -> Any:
-    return yaml.safe_load(v)
+    parsed = yaml.safe_load(v)
+    if isinstance(parsed, str):
+        # YAML 1.1 won't parse "1e-3" as a float; CLI overrides should
+        try:
+            return float(parsed)
+        except ValueError:
+            return parsed
+    return parsed
 
 
 def _set_dotted(cfg: Dict, key: str, value: Any):

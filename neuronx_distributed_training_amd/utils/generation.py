@@ -1,0 +1,41 @@
+"""Greedy/sampling generation over the native models (eval harness use;
+reference sft_evaluation generates on an inference backend — here the
+training model generates directly)."""
+
+from __future__ import annotations
+
+import torch
+
+from ..parallel import state as ps
+from ..parallel.mappings import gather_from_tensor_model_parallel_region
+
+
+@torch.no_grad()
+def generate(
+    model,
+    input_ids: torch.Tensor,
+    max_new_tokens: int = 32,
+    eos_token_id: int | None = None,
+    temperature: float = 0.0,
+    top_k: int = 0,
+) -> torch.Tensor:
+    """input_ids: [b, s]. Returns [b, s + new]. Full-recompute decode (no
+    KV cache) — intended for offline evaluation, not serving."""
+    model.eval()
+    ids = input_ids
+    for _ in range(max_new_tokens):
+        logits = model(ids)  # [b, s, V] (gathered over TP by the model)
+        nxt = logits[:, -1].float()
+        if temperature and temperature > 0:
+            nxt = nxt / temperature
+            if top_k:
+                v, _ = torch.topk(nxt, top_k)
+                nxt[nxt < v[:, [-1]]] = float("-inf")
+            probs = torch.softmax(nxt, dim=-1)
+            tok = torch.multinomial(probs, 1)
+        else:
+            tok = nxt.argmax(-1, keepdim=True)
+        ids = torch.cat([ids, tok], dim=1)
+        if eos_token_id is not None and bool((tok == eos_token_id).all()):
+            break
+    return ids

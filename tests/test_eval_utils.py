@@ -1,0 +1,96 @@
+"""Metrics registry, generation, MFU estimator, config system."""
+
+import json
+import os
+import subprocess
+import sys
+
+import torch
+
+
+def test_metrics():
+    from neuronx_distributed_training_amd.utils.metrics import MetricFactory
+
+    preds = ["the cat sat", "hello world", "42"]
+    labels = ["the cat sat", "world hello", "41"]
+    em = MetricFactory.create("exact_match").compute(preds, labels)
+    assert abs(em - 1 / 3) < 1e-6
+    f1 = MetricFactory.create("f1").compute(preds, labels)
+    assert 0.5 < f1 < 1.0
+    rl = MetricFactory.create("rouge_l").compute(preds, labels)
+    assert 0.0 < rl <= 1.0
+    acc = MetricFactory.create("accuracy").compute(["answer is 42 ok"], ["42"])
+    assert acc == 1.0
+
+
+def test_generation_tiny():
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.llama import (
+        LlamaConfig, LlamaForCausalLM,
+    )
+    from neuronx_distributed_training_amd.utils.generation import generate
+
+    ps.destroy_model_parallel()
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(
+        LlamaConfig(vocab_size=64, hidden_size=32, intermediate_size=64,
+                    num_hidden_layers=2, num_attention_heads=2,
+                    num_key_value_heads=1, max_position_embeddings=64)
+    )
+    ids = torch.randint(0, 64, (2, 8))
+    out = generate(model, ids, max_new_tokens=4)
+    assert out.shape == (2, 12)
+    assert (out[:, :8] == ids).all()
+
+
+def test_mfu_estimator():
+    from neuronx_distributed_training_amd.utils.llama_perf_estimate import (
+        LlamaShape, calculate_mfu, llama_flops_per_seq,
+    )
+
+    shape = LlamaShape.llama3_8b()
+    fwd = llama_flops_per_seq(shape)
+    # ~2N per token forward: N≈8e9 → per-seq fwd ≈ 2*8e9*8192 ≈ 1.3e14
+    assert 1.0e14 < fwd < 2.0e14, fwd
+    out = calculate_mfu(12000, shape, n_gpus=1)
+    assert 0.1 < out["mfu"] < 1.0
+
+
+def test_config_system(tmp_path):
+    import yaml
+    from neuronx_distributed_training_amd.utils.config import (
+        get_attribute_from_cfg, load_config,
+    )
+
+    p = os.path.join(str(tmp_path), "c.yaml")
+    cfg = {
+        "trainer": {"max_steps": 5},
+        "data": {"global_batch_size": 8, "micro_batch_size": 2, "seq_length": 16},
+        "distributed_strategy": {"tensor_model_parallel_size": 1,
+                                 "sequence_parallel": True},
+        "model": {"num_layers": 4},
+    }
+    yaml.safe_dump(cfg, open(p, "w"))
+    loaded = load_config(p, ["model.optim.lr=1e-3", "trainer.max_steps=7"])
+    assert loaded["trainer"]["max_steps"] == 7
+    assert loaded["model"]["optim"]["lr"] == 1e-3
+    # SP forced off at TP=1 (reference megatron_base_model.py:76-80)
+    assert loaded["distributed_strategy"]["sequence_parallel"] is False
+    assert get_attribute_from_cfg(loaded, "lr") == 1e-3
+
+
+def test_config_validation_rejects(tmp_path):
+    import pytest
+    import yaml
+    from neuronx_distributed_training_amd.utils.config import load_config
+
+    p = os.path.join(str(tmp_path), "bad.yaml")
+    yaml.safe_dump(
+        {
+            "data": {"global_batch_size": 7, "micro_batch_size": 2, "seq_length": 8},
+            "model": {},
+        },
+        open(p, "w"),
+    )
+    with pytest.raises(ValueError):
+        load_config(p)
