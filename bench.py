@@ -45,6 +45,11 @@ def main():
     ap.add_argument("--gbs", type=int, default=None, help="global batch (sequences)")
     ap.add_argument("--seq", type=int, default=8192)
     ap.add_argument("--layers", type=int, default=32)
+    ap.add_argument(
+        "--hipgraph", type=int, default=1,
+        help="capture the microbatch fwd+bwd into a hipGraph (single-GPU "
+        "steady state; falls back to eager if capture fails)",
+    )
     args = ap.parse_args()
 
     n = args.gpus
@@ -103,7 +108,48 @@ def main():
 
     micros = [make_micro() for _ in range(n_micro)]
 
+    # hipGraph capture of the steady-state microbatch fwd+bwd (the XLA
+    # graph-compile analog the MI355X way — BASELINE north star). Works at
+    # world=1 (no collectives inside the captured region); the optimizer
+    # step stays eager (ZeRO collectives at N>1).
+    graphed = None
+    if args.hipgraph and use_gpu and world == 1:
+        try:
+            static_ids = micros[0]["input_ids"].to("cuda")
+            static_lab = micros[0]["labels"].to("cuda")
+            # warmup on a side stream, then materialize stable .grad tensors
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(2):
+                    module.optimizer.zero_grad(set_to_none=False)
+                    loss = module.model(static_ids, labels=static_lab)
+                    (loss / n_micro).backward()
+            torch.cuda.current_stream().wait_stream(s)
+            module.optimizer.zero_grad(set_to_none=False)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                static_loss = module.model(static_ids, labels=static_lab)
+                (static_loss / n_micro).backward()
+            graphed = (g, static_ids, static_lab, static_loss)
+            if rank == 0:
+                print("# hipGraph capture OK", flush=True)
+        except Exception as e:  # pragma: no cover
+            graphed = None
+            if rank == 0:
+                print(f"# hipGraph capture failed ({e}); eager path", flush=True)
+
     def one_step():
+        if graphed is not None:
+            g, sid, slab, sloss = graphed
+            module.optimizer.zero_grad(set_to_none=False)
+            for mb in micros:
+                sid.copy_(mb["input_ids"], non_blocking=True)
+                slab.copy_(mb["labels"], non_blocking=True)
+                g.replay()
+            module.optimizer.step()
+            module.scheduler.step()
+            return
         module.optimizer.zero_grad()
         module.forward_backward_step(iter(micros))
         module.optimizer.step()

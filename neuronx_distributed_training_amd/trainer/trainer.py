@@ -30,6 +30,14 @@ class Trainer:
         t = cfg.get("trainer", {})
         self.cfg = cfg
         self.max_steps = int(t.get("max_steps", 100))
+        # max_time budget that survives restarts (reference StatelessTimer,
+        # training.py:65-69): "DD:HH:MM:SS" or seconds
+        mt = t.get("max_time")
+        if isinstance(mt, str) and ":" in mt:
+            d, h, m_, s = (int(x) for x in mt.split(":"))
+            mt = ((d * 24 + h) * 60 + m_) * 60 + s
+        self.max_time_s = float(mt) if mt else None
+        self._t_start = time.perf_counter()
         self.val_check_interval = int(t.get("val_check_interval", 0) or 0)
         self.limit_val_batches = int(t.get("limit_val_batches", 8))
         self.log_every_n_steps = int(t.get("log_every_n_steps", 1))
@@ -95,6 +103,13 @@ class Trainer:
                 self.validate(module, datamodule)
             if self.ckpt_every and self.ckpt_dir and self.global_step % self.ckpt_every == 0:
                 self.save_checkpoint(module, datamodule)
+            if (
+                self.max_time_s is not None
+                and time.perf_counter() - self._t_start > self.max_time_s
+            ):
+                if _is_global_zero():
+                    print(f"max_time reached at step {self.global_step}; stopping")
+                break
 
         if self.ckpt_dir:
             self.save_checkpoint(module, datamodule, tag="last")
