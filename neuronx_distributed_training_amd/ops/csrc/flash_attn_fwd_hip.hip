@@ -81,38 +81,65 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
   if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
     __builtin_amdgcn_s_setprio(1);
 
-  for (int jb = 0; jb < nkb; ++jb) {
-    const int kbase = jb * BN;
-    // ---- stage K [BN][D] (16B writes) + V^T [D][BN] (paired b32 writes) ----
-    {
-      constexpr int KVECS = BN * D / 8;
-      for (int t = threadIdx.x; t < KVECS; t += 512) {
-        const int row = t / (D / 8);
-        const int col8 = (t % (D / 8)) * 8;
-        const int gr = kbase + row;
-        int4 val = (gr < S) ? *(const int4*)(Kp + (long)gr * sKs + col8)
-                            : int4{0, 0, 0, 0};
-        *(int4*)&k_lds[row * KP + col8] = val;
-      }
-      constexpr int VVECS = BN * D / 16;  // row pairs
-      for (int t = threadIdx.x; t < VVECS; t += 512) {
-        const int row = (t / (D / 8)) * 2;
-        const int col8 = (t % (D / 8)) * 8;
-        const int g0 = kbase + row, g1 = g0 + 1;
-        int4 v0 = (g0 < S) ? *(const int4*)(Vp + (long)g0 * sVs + col8)
-                           : int4{0, 0, 0, 0};
-        int4 v1 = (g1 < S) ? *(const int4*)(Vp + (long)g1 * sVs + col8)
-                           : int4{0, 0, 0, 0};
-        const __bf16* e0 = (const __bf16*)&v0;
-        const __bf16* e1 = (const __bf16*)&v1;
+  // T14 async-stage split: issue tile t+1's global loads into registers
+  // BEFORE computing tile t (HBM latency hides under the MFMAs) and write
+  // them to LDS only after the end-of-tile barrier (guide §5.5 T14/G15).
+  // Per-thread staging registers: K = 2 int4 rows-slices, V = 1 row-pair.
+  constexpr int KV_PER_THR = BN * D / 8 / 512;  // int4 K vectors / thread
+  int4 kreg[KV_PER_THR];
+  int4 vreg0, vreg1;
+
+  auto issue_loads = [&](int kbase) {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          __bf16 pair[2] = {e0[j], e1[j]};
-          *(uint*)&vt_lds[(col8 + j) * VP + row] = *(uint*)pair;
-        }
+    for (int u = 0; u < KV_PER_THR; ++u) {
+      const int t = threadIdx.x + u * 512;
+      const int row = t / (D / 8);
+      const int col8 = (t % (D / 8)) * 8;
+      const int gr = kbase + row;
+      kreg[u] = (gr < S) ? *(const int4*)(Kp + (long)gr * sKs + col8)
+                         : int4{0, 0, 0, 0};
+    }
+    {
+      const int t = threadIdx.x;  // BN*D/16 == 512 row-pair slices
+      const int row = (t / (D / 8)) * 2;
+      const int col8 = (t % (D / 8)) * 8;
+      const int g0 = kbase + row, g1 = g0 + 1;
+      vreg0 = (g0 < S) ? *(const int4*)(Vp + (long)g0 * sVs + col8)
+                       : int4{0, 0, 0, 0};
+      vreg1 = (g1 < S) ? *(const int4*)(Vp + (long)g1 * sVs + col8)
+                       : int4{0, 0, 0, 0};
+    }
+  };
+
+  auto write_lds = [&]() {
+#pragma unroll
+    for (int u = 0; u < KV_PER_THR; ++u) {
+      const int t = threadIdx.x + u * 512;
+      const int row = t / (D / 8);
+      const int col8 = (t % (D / 8)) * 8;
+      *(int4*)&k_lds[row * KP + col8] = kreg[u];
+    }
+    {
+      const int t = threadIdx.x;
+      const int row = (t / (D / 8)) * 2;
+      const int col8 = (t % (D / 8)) * 8;
+      const __bf16* e0 = (const __bf16*)&vreg0;
+      const __bf16* e1 = (const __bf16*)&vreg1;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        __bf16 pair[2] = {e0[j], e1[j]};
+        *(uint*)&vt_lds[(col8 + j) * VP + row] = *(uint*)pair;
       }
     }
-    __syncthreads();
+  };
+
+  issue_loads(0);
+  write_lds();
+  __syncthreads();
+
+  for (int jb = 0; jb < nkb; ++jb) {
+    const int kbase = jb * BN;
+    if (jb + 1 < nkb) issue_loads((jb + 1) * BN);
 
     if (!CAUSAL || kbase <= wrow_max) {
       // ---- S = Q K^T for both sub-blocks (B-frags loaded once) ----
@@ -214,7 +241,11 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
         }
       }
     }
-    __syncthreads();
+    __syncthreads();  // all waves done reading this tile's LDS
+    if (jb + 1 < nkb) {
+      write_lds();
+      __syncthreads();  // next tile visible
+    }
   }
 
   // ---- epilogue: O /= l (strided [s,b,h,d] store) + LSE ----
