@@ -212,7 +212,10 @@ class LlamaAttention(nn.Module):
             from ..ops.ring_attn import ring_flash_attn
 
             return ring_flash_attn(q, k, v, scale=self.scale)
-        return flash_attn_func(q, k, v, causal=True, scale=self.scale)
+        return flash_attn_func(
+            q, k, v, causal=True, scale=self.scale,
+            window=getattr(self.cfg, "sliding_window", None),
+        )
 
     def forward(self, x, cos, sin, pos_offset: int = 0):
         # x: [s(, /tp if SP), b, h]

@@ -18,12 +18,12 @@ void launch_rope_fwd(const void*, const void*, const void*, void*, int, int,
 void launch_adamw(void*, const void*, void*, void*, const void*, long, float,
                   float, float, float, float, int, hipStream_t);
 void launch_flash_fwd(const void*, const void*, const void*, void*, void*,
-                      int, int, int, int, int, bool, float, const long*,
+                      int, int, int, int, int, bool, float, int, const long*,
                       const long*, const long*, hipStream_t);
 void launch_flash_bwd(const void*, const void*, const void*, const void*,
                       const void*, const void*, void*, void*, void*, int, int,
-                      int, int, int, bool, float, const long*, const long*,
-                      const long*, const long*, hipStream_t);
+                      int, int, int, bool, float, int, const long*,
+                      const long*, const long*, const long*, hipStream_t);
 void launch_mfma_probe(const void*, const void*, void*, hipStream_t);
 void launch_ce_fwd(const void*, const void*, void*, void*, void*, long, int,
                    long, hipStream_t);
@@ -129,7 +129,7 @@ static void check_bhsd(const torch::Tensor& t, const char* name) {
 
 std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
                                           torch::Tensor v, bool causal,
-                                          double scale) {
+                                          double scale, long window) {
   // q/k/v: logical [b, h, s, d], arbitrary strides (views of the QKV GEMM
   // output). Returns O as a [b, h, s, d] view of [s, b, h, d] storage.
   check_bhsd(q, "q");
@@ -146,15 +146,16 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
   long ks[3] = {k.stride(2), k.stride(0), k.stride(1)};
   long vs[3] = {v.stride(2), v.stride(0), v.stride(1)};
   launch_flash_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o_mem.data_ptr(),
-                   lse.data_ptr(), B, HQ, HKV, S, D, causal, (float)scale, qs,
-                   ks, vs, cur_stream());
+                   lse.data_ptr(), B, HQ, HKV, S, D, causal, (float)scale,
+                   (int)window, qs, ks, vs, cur_stream());
   return {o_mem.permute({1, 2, 0, 3}), lse};
 }
 
 std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                                           torch::Tensor k, torch::Tensor v,
                                           torch::Tensor o, torch::Tensor lse,
-                                          bool causal, double scale) {
+                                          bool causal, double scale,
+                                          long window) {
   if (dout.stride(3) != 1) dout = dout.contiguous();
   check_bhsd(dout, "dout");
   check_bhsd(q, "q");
@@ -174,7 +175,8 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
   launch_flash_bwd(dout.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                    lse.data_ptr(), delta.data_ptr(), dq_mem.data_ptr(),
                    dk_mem.data_ptr(), dv_mem.data_ptr(), B, HQ, HKV, S, D,
-                   causal, (float)scale, qs, ks, vs, ds, cur_stream());
+                   causal, (float)scale, (int)window, qs, ks, vs, ds,
+                   cur_stream());
   return {dq_mem.permute({1, 2, 0, 3}), dk_mem.permute({1, 2, 0, 3}),
           dv_mem.permute({1, 2, 0, 3})};
 }
@@ -227,8 +229,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("rope_fwd", &rope_fwd);
   m.def("adamw_step", &adamw_step);
-  m.def("flash_attn_fwd", &flash_attn_fwd);
-  m.def("flash_attn_bwd", &flash_attn_bwd);
+  m.def("flash_attn_fwd", &flash_attn_fwd, pybind11::arg("q"),
+        pybind11::arg("k"), pybind11::arg("v"), pybind11::arg("causal"),
+        pybind11::arg("scale"), pybind11::arg("window") = 0);
+  m.def("flash_attn_bwd", &flash_attn_bwd, pybind11::arg("dout"),
+        pybind11::arg("q"), pybind11::arg("k"), pybind11::arg("v"),
+        pybind11::arg("o"), pybind11::arg("lse"), pybind11::arg("causal"),
+        pybind11::arg("scale"), pybind11::arg("window") = 0);
   m.def("mfma_probe", &mfma_probe);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);

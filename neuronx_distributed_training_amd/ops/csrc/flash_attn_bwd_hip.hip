@@ -20,8 +20,8 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
     const float* __restrict__ LSE, const float* __restrict__ DELTA,
     bf16* __restrict__ dK,  // [s, b, hkv, d] contiguous
     bf16* __restrict__ dV, int S, int Bb, int HQ, int HKV, float scale,
-    long sQs, long sQb, long sQh, long sKs, long sKb, long sKh, long sVs,
-    long sVb, long sVh, long sDs, long sDb, long sDh) {
+    int window, long sQs, long sQb, long sQh, long sKs, long sKb, long sKh,
+    long sVs, long sVb, long sVh, long sDs, long sDb, long sDh) {
   constexpr int BN = 256;  // keys per block
   constexpr int BM = 64;   // q tile
   constexpr int KP = D + 8;
@@ -81,7 +81,9 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
     const float* Dp = DELTA + ((long)b * HQ + hq) * S;
 
     const int ib0 = CAUSAL ? (kbase / BM) : 0;
-    const int nqb = (S + BM - 1) / BM;
+    int nqb = (S + BM - 1) / BM;
+    if (CAUSAL && window > 0)
+      nqb = min(nqb, (kbase + BN - 1 + window + BM - 1) / BM);
     for (int ib = ib0; ib < nqb; ++ib) {
       const int qbase = ib * BM;
       {  // stage Q, dO row-major (16B) + transposed (paired b32)
@@ -151,8 +153,9 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
               const int krow = krow_w + sb * 16 + (lane >> 4) * 4 + r;
-              const bool dead =
+              bool dead =
                   (qcol >= S) || (krow >= S) || (CAUSAL && qcol < krow);
+              if (CAUSAL && window > 0) dead |= (qcol >= krow + window);
               const float p = dead ? 0.f : __expf(st[sb][r] * scale - lse);
               const float ds = p * (dpt[sb][r] - delta) * scale;
               const int lrow = sb * 16 + (lane >> 4) * 4 + r;
@@ -214,9 +217,9 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
     const bf16* __restrict__ K, const bf16* __restrict__ V,
     const float* __restrict__ LSE, const float* __restrict__ DELTA,
     bf16* __restrict__ dQ,  // [s, b, hq, d] contiguous
-    int S, int Bb, int HQ, int HKV, float scale, long sQs, long sQb, long sQh,
-    long sKs, long sKb, long sKh, long sVs, long sVb, long sVh, long sDs,
-    long sDb, long sDh) {
+    int S, int Bb, int HQ, int HKV, float scale, int window, long sQs,
+    long sQb, long sQh, long sKs, long sKb, long sKh, long sVs, long sVb,
+    long sVh, long sDs, long sDb, long sDh) {
   constexpr int BM = 256, BN = 64;
   constexpr int KP = D + 8;
   constexpr int VP = BN + 8;
@@ -274,9 +277,11 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
 
   const int kend = CAUSAL ? min(S, q0 + BM) : S;
   const int nkb = (kend + BN - 1) / BN;
+  const int jb0 =
+      (CAUSAL && window > 0) ? max(0, (q0 - window + 1) / BN) : 0;
   const int wrow_max = qrow_w + 31;
 
-  for (int jb = 0; jb < nkb; ++jb) {
+  for (int jb = jb0; jb < nkb; ++jb) {
     const int kbase = jb * BN;
     {  // stage K, V row-major + K^T paired
       constexpr int KVECS = BN * D / 8;
@@ -334,7 +339,8 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
             const int qrow = qrow_w + sb * 16 + (lane >> 4) * 4 + r;
-            const bool dead = (kcol >= S) || (CAUSAL && kcol > qrow);
+            bool dead = (kcol >= S) || (CAUSAL && kcol > qrow);
+            if (CAUSAL && window > 0) dead |= (kcol <= qrow - window);
             const float p =
                 dead ? 0.f : __expf(st[sb][r] * scale - lse[sb][r]);
             const float ds = p * (dpt[sb][r] - delta[sb][r]) * scale;
@@ -378,9 +384,9 @@ extern "C" {
 void launch_flash_bwd(const void* dout, const void* q, const void* k,
                       const void* v, const void* lse, const void* delta,
                       void* dq, void* dk, void* dv, int B, int HQ, int HKV,
-                      int S, int D, bool causal, float scale, const long* qstr,
-                      const long* kstr, const long* vstr, const long* dostr,
-                      hipStream_t stream) {
+                      int S, int D, bool causal, float scale, int window,
+                      const long* qstr, const long* kstr, const long* vstr,
+                      const long* dostr, hipStream_t stream) {
   dim3 blk(512);
   dim3 gkv((S + 255) / 256, B * HKV);
   dim3 gq((S + 255) / 256, B * HQ);
@@ -389,13 +395,13 @@ void launch_flash_bwd(const void* dout, const void* q, const void* k,
    hipLaunchKernelGGL(( flash_bwd_dkv_kernel<DD, CC>), dim3(gkv), dim3(blk), 0, stream,                     \
         (const bf16*)dout, (const bf16*)q, (const bf16*)k, (const bf16*)v,    \
         (const float*)lse, (const float*)delta, (bf16*)dk, (bf16*)dv, S, B,   \
-        HQ, HKV, scale, qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2], \
-        vstr[0], vstr[1], vstr[2], dostr[0], dostr[1], dostr[2]);             \
+        HQ, HKV, scale, window, qstr[0], qstr[1], qstr[2], kstr[0], kstr[1],  \
+        kstr[2], vstr[0], vstr[1], vstr[2], dostr[0], dostr[1], dostr[2]);    \
    hipLaunchKernelGGL(( flash_bwd_dq_kernel<DD, CC>), dim3(gq), dim3(blk), 0, stream,                       \
         (const bf16*)dout, (const bf16*)q, (const bf16*)k, (const bf16*)v,    \
         (const float*)lse, (const float*)delta, (bf16*)dq, S, B, HQ, HKV,     \
-        scale, qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2], vstr[0], \
-        vstr[1], vstr[2], dostr[0], dostr[1], dostr[2]);                      \
+        scale, window, qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2],  \
+        vstr[0], vstr[1], vstr[2], dostr[0], dostr[1], dostr[2]);             \
   } while (0)
   if (D == 128) {
     if (causal) CASE(128, true); else CASE(128, false);

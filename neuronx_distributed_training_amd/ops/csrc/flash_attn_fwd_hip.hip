@@ -20,6 +20,7 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
     const bf16* __restrict__ V, bf16* __restrict__ O,
     float* __restrict__ LSE,  // [B, HQ, S] f32
     int S, int Bb, int HQ, int HKV, float scale,
+    int window,                     // sliding window (<=0: disabled)
     long sQs, long sQb, long sQh,   // Q element strides (seq, batch, head)
     long sKs, long sKb, long sKh,   // K strides
     long sVs, long sVb, long sVh) { // V strides
@@ -74,6 +75,8 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
 
   const int kend = CAUSAL ? min(S, q0 + BM) : S;
   const int nkb = (kend + BN - 1) / BN;
+  const int jb0 =
+      (CAUSAL && window > 0) ? max(0, (q0 - window + 1) / BN) : 0;
   const int wrow_max = qrow_w + 31;
 
   // T5 static form: the younger dispatch half gets priority so it is not
@@ -133,11 +136,11 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
     }
   };
 
-  issue_loads(0);
+  issue_loads(jb0 * BN);
   write_lds();
   __syncthreads();
 
-  for (int jb = 0; jb < nkb; ++jb) {
+  for (int jb = jb0; jb < nkb; ++jb) {
     const int kbase = jb * BN;
     if (jb + 1 < nkb) issue_loads((jb + 1) * BN);
 
@@ -163,7 +166,8 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
         // interior tiles (every key visible to every row) skip the mask
         const bool full_tile =
             (kbase + BN <= S) &&
-            (!CAUSAL || (kbase + BN - 1 <= qrow_w + sb * 16));
+            (!CAUSAL || (kbase + BN - 1 <= qrow_w + sb * 16)) &&
+            (window <= 0 || kbase >= qrow_w + sb * 16 + 15 - window + 1);
         float tile_max[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
         float sv[4][4];
         if (full_tile) {
@@ -184,7 +188,8 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
             for (int r = 0; r < 4; ++r) {
               const int qrow = qrow_w + sb * 16 + (lane >> 4) * 4 + r;
               float s = sacc[sb][nk][r] * scale;
-              const bool dead = (kcol >= S) || (CAUSAL && kcol > qrow);
+              bool dead = (kcol >= S) || (CAUSAL && kcol > qrow);
+              if (CAUSAL && window > 0) dead |= (kcol <= qrow - window);
               s = dead ? -1e30f : s;
               sv[nk][r] = s;
               tile_max[r] = fmaxf(tile_max[r], s);
@@ -282,15 +287,15 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
 extern "C" {
 void launch_flash_fwd(const void* q, const void* k, const void* v, void* o,
                       void* lse, int B, int HQ, int HKV, int S, int D,
-                      bool causal, float scale, const long* qstr,
+                      bool causal, float scale, int window, const long* qstr,
                       const long* kstr, const long* vstr, hipStream_t stream) {
   dim3 grid((S + 255) / 256, B * HQ);
   dim3 blk(512);
 #define CASE(DD, CC)                                                          \
  hipLaunchKernelGGL(( flash_fwd_kernel<DD, CC>), dim3(grid), dim3(blk), 0, stream,                          \
       (const bf16*)q, (const bf16*)k, (const bf16*)v, (bf16*)o, (float*)lse, \
-      S, B, HQ, HKV, scale, qstr[0], qstr[1], qstr[2], kstr[0], kstr[1],      \
-      kstr[2], vstr[0], vstr[1], vstr[2])
+      S, B, HQ, HKV, scale, window, qstr[0], qstr[1], qstr[2], kstr[0],       \
+      kstr[1], kstr[2], vstr[0], vstr[1], vstr[2])
   if (D == 128) {
     if (causal) CASE(128, true); else CASE(128, false);
   } else if (D == 64) {

@@ -23,16 +23,26 @@ import torch.nn.functional as F
 from . import kernels_for
 
 
-def _cpu_ref_fwd(q, k, v, causal, scale):
+def _make_mask(sq, sk, causal, window, device):
+    mask = torch.zeros(sq, sk, dtype=torch.bool, device=device)
+    if causal:
+        mask |= torch.ones(sq, sk, dtype=torch.bool, device=device).triu(1)
+        if window and window > 0:
+            mask |= torch.ones(sq, sk, dtype=torch.bool, device=device).tril(
+                -window
+            )
+    return mask
+
+
+def _cpu_ref_fwd(q, k, v, causal, scale, window=0):
     hq, hkv = q.size(1), k.size(1)
     if hq != hkv:
         k = k.repeat_interleave(hq // hkv, dim=1)
         v = v.repeat_interleave(hq // hkv, dim=1)
     qf, kf, vf = q.float(), k.float(), v.float()
     s = torch.matmul(qf, kf.transpose(-1, -2)) * scale
-    if causal:
-        mask = torch.ones(q.size(-2), k.size(-2), dtype=torch.bool, device=q.device).triu(1)
-        s = s.masked_fill(mask, float("-inf"))
+    mask = _make_mask(q.size(-2), k.size(-2), causal, window, q.device)
+    s = s.masked_fill(mask, float("-inf"))
     lse = torch.logsumexp(s, dim=-1)
     p = torch.softmax(s, dim=-1)
     o = torch.matmul(p, vf)
@@ -41,16 +51,17 @@ def _cpu_ref_fwd(q, k, v, causal, scale):
 
 class _FlashAttnFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, causal: bool, scale: float):
+    def forward(ctx, q, k, v, causal: bool, scale: float, window: int = 0):
         kern = kernels_for(q)
         if kern is not None:
             # kernel takes arbitrary-strided [b, h, s, d] views (d contig)
-            o, lse = kern.flash_attn_fwd(q, k, v, causal, scale)
+            o, lse = kern.flash_attn_fwd(q, k, v, causal, scale, window)
         else:
-            o, lse = _cpu_ref_fwd(q, k, v, causal, scale)
+            o, lse = _cpu_ref_fwd(q, k, v, causal, scale, window)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.causal = causal
         ctx.scale = scale
+        ctx.window = window
         return o
 
     @staticmethod
@@ -59,9 +70,9 @@ class _FlashAttnFn(torch.autograd.Function):
         kern = kernels_for(q)
         if kern is not None:
             dq, dk, dv = kern.flash_attn_bwd(
-                do, q, k, v, o, lse, ctx.causal, ctx.scale
+                do, q, k, v, o, lse, ctx.causal, ctx.scale, ctx.window
             )
-            return dq, dk, dv, None, None
+            return dq, dk, dv, None, None, None
         # CPU reference backward (fp32, explicit)
         hq, hkv = q.size(1), k.size(1)
         g = hq // hkv
@@ -69,9 +80,8 @@ class _FlashAttnFn(torch.autograd.Function):
         vx = v.repeat_interleave(g, dim=1).float()
         qf, dof = q.float(), do.float()
         s = torch.matmul(qf, kx.transpose(-1, -2)) * ctx.scale
-        if ctx.causal:
-            mask = torch.ones(q.size(-2), k.size(-2), dtype=torch.bool, device=q.device).triu(1)
-            s = s.masked_fill(mask, float("-inf"))
+        mask = _make_mask(q.size(-2), k.size(-2), ctx.causal, ctx.window, q.device)
+        s = s.masked_fill(mask, float("-inf"))
         p = torch.softmax(s, dim=-1)
         dv = torch.matmul(p.transpose(-1, -2), dof)
         dp = torch.matmul(dof, vx.transpose(-1, -2))
@@ -82,7 +92,7 @@ class _FlashAttnFn(torch.autograd.Function):
         if g > 1:
             dk = dk.reshape(dk.size(0), hkv, g, dk.size(-2), dk.size(-1)).sum(2)
             dv = dv.reshape(dv.size(0), hkv, g, dv.size(-2), dv.size(-1)).sum(2)
-        return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), None, None
+        return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), None, None, None
 
 
 def flash_attn_func(
@@ -91,7 +101,10 @@ def flash_attn_func(
     v: torch.Tensor,
     causal: bool = True,
     scale: float | None = None,
+    window: int | None = None,
 ) -> torch.Tensor:
+    """window: sliding-window size (attend to the last `window` keys);
+    None/0 disables (Mixtral sliding_window parity)."""
     if scale is None:
         scale = 1.0 / math.sqrt(q.size(-1))
-    return _FlashAttnFn.apply(q, k, v, causal, scale)
+    return _FlashAttnFn.apply(q, k, v, causal, scale, int(window or 0))

@@ -187,3 +187,33 @@ def test_fused_cross_entropy_gpu(ext):
     # tolerance = bf16 rounding at |g|≈1, not kernel error
     err = (logits.grad.float() - ref_in.grad).abs().max()
     assert err < 1.5e-2, err
+
+
+def test_flash_attn_sliding_window_gpu(ext):
+    from neuronx_distributed_training_amd.ops import flash_attn_func
+
+    torch.manual_seed(8)
+    b, hq, hkv, s, d, w = 1, 4, 2, 512, 128, 128
+    q = torch.randn(b, hq, s, d, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    k = torch.randn(b, hkv, s, d, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    v = torch.randn(b, hkv, s, d, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    o = flash_attn_func(q, k, v, causal=True, window=w)
+    kx = k.repeat_interleave(hq // hkv, 1).float()
+    vx = v.repeat_interleave(hq // hkv, 1).float()
+    scores = q.float() @ kx.transpose(-1, -2) / (d ** 0.5)
+    mask = torch.ones(s, s, dtype=torch.bool, device="cuda").triu(1) | torch.ones(
+        s, s, dtype=torch.bool, device="cuda"
+    ).tril(-w)
+    scores = scores.masked_fill(mask, float("-inf"))
+    ref = torch.softmax(scores, -1) @ vx
+    err = (o.float() - ref).abs().max()
+    assert err < 0.02, err
+    g = torch.randn_like(o)
+    o.backward(g)
+    # reference grads
+    qr = q.detach().float().clone().requires_grad_(True)
+    kr = kx.detach().clone().requires_grad_(True)
+    vr = vx.detach().clone().requires_grad_(True)
+    s2 = (qr @ kr.transpose(-1, -2) / (d ** 0.5)).masked_fill(mask, float("-inf"))
+    (torch.softmax(s2, -1) @ vr).backward(g.float())
+    assert (q.grad.float() - qr.grad).abs().max() / qr.grad.abs().max().clamp(min=1) < 0.05

@@ -81,3 +81,24 @@ def test_flash_attn_cpu_matches_sdpa():
     assert torch.allclose(q.grad, qr.grad, atol=1e-4)
     assert torch.allclose(k.grad, kr.grad, atol=1e-4)
     assert torch.allclose(v.grad, vr.grad, atol=1e-4)
+
+
+def test_flash_attn_sliding_window_cpu():
+    torch.manual_seed(5)
+    b, h, s, d = 1, 2, 40, 16
+    q = torch.randn(b, h, s, d, requires_grad=True)
+    k = torch.randn(b, h, s, d, requires_grad=True)
+    v = torch.randn(b, h, s, d, requires_grad=True)
+    w = 8
+    o = flash_attn_func(q, k, v, causal=True, window=w)
+    # explicit banded reference
+    qf, kf, vf = q.detach().float(), k.detach().float(), v.detach().float()
+    scores = qf @ kf.transpose(-1, -2) / (d ** 0.5)
+    mask = torch.ones(s, s, dtype=torch.bool).triu(1) | torch.ones(
+        s, s, dtype=torch.bool
+    ).tril(-w)
+    scores = scores.masked_fill(mask, float("-inf"))
+    ref = torch.softmax(scores, -1) @ vf
+    assert torch.allclose(o, ref, atol=1e-4), (o - ref).abs().max()
+    o.sum().backward()
+    assert torch.isfinite(q.grad).all()
