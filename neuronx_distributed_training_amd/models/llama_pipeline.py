@@ -18,6 +18,88 @@ from ..ops.rope import build_rope_cache
 from .llama import LlamaConfig, LlamaDecoderLayer, _init_method
 
 
+class LlamaChunk(nn.Module):
+    """One VIRTUAL pipeline stage (model chunk) for the interleaved
+    schedule: virtual stage v = chunk_idx·pp + pp_rank holds layers
+    [start, end); v==0 additionally owns the embedding, the last virtual
+    stage owns final norm + lm_head + loss."""
+
+    def __init__(self, cfg: LlamaConfig, v: int, n_virtual: int, start: int, end: int):
+        super().__init__()
+        self.cfg = cfg
+        self.is_first = v == 0
+        self.is_last = v == n_virtual - 1
+        self.dtype = cfg.torch_dtype
+        dt = cfg.torch_dtype
+        if self.is_first:
+            self.embed_tokens = ParallelEmbedding(
+                cfg.vocab_size, cfg.hidden_size,
+                init_method=_init_method(cfg.initializer_range), dtype=dt,
+                init_seed=77,
+            )
+        self.layers = nn.ModuleList(
+            [LlamaDecoderLayer(cfg, i) for i in range(start, end)]
+        )
+        if self.is_last:
+            self.norm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, dtype=dt)
+            self.lm_head = ColumnParallelLinear(
+                cfg.hidden_size, cfg.vocab_size, bias=False,
+                init_method=_init_method(cfg.initializer_range), dtype=dt,
+                init_seed=88,
+            )
+        cos, sin = build_rope_cache(
+            cfg.max_position_embeddings, cfg.head_dim, cfg.rope_theta,
+            rope_scaling=cfg.rope_scaling,
+        )
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+        self._batch = {}
+
+    def set_batch(self, batch):
+        dev = next(self.parameters()).device
+        self._batch = {
+            k: (v.to(dev) if torch.is_tensor(v) else v) for k, v in batch.items()
+        }
+
+    def hidden_shape_for(self, batch):
+        b, s = batch["input_ids"].shape
+        return (s, b, self.cfg.hidden_size)
+
+    def forward(self, x):
+        if self.is_first:
+            x = self.embed_tokens(self._batch["input_ids"]).transpose(0, 1).contiguous()
+        for layer in self.layers:
+            x = layer(x, self.rope_cos, self.rope_sin, 0)
+        if not self.is_last:
+            return x
+        x = self.norm(x)
+        logits = self.lm_head(x).transpose(0, 1)
+        labels = self._batch.get("labels", self._batch["input_ids"])
+        loss_mask = self._batch.get("loss_mask")
+        logits = logits[:, :-1]
+        labels = labels[:, 1:]
+        loss_mask = loss_mask[:, 1:] if loss_mask is not None else None
+        per_tok = parallel_cross_entropy(logits, labels)
+        if loss_mask is not None:
+            m = loss_mask.to(per_tok.dtype)
+            return (per_tok * m).sum() / m.sum().clamp(min=1)
+        return per_tok.mean()
+
+
+def build_virtual_chunks(cfg: LlamaConfig, vp: int):
+    """ModuleList of this rank's vp model chunks (interleaved VP)."""
+    pp = ps.get_pipeline_model_parallel_world_size()
+    rank = ps.get_pipeline_model_parallel_rank()
+    n_virtual = pp * vp
+    ranges = partition_layers(cfg.num_hidden_layers, n_virtual)
+    chunks = nn.ModuleList()
+    for c in range(vp):
+        v = c * pp + rank
+        start, end = ranges[v]
+        chunks.append(LlamaChunk(cfg, v, n_virtual, start, end))
+    return chunks
+
+
 def partition_layers(num_layers: int, pp: int, pipeline_cuts: Optional[List[int]] = None):
     """[start, end) layer range per stage. pipeline_cuts: explicit cut
     points (layer index starting each stage>0), else uniform."""

@@ -54,9 +54,19 @@ class BaseModelModule:
         self.model.train()
         self.pp_engine = None
         if ps.get_pipeline_model_parallel_world_size() > 1:
-            from .pipeline import PipelineEngine
+            vp = int(
+                self.cfg.get("distributed_strategy", {}).get(
+                    "virtual_pipeline_model_parallel_size", 1
+                ) or 1
+            )
+            if vp > 1:
+                from .pipeline import InterleavedPipelineEngine
 
-            self.pp_engine = PipelineEngine(self.model)
+                self.pp_engine = InterleavedPipelineEngine(self.model)
+            else:
+                from .pipeline import PipelineEngine
+
+                self.pp_engine = PipelineEngine(self.model)
 
     def configure_optimizers(self, max_steps: int):
         ocfg = self.cfg["model"].get("optim", {})
@@ -247,9 +257,15 @@ class LlamaModule(BaseModelModule):
             dtype=dtype,
         )
         if ps.get_pipeline_model_parallel_world_size() > 1:
-            from ..models.llama_pipeline import LlamaStage
+            vp = int(dstr.get("virtual_pipeline_model_parallel_size", 1) or 1)
+            if vp > 1:
+                from ..models.llama_pipeline import build_virtual_chunks
 
-            model = LlamaStage(cfg, pipeline_cuts=dstr.get("pipeline_cuts"))
+                model = build_virtual_chunks(cfg, vp)
+            else:
+                from ..models.llama_pipeline import LlamaStage
+
+                model = LlamaStage(cfg, pipeline_cuts=dstr.get("pipeline_cuts"))
         else:
             model = LlamaForCausalLM(cfg)
         peft = mcfg.get("peft")

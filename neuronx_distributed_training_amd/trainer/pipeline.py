@@ -164,3 +164,121 @@ class PipelineEngine:
         if grp is None or w.grad is None:
             return
         dist.all_reduce(w.grad, group=grp)
+
+
+class InterleavedPipelineEngine:
+    """Interleaved 1F1B over vp model chunks per rank (virtual pipeline —
+    reference `virtual_pipeline_model_parallel_size`, model/base.py:155).
+
+    Virtual stage v = chunk·pp + rank; the fwd/bwd virtual-step order is
+    rank-independent, so per-channel message order matches on both ends
+    (plain send/recv pairing is safe). Warmup count
+    (pp − rank − 1)·2 + (vp − 1)·pp is the standard interleaved schedule.
+    """
+
+    def __init__(self, chunks):
+        self.chunks = chunks
+        self.rank = ps.get_pipeline_model_parallel_rank()
+        self.world = ps.get_pipeline_model_parallel_world_size()
+        self.vp = len(chunks)
+        self.device = next(chunks[0].parameters()).device
+        self._sends = _SendPool()
+        s = ps._st()
+        self.pp_ranks = s.pp_ranks or [s.rank]
+
+    def run_train(self, microbatches):
+        M = len(microbatches)
+        pp, vp, rank = self.world, self.vp, self.rank
+        assert M % pp == 0, f"num_microbatches {M} must divide by pp {pp}"
+        total = M * vp
+        n_virtual = pp * vp
+        last_v = n_virtual - 1
+        if M == pp:
+            num_warmup = total
+        else:
+            num_warmup = min(total, (pp - rank - 1) * 2 + (vp - 1) * pp)
+        num_steady = total - num_warmup
+
+        store = [dict() for _ in range(vp)]
+        losses = []
+
+        def chunk_of(i):
+            return (i // pp) % vp
+
+        def micro_of(i):
+            return (i // (pp * vp)) * pp + (i % pp)
+
+        def fwd_step(i):
+            c = chunk_of(i)
+            m = micro_of(i)
+            chunk = self.chunks[c]
+            v = c * pp + rank
+            chunk.set_batch(microbatches[m])
+            if v == 0:
+                inp = None
+            else:
+                src = self.pp_ranks[(rank - 1) % pp]
+                inp = _recv(chunk.hidden_shape_for(microbatches[m]),
+                            chunk.dtype, self.device, src)
+                inp.requires_grad_(True)
+            out = chunk(inp)
+            if v == last_v:
+                losses.append(out.detach())
+                out = out / M
+            else:
+                self._sends.send(out.detach(), self.pp_ranks[(rank + 1) % pp])
+            store[c][m] = (inp, out)
+
+        def bwd_step(i):
+            c = vp - 1 - chunk_of(i)
+            m = micro_of(i)
+            v = c * pp + rank
+            inp, out = store[c].pop(m)
+            if v == last_v:
+                torch.autograd.backward(out)
+            else:
+                dout = _recv(tuple(out.shape), out.dtype, self.device,
+                             self.pp_ranks[(rank + 1) % pp])
+                torch.autograd.backward(out, grad_tensors=dout)
+            if v > 0:
+                self._sends.send(inp.grad, self.pp_ranks[(rank - 1) % pp])
+
+        for i in range(num_warmup):
+            fwd_step(i)
+        for k in range(num_steady):
+            fwd_step(num_warmup + k)
+            bwd_step(k)
+        for k in range(num_steady, total):
+            bwd_step(k)
+        self._sends.drain()
+
+        if losses:
+            return torch.stack(losses).mean()
+        return torch.zeros((), device=self.device)
+
+    @torch.no_grad()
+    def run_eval(self, microbatches):
+        M = len(microbatches)
+        pp, vp, rank = self.world, self.vp, self.rank
+        losses = []
+        last_v = pp * vp - 1
+        for c in range(vp):
+            v = c * pp + rank
+            chunk = self.chunks[c]
+            for m, batch in enumerate(microbatches):
+                chunk.set_batch(batch)
+                if v == 0:
+                    inp = None
+                else:
+                    src = self.pp_ranks[(rank - 1) % pp]
+                    inp = _recv(chunk.hidden_shape_for(batch), chunk.dtype,
+                                self.device, src)
+                out = chunk(inp)
+                if v == last_v:
+                    losses.append(out)
+                else:
+                    self._sends.send(out, self.pp_ranks[(rank + 1) % pp])
+        self._sends.drain()
+        if losses:
+            return torch.stack(losses).mean()
+        return torch.zeros((), device=self.device)

@@ -105,3 +105,42 @@ def _pp_train(rank, world):
 def test_pp2_training_decreases():
     res = run_distributed(_pp_train, 2)
     assert res[0] == res[1]
+
+
+def _vp_loss(rank, world, vp):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.llama import LlamaConfig
+    from neuronx_distributed_training_amd.models.llama_pipeline import (
+        build_virtual_chunks,
+    )
+    from neuronx_distributed_training_amd.trainer.pipeline import (
+        InterleavedPipelineEngine,
+    )
+    import torch.distributed as dist
+
+    ps.initialize_model_parallel(pipeline_model_parallel_size=world)
+    torch.manual_seed(7)
+    cfg = LlamaConfig(**TINY)
+    chunks = build_virtual_chunks(cfg, vp)
+    eng = InterleavedPipelineEngine(chunks)
+    g = torch.Generator().manual_seed(99)
+    micro = []
+    for _ in range(4):
+        ids = torch.randint(0, 128, (1, 32), generator=g)
+        micro.append({"input_ids": ids, "labels": ids.clone()})
+    loss = eng.run_train(micro).float()
+    dist.all_reduce(loss, group=ps.get_pipeline_model_parallel_group())
+    gsum = sum(
+        float(p.grad.abs().sum())
+        for p in chunks.parameters()
+        if p.grad is not None
+    )
+    assert gsum > 0
+    return float(loss)
+
+
+def test_vp2_loss_matches_reference():
+    ref = run_distributed(_ref_loss, 1, False)[0]
+    vp = run_distributed(_vp_loss, 2, 2)
+    assert abs(vp[0] - vp[1]) < 1e-6
+    assert abs(vp[0] - ref) < 5e-3, (vp[0], ref)
