@@ -43,6 +43,7 @@ def main():
     ap.add_argument("--steps", type=int, default=8)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--gbs", type=int, default=None, help="global batch (sequences)")
+    ap.add_argument("--mbs", type=int, default=1, help="micro batch size")
     ap.add_argument("--seq", type=int, default=8192)
     ap.add_argument("--layers", type=int, default=32)
     ap.add_argument(
@@ -72,7 +73,7 @@ def main():
     torch.manual_seed(1234)
 
     gbs = args.gbs if args.gbs is not None else 4
-    mbs = 1
+    mbs = args.mbs
     seq = args.seq
     cfg = {
         "data": {"global_batch_size": gbs, "micro_batch_size": mbs, "seq_length": seq},
