@@ -33,6 +33,8 @@ if os.path.exists(_TUNE) and os.environ.get("NXDT_DISABLE_TUNABLEOP") != "1":
         "PYTORCH_TUNABLEOP_FILENAME", os.path.join(_td, "tunableop_gfx950.csv")
     )
 
+os.environ.setdefault("PYTORCH_ALLOC_CONF", "expandable_segments:True")
+
 import torch
 import torch.distributed as dist
 

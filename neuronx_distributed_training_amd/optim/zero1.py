@@ -123,9 +123,9 @@ class ZeRO1AdamW:
             if not is_tp and tp_rank != 0:
                 tp_once_mask[o : o + nel] = False
         self.wd_shard = wd_mask[self.shard_start : self.shard_start + self.shard_size].to(dev)
-        self.normmask_shard = (
-            tp_once_mask[self.shard_start : self.shard_start + self.shard_size].to(dev)
-        )
+        nm = tp_once_mask[self.shard_start : self.shard_start + self.shard_size]
+        self.normmask_all = bool(nm.all())
+        self.normmask_shard = nm.to(dev).to(torch.float32)
 
         # map param -> (offset, numel) for grad fill
         self._grad_views = [
@@ -209,7 +209,16 @@ class ZeRO1AdamW:
 
         # 2) global grad norm: count TP-sharded params on all ranks,
         #    replicated params only on tp rank 0; reduce over DP then TP/PP.
-        sq = (shard * self.normmask_shard).pow(2).sum()
+        #    torch.dot keeps this allocation-free (a masked pow() materializes
+        #    a full fp32 copy of the 32 GB shard and can OOM at large MBS).
+        if bool(self.normmask_all):
+            sq = torch.dot(shard, shard)
+        else:
+            sq = torch.zeros((), dtype=torch.float32, device=shard.device)
+            cs = 1 << 26  # 64M elements (256 MB) per chunk
+            for s0 in range(0, shard.numel(), cs):
+                piece = shard[s0 : s0 + cs] * self.normmask_shard[s0 : s0 + cs]
+                sq += torch.dot(piece, piece)
         if expert_grads and ps.get_tensor_model_parallel_rank() == 0:
             # each expert set appears expert_dp_world times across DP
             sq = sq + sum(g.pow(2).sum() for g in expert_grads) / self.expert_dp_world
