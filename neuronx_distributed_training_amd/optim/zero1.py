@@ -211,14 +211,13 @@ class ZeRO1AdamW:
         #    replicated params only on tp rank 0; reduce over DP then TP/PP.
         #    torch.dot keeps this allocation-free (a masked pow() materializes
         #    a full fp32 copy of the 32 GB shard and can OOM at large MBS).
-        if bool(self.normmask_all):
-            sq = torch.dot(shard, shard)
-        else:
-            sq = torch.zeros((), dtype=torch.float32, device=shard.device)
-            cs = 1 << 26  # 64M elements (256 MB) per chunk
-            for s0 in range(0, shard.numel(), cs):
-                piece = shard[s0 : s0 + cs] * self.normmask_shard[s0 : s0 + cs]
-                sq += torch.dot(piece, piece)
+        sq = torch.zeros((), dtype=torch.float32, device=shard.device)
+        cs = 1 << 28  # 256M elements/chunk (BLAS dot has an int32 bound)
+        for s0 in range(0, shard.numel(), cs):
+            piece = shard[s0 : s0 + cs]
+            if not self.normmask_all:
+                piece = piece * self.normmask_shard[s0 : s0 + cs]
+            sq += torch.dot(piece, piece)
         if expert_grads and ps.get_tensor_model_parallel_rank() == 0:
             # each expert set appears expert_dp_world times across DP
             sq = sq + sum(g.pow(2).sum() for g in expert_grads) / self.expert_dp_world
