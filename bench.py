@@ -45,7 +45,7 @@ def main():
     ap.add_argument("--steps", type=int, default=8)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--gbs", type=int, default=None, help="global batch (sequences)")
-    ap.add_argument("--mbs", type=int, default=1, help="micro batch size")
+    ap.add_argument("--mbs", type=int, default=2, help="micro batch size")
     ap.add_argument("--seq", type=int, default=8192)
     ap.add_argument("--layers", type=int, default=32)
     ap.add_argument(
