@@ -80,9 +80,35 @@ def train(cfg):
     ckpt_path = None
     if cfg.get("exp_manager", {}).get("resume_if_exists") and trainer_obj.ckpt_dir:
         ckpt_path = find_latest_checkpoint(trainer_obj.ckpt_dir)
-        if ckpt_path and dist.is_initialized() is False or ckpt_path:
+        if ckpt_path:
             print(f"resuming from {ckpt_path}")
-    trainer_obj.fit(module, datamodule, ckpt_path=ckpt_path)
+
+    # fine-tune from an HF checkpoint dir: rank 0 converts once, all
+    # ranks load weights only (examples/checkpoint_converter_scripts CLI
+    # does the same offline)
+    init_path = None
+    hf_dir = cfg["model"].get("pretrained_hf_dir")
+    if hf_dir and not ckpt_path:
+        base = trainer_obj.ckpt_dir or "."
+        init_path = os.path.join(base, "hf_init.ckpt")
+        is_zero = (not dist.is_initialized()) or dist.get_rank() == 0
+        if is_zero and not os.path.exists(os.path.join(init_path, "done")):
+            from neuronx_distributed_training_amd.utils.checkpoint_convert import (
+                full_to_sharded_llama, load_hf_state,
+            )
+
+            full_to_sharded_llama(
+                load_hf_state(hf_dir), init_path,
+                tp=ps.get_tensor_model_parallel_world_size(),
+                pp=ps.get_pipeline_model_parallel_world_size(),
+                kv_replicator=int(cfg["model"].get("kv_replicator", 1)),
+                head_dim=int(cfg["model"].get("hidden_size", 4096))
+                // int(cfg["model"].get("num_attention_heads", 32)),
+            )
+        if dist.is_initialized():
+            dist.barrier()
+    trainer_obj.fit(module, datamodule, ckpt_path=ckpt_path,
+                    init_weights_path=init_path)
     return trainer_obj
 
 

@@ -165,10 +165,20 @@ class BaseModelModule:
         return running
 
     def training_step(self, microbatches) -> Dict[str, float]:
+        # phase markers: nvtx maps to roctx on ROCm — visible in
+        # rocprofv3 --sys-trace / omnitrace (tracing parity)
+        rng = torch.cuda.nvtx if self.device.type == "cuda" else None
         self.optimizer.zero_grad()
         t0 = time.perf_counter()
+        if rng:
+            rng.range_push("fwd_bwd")
         loss = self.forward_backward_step(microbatches)
+        if rng:
+            rng.range_pop()
+            rng.range_push("optimizer_step")
         gnorm = self.optimizer.step()
+        if rng:
+            rng.range_pop()
         self.scheduler.step()
         if self.device.type == "cuda":
             torch.cuda.synchronize()

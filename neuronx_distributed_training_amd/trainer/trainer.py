@@ -67,8 +67,13 @@ class Trainer:
         module: BaseModelModule,
         datamodule: BaseDataModule,
         ckpt_path: Optional[str] = None,
+        init_weights_path: Optional[str] = None,
     ):
         module.setup()
+        if init_weights_path and not ckpt_path:
+            # pretrained weights only (converted HF checkpoint); fresh
+            # optimizer/loop state (reference weight_init_only semantics)
+            self.ckpt_io.load(init_weights_path, module, weight_init_only=True)
         module.configure_optimizers(self.max_steps)
         datamodule.setup()
         for cb in self.callbacks:

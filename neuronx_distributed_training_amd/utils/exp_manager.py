@@ -52,6 +52,21 @@ class TensorBoardLogger:
                     self.writer.add_scalar(k, v, step)
 
 
+class HeartbeatCallback:
+    """Per-step heartbeat file: mtime = liveness signal for external
+    failure detection (reference-style watchdog, SURVEY.md §5.3 — the
+    reference relies on NEURON_RT_EXEC_TIMEOUT; here an external monitor
+    can alarm on a stale heartbeat and trigger torchrun restart)."""
+
+    def __init__(self, path: str):
+        self.path = path
+
+    def on_train_batch_end(self, trainer, module, metrics):
+        if _is_global_zero():
+            with open(self.path, "w") as f:
+                f.write(f"{metrics.get('global_step', 0)} {time.time()}\n")
+
+
 class TimingCallback:
     """Per-step wall time into metrics (reference TimingCallback,
     exp_manager.py:64-78, without the XLA step-closure)."""
@@ -112,4 +127,7 @@ def exp_manager(trainer, em_cfg: Dict) -> Tuple[List, Optional[str]]:
         except Exception:
             pass
     trainer.callbacks.append(TimingCallback())
+    if em_cfg.get("heartbeat_file") or em_cfg.get("enable_recovery_time_instrumentation"):
+        hb = em_cfg.get("heartbeat_file") or os.path.join(log_dir, "heartbeat")
+        trainer.callbacks.append(HeartbeatCallback(hb))
     return loggers, ckpt_dir
