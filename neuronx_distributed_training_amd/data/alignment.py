@@ -11,12 +11,9 @@ tests a trivial byte-level tokenizer is used when ``tokenizer: bytes``.
 from __future__ import annotations
 
 import json
-import os
 from typing import Dict, List
 
-import torch
-
-from .datamodule import BaseDataModule, default_collate
+from .datamodule import BaseDataModule
 from .packing import ConcatDataset, PaddedDataset, PaddedDPODataset, IGNORE_INDEX
 
 

@@ -12,7 +12,6 @@ indexed_dataset used by data/datasets/gpt_dataset_patch.py):
 
 from __future__ import annotations
 
-import os
 import struct
 from typing import List
 

@@ -4,8 +4,6 @@ max_steps·GBS, DP-rank sharded samplers, consumed-samples resume)."""
 
 from __future__ import annotations
 
-from typing import Dict, Optional
-
 import torch
 from torch.utils.data import DataLoader
 

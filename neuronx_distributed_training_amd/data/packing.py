@@ -8,7 +8,7 @@ and the DPO variant (left-padded prompts, chosen/rejected field pairs).
 
 from __future__ import annotations
 
-from typing import Dict, List
+from typing import List
 
 import torch
 from torch.utils.data import Dataset

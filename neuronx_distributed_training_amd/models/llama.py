@@ -18,7 +18,7 @@ offsets — but built MI355X-first:
 from __future__ import annotations
 
 import math
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 import torch
@@ -35,11 +35,10 @@ from ..parallel.layers import (
 from ..parallel.loss import parallel_cross_entropy
 from ..parallel.mappings import (
     gather_from_tensor_model_parallel_region,
-    reduce_scatter_to_sequence_parallel_region,
     gather_from_sequence_parallel_region,
     scatter_to_sequence_parallel_region,
 )
-from ..ops import flash_attn_func, rmsnorm, swiglu
+from ..ops import flash_attn_func, swiglu
 from ..ops.rope import apply_rotary_pos_emb, build_rope_cache
 from ..ops.rmsnorm import RMSNorm
 

@@ -17,7 +17,7 @@ convention as our Llama path; the device kernels are shared.
 from __future__ import annotations
 
 import math
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 import torch
@@ -33,7 +33,7 @@ from ..parallel.mappings import (
     scatter_to_sequence_parallel_region,
 )
 from ..modules.moe import ExpertMLPs, MoE, RouterSinkhorn, RouterTopK, load_balancing_loss_func
-from ..ops import flash_attn_func, rmsnorm, swiglu
+from ..ops import flash_attn_func, swiglu
 from ..ops.rmsnorm import RMSNorm
 from ..ops.rope import apply_rotary_pos_emb, build_rope_cache
 

@@ -14,7 +14,6 @@ expert grads then sync only over the expert-DP group (optim/zero1.py).
 
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass
 from typing import Optional
 

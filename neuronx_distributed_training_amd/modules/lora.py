@@ -6,7 +6,6 @@ lora_alpha, lora_dropout, target_modules, save merged/sharded)).
 from __future__ import annotations
 
 import math
-import re
 from dataclasses import dataclass, field
 from typing import List
 
@@ -14,7 +13,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..parallel.layers import ColumnParallelLinear, GQAQKVColumnParallelLinear, RowParallelLinear
+from ..parallel.layers import ColumnParallelLinear, RowParallelLinear
 from ..parallel.mappings import (
     copy_to_tensor_model_parallel_region,
     gather_from_sequence_parallel_region,

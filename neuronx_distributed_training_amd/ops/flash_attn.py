@@ -18,7 +18,6 @@ from __future__ import annotations
 import math
 
 import torch
-import torch.nn.functional as F
 
 from . import kernels_for
 

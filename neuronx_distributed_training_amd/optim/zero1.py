@@ -18,8 +18,7 @@ all-gather writes them in place.
 
 from __future__ import annotations
 
-import math
-from typing import Iterable, List, Optional
+from typing import Optional
 
 import torch
 import torch.distributed as dist
@@ -240,7 +239,7 @@ class ZeRO1AdamW:
         self.step_count += 1
         b1, b2 = self.betas
         t = self.step_count
-        from ..ops import have_extension, _try_load
+        from ..ops import _try_load
 
         k = _try_load() if shard.is_cuda else None
         if k is not None and hasattr(k, "adamw_step"):

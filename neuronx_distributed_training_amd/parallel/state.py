@@ -13,9 +13,8 @@ ranks.
 
 from __future__ import annotations
 
-import os
 from dataclasses import dataclass, field
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 import torch
 import torch.distributed as dist

@@ -19,7 +19,6 @@ import torch
 import torch.nn.functional as F
 from torch.utils.data import DataLoader
 
-from ..parallel import state as ps
 from ..parallel.loss import from_parallel_logits_to_logprobs
 from .module import LlamaModule
 

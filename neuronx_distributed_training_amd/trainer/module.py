@@ -10,7 +10,7 @@ XLA machinery (mark_step, step closures) is gone — eager HIP streams.
 from __future__ import annotations
 
 import time
-from typing import Dict, Iterator, Optional
+from typing import Dict, Optional
 
 import torch
 import torch.distributed as dist

@@ -8,7 +8,6 @@ logging, and max_steps/max_time budgets.
 
 from __future__ import annotations
 
-import os
 import time
 from typing import Dict, List, Optional
 

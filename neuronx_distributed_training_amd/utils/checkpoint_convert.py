@@ -15,10 +15,9 @@ Sharding rules mirror the layer definitions (parallel/layers.py):
 
 from __future__ import annotations
 
-import json
 import os
 import re
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import torch
 

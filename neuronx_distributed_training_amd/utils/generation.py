@@ -6,9 +6,6 @@ from __future__ import annotations
 
 import torch
 
-from ..parallel import state as ps
-from ..parallel.mappings import gather_from_tensor_model_parallel_region
-
 
 @torch.no_grad()
 def generate(
