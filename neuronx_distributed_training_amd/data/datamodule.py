@@ -140,4 +140,7 @@ def build_datamodule(cfg: Dict) -> BaseDataModule:
     if kind == "alignment":
         from .alignment import ModelAlignmentDataModule
         return ModelAlignmentDataModule(cfg)
+    if kind == "megatron":
+        from .megatron_module import MegatronDataModule
+        return MegatronDataModule(cfg)
     raise ValueError(f"unknown data.kind {kind}")

@@ -217,3 +217,31 @@ def test_flash_attn_sliding_window_gpu(ext):
     s2 = (qr @ kr.transpose(-1, -2) / (d ** 0.5)).masked_fill(mask, float("-inf"))
     (torch.softmax(s2, -1) @ vr).backward(g.float())
     assert (q.grad.float() - qr.grad).abs().max() / qr.grad.abs().max().clamp(min=1) < 0.05
+
+
+def test_fused_ce_sharded_stats_combine(ext):
+    """Validate the TP combination math of the fused CE on one GPU by
+    splitting the vocab into two shards and combining stats exactly as
+    parallel/loss.py does over the TP group."""
+    import torch.nn.functional as F
+
+    torch.manual_seed(9)
+    N, V = 256, 4096
+    full = torch.randn(N, 2 * V, device="cuda", dtype=torch.bfloat16)
+    target = torch.randint(0, 2 * V, (N,), device="cuda")
+    parts = [full[:, :V].contiguous(), full[:, V:].contiguous()]
+    stats = [ext.ce_fwd(p, target, i * V) for i, p in enumerate(parts)]
+    gmax = torch.maximum(stats[0][0], stats[1][0])
+    gsum = sum(s[1] * torch.exp(s[0] - gmax) for s in stats)
+    tgt = stats[0][2] + stats[1][2]
+    loss = gsum.log() + gmax - tgt
+    ref = F.cross_entropy(full.float(), target, reduction="none")
+    assert torch.allclose(loss, ref, atol=2e-2, rtol=1e-3), (loss - ref).abs().max()
+    # backward shards vs full-softmax reference
+    go = torch.randn(N, device="cuda")
+    dl0 = ext.ce_bwd(parts[0], target, gmax, gsum, go, 0)
+    dl1 = ext.ce_bwd(parts[1], target, gmax, gsum, go, V)
+    ref_in = full.float().requires_grad_(True)
+    F.cross_entropy(ref_in, target, reduction="none").backward(go)
+    got = torch.cat([dl0, dl1], dim=1).float()
+    assert (got - ref_in.grad).abs().max() < 1.5e-2

@@ -87,3 +87,85 @@ def test_checkpoint_layout(tmp_path):
     assert os.path.exists(os.path.join(root, "optim", "dp_rank_00_tp_rank_00_pp_rank_00.pt"))
     assert os.path.exists(os.path.join(root, "user_content.pt"))
     assert os.path.exists(os.path.join(root, "done"))
+
+
+def _fit_megatron(rank, world, tmpdir):
+    import numpy as np
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.trainer import Trainer
+    from neuronx_distributed_training_amd.trainer.module_megatron import (
+        MegatronGPTModule,
+    )
+    from neuronx_distributed_training_amd.data.datamodule import build_datamodule
+    from neuronx_distributed_training_amd.data.indexed_dataset import (
+        MMapIndexedDatasetBuilder,
+    )
+
+    ps.initialize_model_parallel()
+    torch.manual_seed(0)
+    prefix = os.path.join(tmpdir, "corpus")
+    if rank == 0:
+        b = MMapIndexedDatasetBuilder(prefix)
+        rng = np.random.RandomState(0)
+        for _ in range(60):
+            b.add_document(rng.randint(0, 128, size=rng.randint(20, 80)))
+        b.finalize()
+    cfg = {
+        "trainer": {"max_steps": 2, "limit_val_batches": 1},
+        "data": {
+            "kind": "megatron",
+            "data_prefix": prefix,
+            "splits_string": "90,10,0",
+            "global_batch_size": 4,
+            "micro_batch_size": 2,
+            "seq_length": 32,
+        },
+        "distributed_strategy": {},
+        "model": {
+            "model_source": "megatron",
+            "vocab_size": 128, "hidden_size": 64, "num_layers": 2,
+            "num_attention_heads": 4, "grad_clip": 1.0,
+            "optim": {"lr": 1e-3, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    tr = Trainer(cfg)
+    tr.ckpt_dir = None
+    module = MegatronGPTModule(cfg)
+    dm = build_datamodule(cfg)
+    tr.fit(module, dm)
+    return tr.global_step
+
+
+def test_megatron_trainer_e2e(tmp_path):
+    steps = run_distributed(_fit_megatron, 1, str(tmp_path))[0]
+    assert steps == 2
+
+
+def _fit_async_ckpt(rank, world, tmpdir):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.trainer import Trainer
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+    from neuronx_distributed_training_amd.data.datamodule import build_datamodule
+
+    ps.initialize_model_parallel()
+    torch.manual_seed(0)
+    cfg = _cfg(tmpdir, 3)
+    cfg["exp_manager"]["async_checkpointing"] = True
+    tr = Trainer(cfg)
+    tr.ckpt_dir = os.path.join(tmpdir, "checkpoints")
+    os.makedirs(tr.ckpt_dir, exist_ok=True)
+    module = LlamaModule(cfg)
+    dm = build_datamodule(cfg)
+    tr.fit(module, dm)
+    # all async writes must be committed by finalize()
+    tags = [n for n in os.listdir(tr.ckpt_dir) if n.endswith(".ckpt")]
+    assert tags
+    for t in tags:
+        assert os.path.exists(os.path.join(tr.ckpt_dir, t, "done")), t
+    return len(tags)
+
+
+def test_async_checkpointing(tmp_path):
+    assert run_distributed(_fit_async_ckpt, 1, str(tmp_path))[0] >= 1
