@@ -33,3 +33,25 @@ DEVINL bf16x8_t load_frag_b_rowmajorT(const __bf16* base, int stride, int k0,
   *(int4*)&f = *(const int4*)p;
   return f;
 }
+
+// --- XOR slot swizzle for TRANSPOSED images (V^T / Q^T / dO^T / K^T) ---
+// The transpose WRITER's 32-lane groups step the image row by 8, i.e.
+// 8·stride·2 B ≈ 0 (mod 128 B) for any 16B-aligned stride → every lane of
+// a group lands on ONE bank (16-way ds_write conflict; PMC measured
+// 12-15% of wave cycles). XORing byte bits 4-5 with image-row bits 3-4
+// spreads a group over eight 16 B slots; the same XOR is applied on the
+// b128 reads (both-sides-or-neither, guide §5.4 rule 21).
+DEVINL uint tr_swz(uint byte_off, int img_row) {
+  return byte_off ^ (((uint)(img_row >> 3) & 7u) << 4);
+}
+
+// swizzled 16 B read of a transposed image: abs image row r0 + (lane&15),
+// k-window k0 (+8 per upper lane group).
+DEVINL bf16x8_t load_frag_b_trT_swz(const __bf16* img, int stride, int r0,
+                                    int k0, int lane) {
+  const int r = r0 + (lane & 15);
+  const uint byte = tr_swz((uint)(r * stride + k0 + (lane >> 4) * 8) * 2, r);
+  bf16x8_t f;
+  *(int4*)&f = *(const int4*)((const char*)img + byte);
+  return f;
+}

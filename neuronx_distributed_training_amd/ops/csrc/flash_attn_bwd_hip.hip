@@ -118,8 +118,10 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
           for (int j = 0; j < 8; ++j) {
             __bf16 qp[2] = {qe0[j], qe1[j]};
             __bf16 dp[2] = {de0[j], de1[j]};
-            *(uint*)&qt_lds[(col8 + j) * VP + row] = *(uint*)qp;
-            *(uint*)&dot_lds[(col8 + j) * VP + row] = *(uint*)dp;
+            const int r = col8 + j;
+            const uint byte = tr_swz((uint)(r * VP + row) * 2, r);
+            *(uint*)((char*)qt_lds + byte) = *(uint*)qp;
+            *(uint*)((char*)dot_lds + byte) = *(uint*)dp;
           }
         }
       }
@@ -169,14 +171,14 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
         for (int nj = 0; nj < DN; ++nj) {
 #pragma unroll
           for (int kk = 0; kk < BM / 32; ++kk) {
-            bf16x8_t dob = load_frag_b_rowmajorT(&dot_lds[nj * 16 * VP], VP,
-                                                 kk * 32, lane);
+            bf16x8_t dob =
+                load_frag_b_trT_swz(dot_lds, VP, nj * 16, kk * 32, lane);
             bf16x8_t pa0 = load_frag_a(pw, VP, kk * 32, lane);
             bf16x8_t pa1 = load_frag_a(pw + 16 * VP, VP, kk * 32, lane);
             dvacc[0][nj] = MFMA_16x16x32(pa0, dob, dvacc[0][nj]);
             dvacc[1][nj] = MFMA_16x16x32(pa1, dob, dvacc[1][nj]);
-            bf16x8_t qb2 = load_frag_b_rowmajorT(&qt_lds[nj * 16 * VP], VP,
-                                                 kk * 32, lane);
+            bf16x8_t qb2 =
+                load_frag_b_trT_swz(qt_lds, VP, nj * 16, kk * 32, lane);
             bf16x8_t da0 = load_frag_a(dw, VP, kk * 32, lane);
             bf16x8_t da1 = load_frag_a(dw + 16 * VP, VP, kk * 32, lane);
             dkacc[0][nj] = MFMA_16x16x32(da0, qb2, dkacc[0][nj]);
@@ -309,7 +311,9 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           __bf16 pr[2] = {e0[j], e1[j]};
-          *(uint*)&kt_lds[(col8 + j) * VP + row] = *(uint*)pr;
+          const int r = col8 + j;
+          *(uint*)((char*)kt_lds + tr_swz((uint)(r * VP + row) * 2, r)) =
+              *(uint*)pr;
         }
       }
     }
@@ -354,7 +358,7 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
 #pragma unroll
         for (int kk = 0; kk < BN / 32; ++kk) {
           bf16x8_t kb2 =
-              load_frag_b_rowmajorT(&kt_lds[nj * 16 * VP], VP, kk * 32, lane);
+              load_frag_b_trT_swz(kt_lds, VP, nj * 16, kk * 32, lane);
           bf16x8_t da0 = load_frag_a(dsw, VP, kk * 32, lane);
           bf16x8_t da1 = load_frag_a(dsw + 16 * VP, VP, kk * 32, lane);
           dqacc[0][nj] = MFMA_16x16x32(da0, kb2, dqacc[0][nj]);

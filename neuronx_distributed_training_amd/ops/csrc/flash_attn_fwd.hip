@@ -130,7 +130,9 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         __bf16 pair[2] = {e0[j], e1[j]};
-        *(uint*)&vt_lds[(col8 + j) * VP + row] = *(uint*)pair;
+        const int r = col8 + j;
+        *(uint*)((char*)vt_lds + tr_swz((uint)(r * VP + row) * 2, r)) =
+            *(uint*)pair;
       }
     }
   };
@@ -237,7 +239,7 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
 #pragma unroll
         for (int kk = 0; kk < BN / 32; ++kk) {
           bf16x8_t vb =
-              load_frag_b_rowmajorT(&vt_lds[nj * 16 * VP], VP, kk * 32, lane);
+              load_frag_b_trT_swz(vt_lds, VP, nj * 16, kk * 32, lane);
           bf16x8_t pa0 = load_frag_a(pw, VP, kk * 32, lane);
           bf16x8_t pa1 = load_frag_a(pw + 16 * VP, VP, kk * 32, lane);
           oacc[0][nj] = MFMA_16x16x32(pa0, vb, oacc[0][nj]);
