@@ -58,6 +58,10 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
           *(const int4*)(Kp + row * sKs + kk * 32 + (lane >> 4) * 8);
       *(int4*)&vfrag[sb][kk] =
           *(const int4*)(Vp + row * sVs + kk * 32 + (lane >> 4) * 8);
+      // fold the softmax scale into K once (VALU-bound kernel)
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        kfrag[sb][kk][j] = (__bf16)((float)kfrag[sb][kk][j] * scale);
     }
   }
 
@@ -158,7 +162,7 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
               bool dead =
                   (qcol >= S) || (krow >= S) || (CAUSAL && qcol < krow);
               if (CAUSAL && window > 0) dead |= (qcol >= krow + window);
-              const float p = dead ? 0.f : __expf(st[sb][r] * scale - lse);
+              const float p = dead ? 0.f : __expf(st[sb][r] - lse);
               const float ds = p * (dpt[sb][r] - delta) * scale;
               const int lrow = sb * 16 + (lane >> 4) * 4 + r;
               pw[lrow * VP + nq * 16 + (lane & 15)] = (__bf16)p;
@@ -261,6 +265,10 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
           *(const int4*)(Qp + row * sQs + kk * 32 + (lane >> 4) * 8);
       *(int4*)&dofrag[sb][kk] =
           *(const int4*)(dOp + row * sDs + kk * 32 + (lane >> 4) * 8);
+      // fold the softmax scale into Q once (VALU-bound kernel)
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        qfrag[sb][kk][j] = (__bf16)((float)qfrag[sb][kk][j] * scale);
     }
 #pragma unroll
     for (int rr = 0; rr < 4; ++rr) {
@@ -345,8 +353,7 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
             const int qrow = qrow_w + sb * 16 + (lane >> 4) * 4 + r;
             bool dead = (kcol >= S) || (CAUSAL && kcol > qrow);
             if (CAUSAL && window > 0) dead |= (kcol <= qrow - window);
-            const float p =
-                dead ? 0.f : __expf(st[sb][r] * scale - lse[sb][r]);
+            const float p = dead ? 0.f : __expf(st[sb][r] - lse[sb][r]);
             const float ds = p * (dpt[sb][r] - delta[sb][r]) * scale;
             dsw[(sb * 16 + (lane >> 4) * 4 + r) * VP + nk * 16 +
                 (lane & 15)] = (__bf16)ds;

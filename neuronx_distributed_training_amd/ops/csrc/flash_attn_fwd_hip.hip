@@ -56,10 +56,22 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
     for (int kk = 0; kk < DK; ++kk) {
       const bf16* p = Qp + row * sQs + kk * 32 + (lane >> 4) * 8;
       *(int4*)&qfrag[sb][kk] = *(const int4*)p;
+      // fold the softmax scale into Q once (saves a VALU mul per score
+      // per tile; PMC: these kernels are VALU-bound at ~8 VALU/MFMA)
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        qfrag[sb][kk][j] = (__bf16)((float)qfrag[sb][kk][j] * scale);
     }
   }
 
   float m_i[2][4], l_i[2][4];
+  float alpha_s[2][4];
+  // all-ones B fragment: one MFMA per 32-key chunk computes the P row-sums
+  // into every lane's accumulator (replaces 16 adds + 16 shuffles per
+  // sub-block of VALU reduction)
+  bf16x8_t ones_frag;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) ones_frag[j] = (__bf16)1.0f;
   constexpr int DN = D / 16;
   f32x4_t oacc[2][DN];
 #pragma unroll
@@ -177,7 +189,7 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
           for (int nk = 0; nk < 4; ++nk) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-              const float s = sacc[sb][nk][r] * scale;
+              const float s = sacc[sb][nk][r];  // scale folded into Q
               sv[nk][r] = s;
               tile_max[r] = fmaxf(tile_max[r], s);
             }
@@ -189,7 +201,7 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
               const int qrow = qrow_w + sb * 16 + (lane >> 4) * 4 + r;
-              float s = sacc[sb][nk][r] * scale;
+              float s = sacc[sb][nk][r];  // scale folded into Q
               bool dead = (kcol >= S) || (CAUSAL && kcol > qrow);
               if (CAUSAL && window > 0) dead |= (kcol <= qrow - window);
               s = dead ? -1e30f : s;
@@ -204,37 +216,41 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
           for (int off = 1; off < 16; off <<= 1)
             tile_max[r] = fmaxf(tile_max[r], __shfl_xor(tile_max[r], off, 64));
         }
-        float alpha[4], rsum[4] = {0.f, 0.f, 0.f, 0.f};
+        // rescale is EXACTLY the identity when no row's max grew — skip
+        // the alpha exps and the O/l rescale entirely then (wave-uniform
+        // vote; most interior tiles after the first few don't move m).
+        bool grew = false;
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const float mn = fmaxf(m_i[sb][r], tile_max[r]);
-          alpha[r] = __expf(m_i[sb][r] - mn);
-          m_i[sb][r] = mn;
+        for (int r = 0; r < 4; ++r) grew |= tile_max[r] > m_i[sb][r];
+        if (__builtin_amdgcn_ballot_w64(grew) != 0ull) {
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const float mn = fmaxf(m_i[sb][r], tile_max[r]);
+            alpha_s[sb][r] = __expf(m_i[sb][r] - mn);
+            m_i[sb][r] = mn;
+          }
+#pragma unroll
+          for (int nj = 0; nj < DN; ++nj) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) oacc[sb][nj][r] *= alpha_s[sb][r];
+          }
+        } else {
+#pragma unroll
+          for (int r = 0; r < 4; ++r) alpha_s[sb][r] = 1.0f;
         }
 #pragma unroll
         for (int nk = 0; nk < 4; ++nk) {
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
             const float p = __expf(sv[nk][r] - m_i[sb][r]);
-            rsum[r] += p;
             pw[(sb * 16 + (lane >> 4) * 4 + r) * VP + nk * 16 + (lane & 15)] =
                 (__bf16)p;
           }
         }
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-#pragma unroll
-          for (int off = 1; off < 16; off <<= 1)
-            rsum[r] += __shfl_xor(rsum[r], off, 64);
-          l_i[sb][r] = l_i[sb][r] * alpha[r] + rsum[r];
-        }
-#pragma unroll
-        for (int nj = 0; nj < DN; ++nj) {
-#pragma unroll
-          for (int r = 0; r < 4; ++r) oacc[sb][nj][r] *= alpha[r];
-        }
       }
       // ---- O += P V (V^T B-frags loaded once per sub-block pair) ----
+      f32x4_t racc[2];
+      racc[0] = racc[1] = f32x4_t{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int nj = 0; nj < DN; ++nj) {
 #pragma unroll
@@ -245,8 +261,17 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
           bf16x8_t pa1 = load_frag_a(pw + 16 * VP, VP, kk * 32, lane);
           oacc[0][nj] = MFMA_16x16x32(pa0, vb, oacc[0][nj]);
           oacc[1][nj] = MFMA_16x16x32(pa1, vb, oacc[1][nj]);
+          if (nj == 0) {
+            racc[0] = MFMA_16x16x32(pa0, ones_frag, racc[0]);
+            racc[1] = MFMA_16x16x32(pa1, ones_frag, racc[1]);
+          }
         }
       }
+#pragma unroll
+      for (int sb = 0; sb < 2; ++sb)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          l_i[sb][r] = l_i[sb][r] * alpha_s[sb][r] + racc[sb][r];
     }
     __syncthreads();  // all waves done reading this tile's LDS
     if (jb + 1 < nkb) {
