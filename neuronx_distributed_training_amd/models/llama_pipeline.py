@@ -63,15 +63,26 @@ class LlamaChunk(nn.Module):
 
     def hidden_shape_for(self, batch):
         b, s = batch["input_ids"].shape
+        if self.cfg.sequence_parallel:
+            s = s // ps.get_tensor_model_parallel_world_size()
         return (s, b, self.cfg.hidden_size)
 
     def forward(self, x):
         if self.is_first:
             x = self.embed_tokens(self._batch["input_ids"]).transpose(0, 1).contiguous()
+            if self.cfg.sequence_parallel:
+                from ..parallel.mappings import (
+                    scatter_to_sequence_parallel_region,
+                )
+                x = scatter_to_sequence_parallel_region(x)
         for layer in self.layers:
             x = layer(x, self.rope_cos, self.rope_sin, 0)
         if not self.is_last:
             return x
+        if self.cfg.sequence_parallel:
+            from ..parallel.mappings import gather_from_sequence_parallel_region
+
+            x = gather_from_sequence_parallel_region(x)
         x = self.norm(x)
         logits = self.lm_head(x).transpose(0, 1)
         labels = self._batch.get("labels", self._batch["input_ids"])
@@ -174,6 +185,9 @@ class LlamaStage(nn.Module):
 
     def hidden_shape_for(self, batch):
         b, s = batch["input_ids"].shape
+        # under SP the stage-boundary activation is the [s/tp, b, h] shard
+        if self.cfg.sequence_parallel:
+            s = s // ps.get_tensor_model_parallel_world_size()
         return (s, b, self.cfg.hidden_size)
 
     def forward(self, x: Optional[torch.Tensor]):
@@ -181,11 +195,21 @@ class LlamaStage(nn.Module):
         if self.is_first:
             ids = self._batch["input_ids"]
             x = self.embed_tokens(ids).transpose(0, 1).contiguous()
-        pos_offset = cp_rank * x.size(0)
+            if self.cfg.sequence_parallel:
+                from ..parallel.mappings import (
+                    scatter_to_sequence_parallel_region,
+                )
+                x = scatter_to_sequence_parallel_region(x)
+        seq_full = self._batch["input_ids"].size(1)
+        pos_offset = cp_rank * seq_full
         for layer in self.layers:
             x = layer(x, self.rope_cos, self.rope_sin, pos_offset)
         if not self.is_last:
             return x
+        if self.cfg.sequence_parallel:
+            from ..parallel.mappings import gather_from_sequence_parallel_region
+
+            x = gather_from_sequence_parallel_region(x)
         x = self.norm(x)
         logits = self.lm_head(x).transpose(0, 1)  # [b, s, v/tp]
         labels = self._batch.get("labels", self._batch["input_ids"])

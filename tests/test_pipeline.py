@@ -144,3 +144,34 @@ def test_vp2_loss_matches_reference():
     vp = run_distributed(_vp_loss, 2, 2)
     assert abs(vp[0] - vp[1]) < 1e-6
     assert abs(vp[0] - ref) < 5e-3, (vp[0], ref)
+
+
+def _pp_sp_loss(rank, world):
+    """PP=2 × TP=2 × SP on 4 ranks vs the non-distributed reference."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.llama import LlamaConfig
+    from neuronx_distributed_training_amd.models.llama_pipeline import LlamaStage
+    from neuronx_distributed_training_amd.trainer.pipeline import PipelineEngine
+    import torch.distributed as dist
+
+    ps.initialize_model_parallel(
+        tensor_model_parallel_size=2, pipeline_model_parallel_size=2
+    )
+    torch.manual_seed(7)
+    cfg = LlamaConfig(**TINY, sequence_parallel=True)
+    stage = LlamaStage(cfg)
+    eng = PipelineEngine(stage)
+    g = torch.Generator().manual_seed(99)
+    micro = []
+    for _ in range(4):
+        ids = torch.randint(0, 128, (1, 32), generator=g)
+        micro.append({"input_ids": ids, "labels": ids.clone()})
+    loss = eng.run_train(micro).float()
+    dist.all_reduce(loss, group=ps.get_pipeline_model_parallel_group())
+    return float(loss)
+
+
+def test_pp2_tp2_sp_matches_reference():
+    ref = run_distributed(_ref_loss, 1, False)[0]
+    res = run_distributed(_pp_sp_loss, 4)
+    assert abs(res[0] - ref) < 5e-3, (res[0], ref)
