@@ -88,3 +88,32 @@ def _train_decreases(rank, world):
 @pytest.mark.parametrize("world", [1, 2])
 def test_training_loss_decreases(world):
     run_distributed(_train_decreases, world)
+
+
+def _fused_qkv_loss(rank, world):
+    """num_heads == num_kv_heads → single fused stride-3 QKV GEMM path."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.llama import (
+        LlamaConfig, LlamaForCausalLM,
+    )
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(7)
+    cfg = LlamaConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=4,
+        max_position_embeddings=32, fuse_qkv=True,
+    )
+    model = LlamaForCausalLM(cfg)
+    assert hasattr(model.model.layers[0].self_attn, "qkv_proj")
+    g = torch.Generator().manual_seed(99)
+    ids = torch.randint(0, 128, (2, 32), generator=g)
+    loss = model(ids, labels=ids)
+    loss.backward()
+    return float(loss.detach())
+
+
+def test_fused_qkv_tp2_matches_tp1():
+    l1 = run_distributed(_fused_qkv_loss, 1)[0]
+    l2 = run_distributed(_fused_qkv_loss, 2)
+    assert abs(l1 - l2[0]) < 5e-3, (l1, l2[0])
