@@ -91,11 +91,25 @@ class ModelAlignmentDataModule(BaseDataModule):
             return
 
         # SFT: prompt tokens get IGNORE_INDEX labels (loss on completion
-        # only — reference model_alignment_data_module.py:148-160)
+        # only — reference model_alignment_data_module.py:148-160).
+        # Optional prompt/completion templates replace the reference's
+        # promptsource integration: "{field}"-style format strings over
+        # the row's columns.
+        sft_cfg = self.align.get("sft", {}) or {}
+        p_tmpl = sft_cfg.get("prompt_template")
+        c_tmpl = sft_cfg.get("completion_template")
         samples = []
         for r in rows:
-            p_ids = tok.encode(_get(r, "prompt", "instruction", "input"))
-            c_ids = tok.encode(_get(r, "completion", "output", "response")) + [eos]
+            if p_tmpl:
+                prompt = p_tmpl.format(**r)
+            else:
+                prompt = _get(r, "prompt", "instruction", "input")
+            if c_tmpl:
+                completion = c_tmpl.format(**r)
+            else:
+                completion = _get(r, "completion", "output", "response")
+            p_ids = tok.encode(prompt)
+            c_ids = tok.encode(completion) + [eos]
             samples.append(
                 {
                     "input_ids": p_ids + c_ids,
