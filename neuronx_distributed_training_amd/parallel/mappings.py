@@ -30,8 +30,24 @@ def _tp_world() -> int:
     return ps.get_tensor_model_parallel_world_size()
 
 
+_REDUCE_DTYPE = None  # None = reduce in the activation dtype
+
+
+def set_reduce_dtype(dtype):
+    """Optional wider dtype for TP activation reductions (reference
+    ``reduce_dtype`` knob, llama_model.py:67-74): e.g. float32 to remove
+    bf16 summation error across the TP group at ~2× comm cost."""
+    global _REDUCE_DTYPE
+    _REDUCE_DTYPE = dtype
+
+
 def _all_reduce(x: torch.Tensor) -> torch.Tensor:
     if _tp_world() == 1:
+        return x
+    if _REDUCE_DTYPE is not None and x.dtype != _REDUCE_DTYPE:
+        xr = x.to(_REDUCE_DTYPE)
+        dist.all_reduce(xr, group=ps.get_tensor_model_parallel_group())
+        x.copy_(xr.to(x.dtype))
         return x
     dist.all_reduce(x, group=ps.get_tensor_model_parallel_group())
     return x

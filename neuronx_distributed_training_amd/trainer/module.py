@@ -244,6 +244,11 @@ class LlamaModule(BaseModelModule):
 
     def build_model(self) -> torch.nn.Module:
         mcfg = self.cfg["model"]
+        rd = mcfg.get("reduce_dtype")
+        if rd:
+            from ..parallel.mappings import set_reduce_dtype
+
+            set_reduce_dtype(getattr(torch, str(rd)))
         precision = str(self.cfg.get("precision", {}).get("type", "bf16"))
         want_bf16 = ("bf16" in precision) or ("mixed" in precision)
         # CPU runs stay fp32 (bf16 matmul is unusably slow off-GPU)
