@@ -1,0 +1,6 @@
+from .datamodule import (  # noqa: F401
+    BaseDataModule,
+    HFDataModule,
+    SyntheticDataModule,
+    build_datamodule,
+)
