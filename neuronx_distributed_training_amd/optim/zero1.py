@@ -40,6 +40,8 @@ class ZeRO1AdamW:
         weight_decay: float = 0.01,
         grad_clip: float = 1.0,
         no_decay_keys=("bias", "norm"),
+        overlap_grad_reduce: bool = False,
+        bucket_cap_mb: int = 128,
     ):
         self.lr = lr
         self.betas = betas
@@ -150,6 +152,97 @@ class ZeRO1AdamW:
             1, self.dp_world // ps.get_expert_model_parallel_world_size()
         )
 
+        # ---- optional comm/compute overlap: bucketed grad all-reduce
+        # launched from backward hooks as each bucket's grads complete
+        # (the eager analog of the compiler-scheduled overlap the
+        # reference gets for free — SURVEY.md §7 hard-parts). Uses
+        # all-reduce (not reduce-scatter) so the global shard layout,
+        # masks and checkpoints are unchanged; step() then skips its own
+        # reduction. Only active at DP > 1.
+        self.overlap_grad_reduce = bool(overlap_grad_reduce) and self.dp_world > 1
+        self._pending_works = []
+        self._sync_enabled = False  # armed for the LAST microbatch only
+        if self.overlap_grad_reduce:
+            bucket_elems = max(bucket_cap_mb, 1) * (1 << 20) // 4
+            # grads arrive roughly in reverse parameter order during
+            # backward; bucket by reversed order so early buckets fill first
+            order = list(range(len(self.named_params)))[::-1]
+            self._param_bucket = {}
+            self._buckets = []  # list of dicts: {params, remaining, lo, hi}
+            cur, cur_elems = [], 0
+            for i in order:
+                cur.append(i)
+                cur_elems += self.named_params[i][1].numel()
+                if cur_elems >= bucket_elems:
+                    self._buckets.append(cur)
+                    cur, cur_elems = [], 0
+            if cur:
+                self._buckets.append(cur)
+            self._bucket_state = []
+            for bi, idxs in enumerate(self._buckets):
+                lo = min(self.offsets[i] for i in idxs)
+                hi = max(
+                    self.offsets[i] + _pad_to(self.named_params[i][1].numel(), 128)
+                    for i in idxs
+                )
+                self._bucket_state.append({"idxs": idxs, "lo": lo, "hi": hi,
+                                           "remaining": len(idxs)})
+                for i in idxs:
+                    self._param_bucket[i] = bi
+            self._hook_handles = []
+            for i, (n, p) in enumerate(self.named_params):
+                self._hook_handles.append(
+                    p.register_post_accumulate_grad_hook(self._make_hook(i))
+                )
+
+    def _make_hook(self, i):
+        def hook(p):
+            # fires after EVERY grad accumulation; only the final
+            # microbatch (enable_grad_sync armed) launches communication
+            if not self._sync_enabled:
+                return
+            o = self.offsets[i]
+            self.grad_flat[o : o + p.numel()].copy_(p.grad.reshape(-1))
+            bi = self._param_bucket[i]
+            st = self._bucket_state[bi]
+            st["remaining"] -= 1
+            if st["remaining"] == 0:
+                w = dist.all_reduce(
+                    self.grad_flat[st["lo"] : st["hi"]],
+                    group=self.dp_group, async_op=True,
+                )
+                self._pending_works.append(w)
+        return hook
+
+    def enable_grad_sync(self):
+        """Arm the backward hooks before the LAST microbatch's backward
+        (grad-accumulation boundary)."""
+        if self.overlap_grad_reduce:
+            self._sync_enabled = True
+
+    def _finish_overlap_reduce(self):
+        # flush buckets whose params produced no grad this step
+        for st in self._bucket_state:
+            if st["remaining"] > 0:
+                for i in st["idxs"]:
+                    p = self.named_params[i][1]
+                    o = self.offsets[i]
+                    if p.grad is not None:
+                        self.grad_flat[o : o + p.numel()].copy_(p.grad.reshape(-1))
+                self._pending_works.append(
+                    dist.all_reduce(
+                        self.grad_flat[st["lo"] : st["hi"]],
+                        group=self.dp_group, async_op=True,
+                    )
+                )
+                st["remaining"] = 0
+        for w in self._pending_works:
+            w.wait()
+        self._pending_works = []
+        for st in self._bucket_state:
+            st["remaining"] = len(st["idxs"])
+        self._sync_enabled = False
+
     # -- hooks the trainer uses --
     def zero_grad(self, set_to_none: bool = True):
         for _, p in self.named_params:
@@ -172,15 +265,23 @@ class ZeRO1AdamW:
 
     @torch.no_grad()
     def step(self) -> torch.Tensor:
-        self._collect_grads()
-        # 1) reduce-scatter fp32 grads over DP (average)
-        if self.dp_world > 1:
+        # 1) grads → fp32, synced over DP: either the backward-overlapped
+        #    bucketed all-reduce (flag) or one reduce-scatter here
+        if self.overlap_grad_reduce:
+            self._finish_overlap_reduce()
+            shard = self.grad_flat[
+                self.shard_start : self.shard_start + self.shard_size
+            ]
+            shard.div_(self.dp_world)
+        elif self.dp_world > 1:
+            self._collect_grads()
             shard = torch.empty(
                 self.shard_size, dtype=torch.float32, device=self.device
             )
             dist.reduce_scatter_tensor(shard, self.grad_flat, group=self.dp_group)
             shard.div_(self.dp_world)
         else:
+            self._collect_grads()
             shard = self.grad_flat[
                 self.shard_start : self.shard_start + self.shard_size
             ]

@@ -70,6 +70,7 @@ class BaseModelModule:
 
     def configure_optimizers(self, max_steps: int):
         ocfg = self.cfg["model"].get("optim", {})
+        dstr = self.cfg.get("distributed_strategy", {})
         lr = float(ocfg.get("lr", 3e-4))
         self.optimizer = ZeRO1AdamW(
             list(self.model.named_parameters()),
@@ -78,6 +79,10 @@ class BaseModelModule:
             eps=float(ocfg.get("eps", 1e-8)),
             weight_decay=float(ocfg.get("weight_decay", 0.01)),
             grad_clip=float(self.cfg["model"].get("grad_clip", 1.0)),
+            overlap_grad_reduce=bool(dstr.get("overlap_grad_reduce", False)),
+            bucket_cap_mb=int(
+                self.cfg.get("runtime", {}).get("bucket_cap_mb", 128) or 128
+            ),
         )
         sched_cfg = ocfg.get("sched", {})
         self.scheduler = build_scheduler(
@@ -146,13 +151,17 @@ class BaseModelModule:
             return running
         running = torch.zeros((), dtype=torch.float32, device=self.device)
         n = self.num_microbatches
-        for batch in microbatches:
+        batches = list(microbatches)
+        for bi, batch in enumerate(batches):
             batch = {
                 k: (v.to(self.device, non_blocking=True) if torch.is_tensor(v) else v)
                 for k, v in batch.items()
             }
             batch = self.get_batch_on_this_context_parallel_rank(batch)
             loss = self.model_fwd_calc_loss(batch)
+            if bi == len(batches) - 1 and hasattr(self.optimizer, "enable_grad_sync"):
+                # final microbatch: overlap the DP grad reduce with backward
+                self.optimizer.enable_grad_sync()
             (loss / n).backward()
             running += loss.detach().float()
         running /= n

@@ -99,3 +99,43 @@ def _zero1_ckpt(rank, world):
 
 def test_zero1_state_roundtrip():
     run_distributed(_zero1_ckpt, 1)
+
+
+def _zero1_overlap(rank, world):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.optim.zero1 import ZeRO1AdamW
+
+    ps.initialize_model_parallel()
+    m = _make_model()
+    opt = ZeRO1AdamW(
+        list(m.named_parameters()), lr=1e-2, grad_clip=1.0,
+        overlap_grad_reduce=True, bucket_cap_mb=1,
+    )
+    torch.manual_seed(50)
+    for step in range(3):
+        xfull = torch.randn(8, 16)
+        yfull = torch.randn(8, 4)
+        per = 8 // world
+        x = xfull[rank * per : (rank + 1) * per]
+        y = yfull[rank * per : (rank + 1) * per]
+        opt.zero_grad()
+        # two microbatches: only the second arms the sync
+        for mb in range(2):
+            xm = x[mb::2]
+            ym = y[mb::2]
+            if mb == 1:
+                opt.enable_grad_sync()
+            (((m(xm) - ym) ** 2).mean() / 2).backward()
+        opt.step()
+    return torch.cat([p.detach().reshape(-1) for p in m.parameters()])
+
+
+def test_zero1_overlap_matches_plain():
+    base = run_distributed(_zero1_dp, 2)  # plain reduce-scatter path
+    # overlapped path must land on the same weights modulo averaging of
+    # half-batches (same data split, same microbatching as _zero1_dp? no —
+    # _zero1_dp does a single backward; compare overlap dp2 vs overlap dp1)
+    o1 = run_distributed(_zero1_overlap, 1)
+    o2 = run_distributed(_zero1_overlap, 2)
+    assert torch.allclose(o1[0], o2[0], atol=1e-5), (o1[0] - o2[0]).abs().max()
+    assert torch.allclose(o2[0], o2[1])
