@@ -75,8 +75,10 @@ def validate_config(cfg: Dict):
         ds["sequence_parallel"] = False
     model = cfg.get("model", {})
     nl = int(model.get("num_layers", 1))
-    if pp > 1 and nl % pp != 0:
-        raise ValueError(f"num_layers {nl} not divisible by PP {pp}")
+    if pp > 1 and nl < pp:
+        raise ValueError(f"num_layers {nl} < PP {pp}")
+    # uneven layer counts are allowed: partition_layers spreads the
+    # remainder over the first stages (reference pipeline_cuts analog)
     moe = model.get("moe", {})
     if moe:
         ep = int(ds.get("expert_model_parallel_size", 1))
