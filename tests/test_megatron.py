@@ -177,3 +177,33 @@ def _gpt_moe(rank, world):
 
 def test_gpt_moe_layer():
     run_distributed(_gpt_moe, 1)
+
+
+def _cca(rank, world):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.modules.chunked_cross_attention import (
+        ParallelChunkedCrossAttention,
+    )
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(0)
+    s, b, h, m = 16, 2, 32, 4
+    num_chunks, r_tot = s // m, 6
+    cca = ParallelChunkedCrossAttention(h, num_attention_heads=4, chunk_size=m,
+                                        init_seed=42)
+    hidden = torch.randn(s, b, h, requires_grad=True)
+    retrieved = torch.randn(num_chunks, r_tot, b, h, requires_grad=True)
+    out = cca(hidden, retrieved)
+    assert out.shape == (s, b, h)
+    # RETRO shift: first chunk_size-1 positions see no retrieval
+    assert (out[: m - 1] == 0).all()
+    out.sum().backward()
+    assert torch.isfinite(hidden.grad).all()
+    assert torch.isfinite(retrieved.grad).all()
+    return out.detach().sum()
+
+
+def test_chunked_cross_attention():
+    r1 = run_distributed(_cca, 1)
+    r2 = run_distributed(_cca, 2)
+    assert torch.allclose(r1[0], r2[0], atol=1e-3), (r1[0], r2[0])
