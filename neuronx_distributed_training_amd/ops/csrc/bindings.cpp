@@ -166,17 +166,23 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
                    .sum(-1)
                    .contiguous();  // [B, HQ, S] f32
   auto dq_mem = torch::empty({S, B, HQ, D}, q.options());
-  auto dk_mem = torch::empty({S, B, HKV, D}, q.options());
-  auto dv_mem = torch::empty({S, B, HKV, D}, q.options());
+  // dkv writes one fp32 partial slab per q-head of each GQA group (full
+  // grid occupancy at high TP); sum + cast here
+  int group = HQ / HKV;
+  auto f32 = q.options().dtype(torch::kFloat32);
+  auto dk_part = torch::empty({group, S, B, HKV, D}, f32);
+  auto dv_part = torch::empty({group, S, B, HKV, D}, f32);
   long qs[3] = {q.stride(2), q.stride(0), q.stride(1)};
   long ks[3] = {k.stride(2), k.stride(0), k.stride(1)};
   long vs[3] = {v.stride(2), v.stride(0), v.stride(1)};
   long ds[3] = {dout.stride(2), dout.stride(0), dout.stride(1)};
   launch_flash_bwd(dout.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                    lse.data_ptr(), delta.data_ptr(), dq_mem.data_ptr(),
-                   dk_mem.data_ptr(), dv_mem.data_ptr(), B, HQ, HKV, S, D,
+                   dk_part.data_ptr(), dv_part.data_ptr(), B, HQ, HKV, S, D,
                    causal, (float)scale, (int)window, qs, ks, vs, ds,
                    cur_stream());
+  auto dk_mem = dk_part.sum(0).to(torch::kBFloat16);
+  auto dv_mem = dv_part.sum(0).to(torch::kBFloat16);
   return {dq_mem.permute({1, 2, 0, 3}), dk_mem.permute({1, 2, 0, 3}),
           dv_mem.permute({1, 2, 0, 3})};
 }

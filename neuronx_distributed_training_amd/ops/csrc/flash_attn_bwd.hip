@@ -17,8 +17,12 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
     const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     const bf16* __restrict__ K, const bf16* __restrict__ V,
     const float* __restrict__ LSE, const float* __restrict__ DELTA,
-    bf16* __restrict__ dK,  // [s, b, hkv, d] contiguous
-    bf16* __restrict__ dV, int S, int Bb, int HQ, int HKV, float scale,
+    float* __restrict__ dK,  // [group, s, b, hkv, d] fp32 PARTIALS (one
+    float* __restrict__ dV,  // slab per q-head of the GQA group; the
+                             // binding sums over dim 0) — one block per
+                             // (key-block, b, hkv, g) keeps the grid full
+                             // even at TP=8 where HKV_local = 1
+    int S, int Bb, int HQ, int HKV, float scale,
     int window, long sQs, long sQb, long sQh, long sKs, long sKb, long sKh,
     long sVs, long sVb, long sVh, long sDs, long sDb, long sDh) {
   constexpr int BN = 256;  // keys per block
@@ -35,10 +39,11 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
   const int kblock = blockIdx.x;
-  const int bh = blockIdx.y;  // b * HKV + hkv
-  const int hkv = bh % HKV;
-  const int b = bh / HKV;
   const int group = HQ / HKV;
+  const int bhg = blockIdx.y;  // (b * HKV + hkv) * group + g
+  const int g = bhg % group;
+  const int hkv = (bhg / group) % HKV;
+  const int b = bhg / (group * HKV);
 
   const bf16* Kp = K + b * sKb + hkv * sKh;
   const bf16* Vp = V + b * sVb + hkv * sVh;
@@ -76,7 +81,7 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
 
   const int wkey_min = krow_w;  // first key of this wave
 
-  for (int g = 0; g < group; ++g) {
+  {
     const int hq = hkv * group + g;
     const bf16* Qp = Q + b * sQb + hq * sQh;
     const bf16* dOp = dO + b * sDb + hq * sDh;
@@ -193,10 +198,11 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
     }
   }
 
-  // ---- store dK, dV as [s, b, hkv, d] contiguous ----
+  // ---- store fp32 partials: [group][s][b][hkv][d] ----
   const long sOs = (long)Bb * HKV * D;
-  bf16* dKp = dK + ((long)b * HKV + hkv) * D;
-  bf16* dVp = dV + ((long)b * HKV + hkv) * D;
+  const long slab = (long)g * S * sOs + ((long)b * HKV + hkv) * D;
+  float* dKp = dK + slab;
+  float* dVp = dV + slab;
 #pragma unroll
   for (int sb = 0; sb < 2; ++sb) {
 #pragma unroll
@@ -205,10 +211,8 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
       for (int r = 0; r < 4; ++r) {
         const int krow = krow_w + sb * 16 + (lane >> 4) * 4 + r;
         if (krow < S) {
-          dKp[(long)krow * sOs + nj * 16 + (lane & 15)] =
-              f2bf(dkacc[sb][nj][r]);
-          dVp[(long)krow * sOs + nj * 16 + (lane & 15)] =
-              f2bf(dvacc[sb][nj][r]);
+          dKp[(long)krow * sOs + nj * 16 + (lane & 15)] = dkacc[sb][nj][r];
+          dVp[(long)krow * sOs + nj * 16 + (lane & 15)] = dvacc[sb][nj][r];
         }
       }
     }
@@ -398,14 +402,15 @@ void launch_flash_bwd(const void* dout, const void* q, const void* k,
                       const long* qstr, const long* kstr, const long* vstr,
                       const long* dostr, hipStream_t stream) {
   dim3 blk(512);
-  dim3 gkv((S + 255) / 256, B * HKV);
+  dim3 gkv((S + 255) / 256, B * HQ);  // one block per (kblock, b, hkv, g)
   dim3 gq((S + 255) / 256, B * HQ);
 #define CASE(DD, CC)                                                          \
   do {                                                                        \
     flash_bwd_dkv_kernel<DD, CC><<<gkv, blk, 0, stream>>>(                    \
         (const bf16*)dout, (const bf16*)q, (const bf16*)k, (const bf16*)v,    \
-        (const float*)lse, (const float*)delta, (bf16*)dk, (bf16*)dv, S, B,   \
-        HQ, HKV, scale, window, qstr[0], qstr[1], qstr[2], kstr[0], kstr[1],  \
+        (const float*)lse, (const float*)delta, (float*)dk, (float*)dv, S,   \
+        B, HQ, HKV, scale, window, qstr[0], qstr[1], qstr[2], kstr[0],        \
+        kstr[1],                                                              \
         kstr[2], vstr[0], vstr[1], vstr[2], dostr[0], dostr[1], dostr[2]);    \
     flash_bwd_dq_kernel<DD, CC><<<gq, blk, 0, stream>>>(                      \
         (const bf16*)dout, (const bf16*)q, (const bf16*)k, (const bf16*)v,    \
