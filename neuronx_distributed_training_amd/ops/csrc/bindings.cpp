@@ -25,6 +25,7 @@ void launch_flash_bwd(const void*, const void*, const void*, const void*,
                       int, int, int, bool, float, int, const long*,
                       const long*, const long*, const long*, hipStream_t);
 void launch_mfma_probe(const void*, const void*, void*, hipStream_t);
+void launch_mfma_probe32(const void*, const void*, void*, hipStream_t);
 void launch_ce_fwd(const void*, const void*, void*, void*, void*, long, int,
                    long, hipStream_t);
 void launch_ce_bwd(const void*, const void*, const void*, const void*,
@@ -195,6 +196,14 @@ torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
+torch::Tensor mfma_probe32(torch::Tensor a, torch::Tensor b) {
+  CHECK_IN(a);
+  CHECK_IN(b);
+  auto c = torch::empty({32, 32}, a.options().dtype(torch::kFloat32));
+  launch_mfma_probe32(a.data_ptr(), b.data_ptr(), c.data_ptr(), cur_stream());
+  return c;
+}
+
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor targets,
                                   long vocab_start) {
   CHECK_IN(logits);
@@ -243,6 +252,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("o"), pybind11::arg("lse"), pybind11::arg("causal"),
         pybind11::arg("scale"), pybind11::arg("window") = 0);
   m.def("mfma_probe", &mfma_probe);
+  m.def("mfma_probe32", &mfma_probe32);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
 }

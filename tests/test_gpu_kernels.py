@@ -2,6 +2,7 @@
 All tests here are @gpu (run via gpurun / the driver's GPU tier)."""
 
 import math
+import os
 
 import pytest
 import torch
@@ -245,3 +246,19 @@ def test_fused_ce_sharded_stats_combine(ext):
     F.cross_entropy(ref_in, target, reduction="none").backward(go)
     got = torch.cat([dl0, dl1], dim=1).float()
     assert (got - ref_in.grad).abs().max() < 1.5e-2
+
+
+@pytest.mark.skipif(
+    os.environ.get("NXDT_ATTN_V3") != "1",
+    reason="32x32 maps are v3 groundwork; enable with NXDT_ATTN_V3=1",
+)
+def test_mfma32_layout_probe(ext):
+    """32×32×16 fragment-map probe for the planned v3 attention
+    (ROADMAP.md §1). Asymmetric operands catch transposes."""
+    torch.manual_seed(0)
+    a = (torch.randn(32, 16) * 0.5).bfloat16().cuda()
+    b = (torch.arange(16 * 32).float().reshape(16, 32) % 7 - 3).bfloat16().cuda()
+    c = ext.mfma_probe32(a.contiguous(), b.contiguous())
+    ref = a.float() @ b.float()
+    torch.cuda.synchronize()
+    assert torch.allclose(c, ref, atol=1e-2, rtol=1e-2), (c - ref).abs().max()
