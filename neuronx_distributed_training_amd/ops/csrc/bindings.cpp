@@ -26,6 +26,9 @@ void launch_flash_bwd(const void*, const void*, const void*, const void*,
                       const long*, const long*, const long*, hipStream_t);
 void launch_mfma_probe(const void*, const void*, void*, hipStream_t);
 void launch_mfma_probe32(const void*, const void*, void*, hipStream_t);
+void launch_flash_fwd_v3(const void*, const void*, const void*, void*, void*,
+                         int, int, int, int, int, bool, float, int,
+                         const long*, const long*, const long*, hipStream_t);
 void launch_ce_fwd(const void*, const void*, void*, void*, void*, long, int,
                    long, hipStream_t);
 void launch_ce_bwd(const void*, const void*, const void*, const void*,
@@ -126,6 +129,29 @@ static void check_bhsd(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda() && t.dim() == 4 && t.stride(3) == 1,
               name, " must be [b, h, s, d] on GPU with d contiguous");
   TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+}
+
+std::vector<torch::Tensor> flash_attn_fwd_v3(torch::Tensor q, torch::Tensor k,
+                                             torch::Tensor v, bool causal,
+                                             double scale, long window) {
+  // T12 swapped-operand forward (ROADMAP §1); dark until HW-validated
+  check_bhsd(q, "q");
+  check_bhsd(k, "k");
+  check_bhsd(v, "v");
+  int B = (int)q.size(0), HQ = (int)q.size(1), S = (int)q.size(2),
+      D = (int)q.size(3);
+  int HKV = (int)k.size(1);
+  TORCH_CHECK(D == 128 || D == 64, "head_dim must be 64 or 128");
+  auto o_mem = torch::empty({S, B, HQ, D}, q.options());
+  auto lse = torch::empty({B, HQ, S}, q.options().dtype(torch::kFloat32));
+  long qs[3] = {q.stride(2), q.stride(0), q.stride(1)};
+  long ks[3] = {k.stride(2), k.stride(0), k.stride(1)};
+  long vs[3] = {v.stride(2), v.stride(0), v.stride(1)};
+  launch_flash_fwd_v3(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                      o_mem.data_ptr(), lse.data_ptr(), B, HQ, HKV, S, D,
+                      causal, (float)scale, (int)window, qs, ks, vs,
+                      cur_stream());
+  return {o_mem.permute({1, 2, 0, 3}), lse};
 }
 
 std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
@@ -253,6 +279,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("scale"), pybind11::arg("window") = 0);
   m.def("mfma_probe", &mfma_probe);
   m.def("mfma_probe32", &mfma_probe32);
+  m.def("flash_attn_fwd_v3", &flash_attn_fwd_v3, pybind11::arg("q"),
+        pybind11::arg("k"), pybind11::arg("v"), pybind11::arg("causal"),
+        pybind11::arg("scale"), pybind11::arg("window") = 0);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
 }
