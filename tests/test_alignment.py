@@ -169,3 +169,33 @@ def _lora_run(rank, world):
 @pytest.mark.parametrize("world", [1, 2])
 def test_lora(world):
     run_distributed(_lora_run, world)
+
+
+def _dpo_tp_run(rank, world, tmpdir, mode):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.alignment import DPOModule
+    from neuronx_distributed_training_amd.data.datamodule import build_datamodule
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(0)
+    cfg = _dpo_cfg(tmpdir, os.path.join(tmpdir, "dpo.jsonl"), mode)
+    cfg["distributed_strategy"] = {"tensor_model_parallel_size": world}
+    module = DPOModule(cfg)
+    module.setup()
+    module.configure_optimizers(max_steps=4)
+    dm = build_datamodule(cfg)
+    dm.setup()
+    module.on_train_start(dm)
+    loader = iter(dm.train_dataloader())
+    micro = list(dm.microbatch_iterator(loader))
+    metrics = module.training_step(micro)
+    return metrics["reduced_train_loss"]
+
+
+def test_dpo_tp2(tmp_path):
+    """DPO training step under TP=2 (TP-aware sequence logprobs)."""
+    path = os.path.join(str(tmp_path), "dpo.jsonl")
+    _write_dpo(path)
+    l1 = run_distributed(_dpo_tp_run, 1, str(tmp_path), "dpo")[0]
+    l2 = run_distributed(_dpo_tp_run, 2, str(tmp_path), "dpo")
+    assert abs(l1 - l2[0]) < 5e-3, (l1, l2[0])

@@ -82,3 +82,41 @@ def test_cp2_loss_matches_cp1():
     l2 = run_distributed(_cp_train_loss, 2)
     assert abs(l2[0] - l2[1]) < 1e-6
     assert abs(l1 - l2[0]) < 0.05, (l1, l2[0])
+
+
+def _cp_tp_train(rank, world):
+    """CP=2 × TP=2 on 4 ranks: training step loss vs single-rank."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+
+    ps.initialize_model_parallel(
+        tensor_model_parallel_size=2, context_parallel_size=2
+    )
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 2, "seq_length": 32},
+        "distributed_strategy": {"tensor_model_parallel_size": 2,
+                                 "context_parallel_size": 2},
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 2, "num_attention_heads": 4, "num_kv_heads": 2,
+            "grad_clip": 1.0, "optim": {"lr": 1e-3, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(3)
+    mod = LlamaModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=10)
+    g = torch.Generator().manual_seed(5)
+    ids = torch.randint(0, 128, (2, 32), generator=g)
+    micro = [{"input_ids": ids, "labels": ids.clone()}]
+    m = mod.training_step(micro)
+    return m["reduced_train_loss"]
+
+
+def test_cp2_tp2_loss_matches_single():
+    l1 = run_distributed(_cp_train_loss, 1)[0]
+    l4 = run_distributed(_cp_tp_train, 4)
+    assert max(abs(l - l4[0]) for l in l4) < 1e-6  # consistent across ranks
+    assert abs(l1 - l4[0]) < 0.05, (l1, l4[0])
