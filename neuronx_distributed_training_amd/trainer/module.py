@@ -198,6 +198,7 @@ class BaseModelModule:
             "reduced_train_loss": float(loss),
             "lr": self.optimizer.lr,
             "throughput_seq_s": self.throughput.value,
+            "throughput_peak_seq_s": self.throughput.peak,
             "step_time_s": dt,
         }
         if self.log_grad_norm and gnorm is not None:
