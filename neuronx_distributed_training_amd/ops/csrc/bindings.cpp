@@ -29,6 +29,10 @@ void launch_mfma_probe32(const void*, const void*, void*, hipStream_t);
 void launch_flash_fwd_v3(const void*, const void*, const void*, void*, void*,
                          int, int, int, int, int, bool, float, int,
                          const long*, const long*, const long*, hipStream_t);
+void launch_flash_bwd_v3(const void*, const void*, const void*, const void*,
+                         const void*, const void*, void*, void*, void*, int,
+                         int, int, int, int, bool, float, int, const long*,
+                         const long*, const long*, const long*, hipStream_t);
 void launch_ce_fwd(const void*, const void*, void*, void*, void*, long, int,
                    long, hipStream_t);
 void launch_ce_bwd(const void*, const void*, const void*, const void*,
@@ -152,6 +156,41 @@ std::vector<torch::Tensor> flash_attn_fwd_v3(torch::Tensor q, torch::Tensor k,
                       causal, (float)scale, (int)window, qs, ks, vs,
                       cur_stream());
   return {o_mem.permute({1, 2, 0, 3}), lse};
+}
+
+std::vector<torch::Tensor> flash_attn_bwd_v3(torch::Tensor dout,
+                                             torch::Tensor q, torch::Tensor k,
+                                             torch::Tensor v, torch::Tensor o,
+                                             torch::Tensor lse, bool causal,
+                                             double scale, long window) {
+  if (dout.stride(3) != 1) dout = dout.contiguous();
+  check_bhsd(dout, "dout");
+  check_bhsd(q, "q");
+  int B = (int)q.size(0), HQ = (int)q.size(1), S = (int)q.size(2),
+      D = (int)q.size(3);
+  int HKV = (int)k.size(1);
+  auto delta = (dout.to(torch::kFloat32) * o.to(torch::kFloat32))
+                   .sum(-1)
+                   .contiguous();
+  auto dq_mem = torch::empty({S, B, HQ, D}, q.options());
+  int group = HQ / HKV;
+  auto f32 = q.options().dtype(torch::kFloat32);
+  auto dk_part = torch::empty({group, S, B, HKV, D}, f32);
+  auto dv_part = torch::empty({group, S, B, HKV, D}, f32);
+  long qs[3] = {q.stride(2), q.stride(0), q.stride(1)};
+  long ks[3] = {k.stride(2), k.stride(0), k.stride(1)};
+  long vs[3] = {v.stride(2), v.stride(0), v.stride(1)};
+  long dstr[3] = {dout.stride(2), dout.stride(0), dout.stride(1)};
+  launch_flash_bwd_v3(dout.data_ptr(), q.data_ptr(), k.data_ptr(),
+                      v.data_ptr(), lse.data_ptr(), delta.data_ptr(),
+                      dq_mem.data_ptr(), dk_part.data_ptr(),
+                      dv_part.data_ptr(), B, HQ, HKV, S, D, causal,
+                      (float)scale, (int)window, qs, ks, vs, dstr,
+                      cur_stream());
+  auto dk_mem = dk_part.sum(0).to(torch::kBFloat16);
+  auto dv_mem = dv_part.sum(0).to(torch::kBFloat16);
+  return {dq_mem.permute({1, 2, 0, 3}), dk_mem.permute({1, 2, 0, 3}),
+          dv_mem.permute({1, 2, 0, 3})};
 }
 
 std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
@@ -281,6 +320,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe32", &mfma_probe32);
   m.def("flash_attn_fwd_v3", &flash_attn_fwd_v3, pybind11::arg("q"),
         pybind11::arg("k"), pybind11::arg("v"), pybind11::arg("causal"),
+        pybind11::arg("scale"), pybind11::arg("window") = 0);
+  m.def("flash_attn_bwd_v3", &flash_attn_bwd_v3, pybind11::arg("dout"),
+        pybind11::arg("q"), pybind11::arg("k"), pybind11::arg("v"),
+        pybind11::arg("o"), pybind11::arg("lse"), pybind11::arg("causal"),
         pybind11::arg("scale"), pybind11::arg("window") = 0);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);

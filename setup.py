@@ -27,6 +27,7 @@ sources = [
         "adamw.hip",
         "flash_attn_fwd.hip",
         "flash_attn_fwd_v3.hip",
+        "flash_attn_bwd_v3.hip",
         "flash_attn_bwd.hip",
         "cross_entropy.hip",
         "probe.hip",
