@@ -207,27 +207,29 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_v3_kernel(
 }
 
 // ============================ dK/dV (v3) ============================
-template <int D, bool CAUSAL>
+// Split into two single-output instantiations (WHICH: 0 = dV, 1 = dK) —
+// a fused dkv at this tiling allocates 256 VGPR + ~200 B/lane scratch;
+// recomputing S in a second launch is cheaper than the spill.
+template <int D, bool CAUSAL, int WHICH>
 __global__ __launch_bounds__(512) void flash_bwd_dkv_v3_kernel(
     const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     const bf16* __restrict__ K, const bf16* __restrict__ V,
     const float* __restrict__ LSE, const float* __restrict__ DELTA,
-    float* __restrict__ dK,  // [group][s][b][hkv][d] fp32 partials
-    float* __restrict__ dV, int S, int Bb, int HQ, int HKV, float scale,
-    int window, long sQs, long sQb, long sQh, long sKs, long sKb, long sKh,
-    long sVs, long sVb, long sVh, long sDs, long sDb, long sDh) {
+    float* __restrict__ OUT,  // [group][s][b][hkv][d] fp32 partials
+    int S, int Bb, int HQ, int HKV, float scale, int window, long sQs,
+    long sQb, long sQh, long sKs, long sKb, long sKh, long sVs, long sVb,
+    long sVh, long sDs, long sDb, long sDh) {
   constexpr int BN = 256, BM = 32;  // keys per block, q tile
   constexpr int KP = D + 8;
   constexpr int VP = BM + 8;
   __shared__ __bf16 q_lds[BM * KP];
-  __shared__ __bf16 do_lds[BM * KP];
-  __shared__ __bf16 qt_lds[D * VP];   // Q^T  (tr_swz) for dK B-frags
-  __shared__ __bf16 dot_lds[D * VP];  // dO^T (tr_swz) for dV B-frags
+  __shared__ __bf16 aux_lds[BM * KP];  // dO rows (WHICH==1 only)
+  __shared__ __bf16 t_lds[D * VP];     // dO^T (dV) / Q^T (dK), tr_swz
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
   const int hi = lane >> 5;
-  const int kcol = lane & 31;  // this lane's key within the wave tile
+  const int kcol = lane & 31;
   const int kblock = blockIdx.x;
   const int group = HQ / HKV;
   const int bhg = blockIdx.y;  // (b * HKV + hkv) * group + g
@@ -247,9 +249,9 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_v3_kernel(
   const int krow_w = kbase + wid * 32;
   const int my_k = krow_w + kcol;
 
-  // per-lane B-frags of K^T and V^T (own key's row, scale folded into K)
   constexpr int CK = D / 16;
-  bf16x8_t kb[CK], vb[CK];
+  bf16x8_t kb[CK];           // K^T B-frags (own key row, scale folded)
+  bf16x8_t vb[WHICH ? CK : 1];  // V^T B-frags only needed for dK (dP)
   {
     const long row = (my_k < S) ? my_k : (S - 1);
 #pragma unroll
@@ -258,17 +260,15 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_v3_kernel(
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         kb[ck][j] = (__bf16)((float)kb[ck][j] * scale);
-      *(int4*)&vb[ck] = *(const int4*)(Vp + row * sVs + ck * 16 + hi * 8);
+      if constexpr (WHICH == 1)
+        *(int4*)&vb[ck] = *(const int4*)(Vp + row * sVs + ck * 16 + hi * 8);
     }
   }
 
   constexpr int DC = D / 32;
-  f32x16_t dkacc[DC], dvacc[DC];  // C: col = d, rows = key
+  f32x16_t acc[DC];  // C: col = d, rows = key
 #pragma unroll
-  for (int dc = 0; dc < DC; ++dc) {
-    dkacc[dc] = f32x16_t{};
-    dvacc[dc] = f32x16_t{};
-  }
+  for (int dc = 0; dc < DC; ++dc) acc[dc] = f32x16_t{};
 
   const int ib0 = CAUSAL ? (kbase / BM) : 0;
   int nqb = (S + BM - 1) / BM;
@@ -278,75 +278,76 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_v3_kernel(
 
   for (int ib = ib0; ib < nqb; ++ib) {
     const int qbase = ib * BM;
-    {  // stage Q, dO row-major + transposed (tr_swz)
+    {  // stage: q rows always; dO rows (dK); dO^T or Q^T image
       const int t = threadIdx.x;
       const int row = t / (D / 8);
       const int col8 = (t % (D / 8)) * 8;
       const int gr = qbase + row;
       int4 qv = (gr < S) ? *(const int4*)(Qp + (long)gr * sQs + col8)
                          : int4{0, 0, 0, 0};
-      int4 dv = (gr < S) ? *(const int4*)(dOp + (long)gr * sDs + col8)
-                         : int4{0, 0, 0, 0};
       *(int4*)&q_lds[row * KP + col8] = qv;
-      *(int4*)&do_lds[row * KP + col8] = dv;
+      if constexpr (WHICH == 1) {
+        int4 dv = (gr < S) ? *(const int4*)(dOp + (long)gr * sDs + col8)
+                           : int4{0, 0, 0, 0};
+        *(int4*)&aux_lds[row * KP + col8] = dv;
+      }
       if (t < BM * D / 16) {
         const int prow = (t / (D / 8)) * 2;
         const int pcol8 = (t % (D / 8)) * 8;
         const int g0 = qbase + prow, g1 = g0 + 1;
-        int4 a0 = (g0 < S) ? *(const int4*)(Qp + (long)g0 * sQs + pcol8)
+        const bf16* src = WHICH ? Qp : dOp;
+        const long stride = WHICH ? sQs : sDs;
+        int4 a0 = (g0 < S) ? *(const int4*)(src + (long)g0 * stride + pcol8)
                            : int4{0, 0, 0, 0};
-        int4 a1 = (g1 < S) ? *(const int4*)(Qp + (long)g1 * sQs + pcol8)
+        int4 a1 = (g1 < S) ? *(const int4*)(src + (long)g1 * stride + pcol8)
                            : int4{0, 0, 0, 0};
-        int4 d0 = (g0 < S) ? *(const int4*)(dOp + (long)g0 * sDs + pcol8)
-                           : int4{0, 0, 0, 0};
-        int4 d1 = (g1 < S) ? *(const int4*)(dOp + (long)g1 * sDs + pcol8)
-                           : int4{0, 0, 0, 0};
-        const __bf16 *qe0 = (const __bf16*)&a0, *qe1 = (const __bf16*)&a1;
-        const __bf16 *de0 = (const __bf16*)&d0, *de1 = (const __bf16*)&d1;
+        const __bf16 *e0 = (const __bf16*)&a0, *e1 = (const __bf16*)&a1;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          __bf16 qp[2] = {qe0[j], qe1[j]};
-          __bf16 dp[2] = {de0[j], de1[j]};
+          __bf16 pr[2] = {e0[j], e1[j]};
           const int r = pcol8 + j;
-          const uint byte = tr_swz((uint)(r * VP + prow) * 2, r);
-          *(uint*)((char*)qt_lds + byte) = *(uint*)qp;
-          *(uint*)((char*)dot_lds + byte) = *(uint*)dp;
+          *(uint*)((char*)t_lds + tr_swz((uint)(r * VP + prow) * 2, r)) =
+              *(uint*)pr;
         }
       }
     }
     __syncthreads();
 
     if (!CAUSAL || wkey_min <= qbase + BM - 1) {
-      // ---- S = Q·K^T and dP = dO·V^T (C: col = key, rows = q) ----
+      // ---- S = Q·K^T (C: col = key, rows = q); dP = dO·V^T (dK) ----
       f32x16_t st = f32x16_t{};
       f32x16_t dpt = f32x16_t{};
 #pragma unroll
       for (int ck = 0; ck < CK; ++ck) {
-        bf16x8_t qa, da;
+        bf16x8_t qa;
         *(int4*)&qa = *(const int4*)&q_lds[(lane & 31) * KP + ck * 16 + hi * 8];
-        *(int4*)&da =
-            *(const int4*)&do_lds[(lane & 31) * KP + ck * 16 + hi * 8];
         st = MFMA_32x32x16(qa, kb[ck], st);
-        dpt = MFMA_32x32x16(da, vb[ck], dpt);
+        if constexpr (WHICH == 1) {
+          bf16x8_t da;
+          *(int4*)&da =
+              *(const int4*)&aux_lds[(lane & 31) * KP + ck * 16 + hi * 8];
+          dpt = MFMA_32x32x16(da, vb[ck], dpt);
+        }
       }
-      // per-reg q row; P and dS in the same layout
-      float pv[16], dsv[16];
+      float val[16];  // P (dV) or dS (dK), per q row of the C map
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qrow = qbase + (r & 3) + 8 * (r >> 2) + 4 * hi;
         const float lse = (qrow < S) ? Lp[qrow] : 1e30f;
-        const float delta = (qrow < S) ? Dp[qrow] : 0.f;
         bool dead = (qrow >= S) || (my_k >= S) || (CAUSAL && my_k > qrow);
         if (CAUSAL && window > 0) dead |= (my_k <= qrow - window);
         const float p = dead ? 0.f : __expf(st[r] - lse);
-        pv[r] = p;
-        dsv[r] = p * (dpt[r] - delta) * scale;
+        if constexpr (WHICH == 1) {
+          const float delta = (qrow < S) ? Dp[qrow] : 0.f;
+          val[r] = p * (dpt[r] - delta) * scale;
+        } else {
+          val[r] = p;
+        }
       }
-      // A-frags (rows = key = lane&31) of P^T and dS^T via half exchange
-      bf16x8_t pf[2], dsf[2];
-      build_halffrags(pv, pf);
-      build_halffrags(dsv, dsf);
-      // ---- dV += P^T·dO ; dK += dS^T·Q ----
+      // A-frags (rows = key = lane&31) via the wave-half exchange
+      bf16x8_t af[2];
+      build_halffrags(val, af);
+      // ---- acc += P^T·dO (dV) or dS^T·Q (dK) ----
 #pragma unroll
       for (int dc = 0; dc < DC; ++dc) {
 #pragma unroll
@@ -354,11 +355,9 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_v3_kernel(
           const int drow = dc * 32 + (lane & 31);
           const uint byte =
               tr_swz((uint)(drow * VP + kk * 16 + hi * 8) * 2, drow);
-          bf16x8_t dofr, qfr;
-          *(int4*)&dofr = *(const int4*)((const char*)dot_lds + byte);
-          *(int4*)&qfr = *(const int4*)((const char*)qt_lds + byte);
-          dvacc[dc] = MFMA_32x32x16(pf[kk], dofr, dvacc[dc]);
-          dkacc[dc] = MFMA_32x32x16(dsf[kk], qfr, dkacc[dc]);
+          bf16x8_t bf;
+          *(int4*)&bf = *(const int4*)((const char*)t_lds + byte);
+          acc[dc] = MFMA_32x32x16(af[kk], bf, acc[dc]);
         }
       }
     }
@@ -367,18 +366,14 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_v3_kernel(
 
   // ---- store fp32 partials [group][s][b][hkv][d]; C rows = key ----
   const long sOs = (long)Bb * HKV * D;
-  const long slab = (long)g * S * sOs + ((long)b * HKV + hkv) * D;
-  float* dKp = dK + slab;
-  float* dVp = dV + slab;
+  float* Op = OUT + (long)g * S * sOs + ((long)b * HKV + hkv) * D;
 #pragma unroll
   for (int dc = 0; dc < DC; ++dc) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int krow = krow_w + (r & 3) + 8 * (r >> 2) + 4 * hi;
-      if (krow < S) {
-        dKp[(long)krow * sOs + dc * 32 + (lane & 31)] = dkacc[dc][r];
-        dVp[(long)krow * sOs + dc * 32 + (lane & 31)] = dvacc[dc][r];
-      }
+      if (krow < S)
+        Op[(long)krow * sOs + dc * 32 + (lane & 31)] = acc[dc][r];
     }
   }
 }
@@ -396,12 +391,16 @@ void launch_flash_bwd_v3(const void* dout, const void* q, const void* k,
   dim3 gq((S + 255) / 256, B * HQ);
 #define CASEB3(DD, CC)                                                        \
   do {                                                                        \
-   hipLaunchKernelGGL(( flash_bwd_dkv_v3_kernel<DD, CC>), dim3(gkv), dim3(blk), 0, stream,                  \
+   hipLaunchKernelGGL(( flash_bwd_dkv_v3_kernel<DD, CC, 0>), dim3(gkv), dim3(blk), 0, stream,               \
         (const bf16*)dout, (const bf16*)q, (const bf16*)k, (const bf16*)v,    \
-        (const float*)lse, (const float*)delta, (float*)dk, (float*)dv, S,    \
-        B, HQ, HKV, scale, window, qstr[0], qstr[1], qstr[2], kstr[0],        \
-        kstr[1], kstr[2], vstr[0], vstr[1], vstr[2], dostr[0], dostr[1],      \
-        dostr[2]);                                                            \
+        (const float*)lse, (const float*)delta, (float*)dv, S, B, HQ, HKV,    \
+        scale, window, qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2],  \
+        vstr[0], vstr[1], vstr[2], dostr[0], dostr[1], dostr[2]);             \
+   hipLaunchKernelGGL(( flash_bwd_dkv_v3_kernel<DD, CC, 1>), dim3(gkv), dim3(blk), 0, stream,               \
+        (const bf16*)dout, (const bf16*)q, (const bf16*)k, (const bf16*)v,    \
+        (const float*)lse, (const float*)delta, (float*)dk, S, B, HQ, HKV,    \
+        scale, window, qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2],  \
+        vstr[0], vstr[1], vstr[2], dostr[0], dostr[1], dostr[2]);             \
    hipLaunchKernelGGL(( flash_bwd_dq_v3_kernel<DD, CC>), dim3(gq), dim3(blk), 0, stream,                    \
         (const bf16*)dout, (const bf16*)q, (const bf16*)k, (const bf16*)v,    \
         (const float*)lse, (const float*)delta, (bf16*)dq, S, B, HQ, HKV,     \
