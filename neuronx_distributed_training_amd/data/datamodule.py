@@ -44,6 +44,15 @@ def default_collate(samples: List[Dict[str, torch.Tensor]]):
     return {k: torch.stack([s[k] for s in samples]) for k in samples[0]}
 
 
+def pad_vocab_size(orig_vocab: int, make_divisible_by: int = 8,
+                   tp_size: int = 1) -> int:
+    """Pad the vocabulary so it divides evenly over TP ranks with
+    GEMM-friendly alignment (reference data/base.py:66-89: pad to a
+    multiple of ``make_vocab_size_divisible_by * tp``)."""
+    mult = make_divisible_by * tp_size
+    return ((orig_vocab + mult - 1) // mult) * mult
+
+
 class BaseDataModule:
     """DP-size math + consumed-samples bookkeeping (reference data/base.py)."""
 

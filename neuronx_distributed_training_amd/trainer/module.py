@@ -264,8 +264,16 @@ class LlamaModule(BaseModelModule):
         # CPU runs stay fp32 (bf16 matmul is unusably slow off-GPU)
         dtype = "bfloat16" if (want_bf16 and torch.cuda.is_available()) else "float32"
         dstr = self.cfg.get("distributed_strategy", {})
+        from ..data.datamodule import pad_vocab_size
+
+        tp = ps.get_tensor_model_parallel_world_size()
+        vocab = pad_vocab_size(
+            int(mcfg.get("vocab_size", 128256)),
+            int(mcfg.get("make_vocab_size_divisible_by", 8)),
+            tp,
+        )
         cfg = LlamaConfig(
-            vocab_size=int(mcfg.get("vocab_size", 128256)),
+            vocab_size=vocab,
             hidden_size=int(mcfg.get("hidden_size", 4096)),
             intermediate_size=int(mcfg.get("intermediate_size", 14336)),
             num_hidden_layers=int(mcfg.get("num_layers", 32)),

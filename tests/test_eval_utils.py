@@ -94,3 +94,12 @@ def test_config_validation_rejects(tmp_path):
     )
     with pytest.raises(ValueError):
         load_config(p)
+
+
+def test_pad_vocab_size():
+    from neuronx_distributed_training_amd.data.datamodule import pad_vocab_size
+
+    assert pad_vocab_size(128256, 8, 8) == 128256      # already aligned
+    assert pad_vocab_size(32000, 8, 8) == 32000
+    assert pad_vocab_size(32003, 8, 8) == 32064        # padded up
+    assert pad_vocab_size(50257, 128, 1) == 50304      # classic GPT-2 pad
