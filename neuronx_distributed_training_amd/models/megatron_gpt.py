@@ -67,6 +67,7 @@ class GPTConfig:
     moe_frequency: int = 1
     moe_router_type: str = "top_k"
     moe_capacity_factor: Optional[float] = None
+    token_shuffle_group_size: int = 1
     moe_aux_loss_coeff: float = 0.01
 
     @property
@@ -146,6 +147,7 @@ class NeuronSwitchMLP(nn.Module):
             ExpertMLPs(cfg.num_moe_experts, cfg.hidden_size, cfg.ffn_size,
                        dtype=cfg.torch_dtype, init_seed=4001 + layer_idx),
             capacity_factor=cfg.moe_capacity_factor,
+            token_shuffle_group_size=cfg.token_shuffle_group_size,
         )
 
     def forward(self, x):

@@ -43,6 +43,7 @@ class MixtralConfig(LlamaConfig):
     router_aux_loss_coef: float = 0.02
     router_type: str = "top_k"       # top_k | sinkhorn
     capacity_factor: Optional[float] = None  # None = dropless
+    token_shuffle_group_size: int = 1
     sliding_window: Optional[int] = None
 
     @classmethod
@@ -80,6 +81,7 @@ class MixtralSparseMoeBlock(nn.Module):
                 dtype=cfg.torch_dtype, init_seed=seed + 1,
             ),
             capacity_factor=cfg.capacity_factor,
+            token_shuffle_group_size=cfg.token_shuffle_group_size,
         )
 
     def forward(self, x):

@@ -29,7 +29,7 @@ from ..parallel import state as ps
 from ..ops import swiglu
 
 __all__ = ["RouterTopK", "RouterSinkhorn", "ExpertMLPs", "MoE",
-           "load_balancing_loss_func"]
+           "load_balancing_loss_func", "token_shuffle", "token_unshuffle"]
 
 
 class _AllToAll(torch.autograd.Function):
@@ -51,6 +51,63 @@ class _AllToAll(torch.autograd.Function):
             out, grad.contiguous(), ctx.in_splits, ctx.out_splits, group=ctx.group
         )
         return out, None, None, None
+
+
+# shared call counter so every rank in a shuffle group derives the same
+# permutation seed without communication (layers step in lockstep).
+_SHUFFLE_CALLS = 0
+
+
+def _route_tokens(x: torch.Tensor, slots_all: torch.Tensor, group, r: int):
+    """Move token i of rank src to global slot slots_all[src, i] within the
+    shuffle group (slot g*T..(g+1)*T-1 lives on rank g). slots_all is a
+    permutation of range(g*T), identical on every rank. Differentiable."""
+    g, T = slots_all.shape
+    slots = slots_all[r]
+    dest = torch.div(slots, T, rounding_mode="floor")
+    order = torch.argsort(dest, stable=True)
+    in_splits = torch.bincount(dest, minlength=g).tolist()
+    dest_all = torch.div(slots_all, T, rounding_mode="floor")
+    out_splits = [(dest_all[src] == r).sum().item() for src in range(g)]
+    recv = _AllToAll.apply(
+        x[order.to(x.device)].contiguous(), out_splits, in_splits, group
+    )
+    # received rows arrive grouped by src rank, original-order within src;
+    # place each at its target in-slab position.
+    recv_slots = torch.cat([slots_all[src][dest_all[src] == r] for src in range(g)])
+    pos = recv_slots - r * T
+    inv = torch.empty(T, dtype=torch.long)
+    inv[pos] = torch.arange(T)
+    return recv[inv.to(x.device)]
+
+
+def token_shuffle(x: torch.Tensor, group_size: int):
+    """Randomly permute tokens across `group_size` adjacent DP ranks before
+    MoE routing so capacity-mode load is balanced across the group
+    (reference transformer.py:463 / megatron_gpt_model.py:134
+    ``token_shuffle_group_size``). Returns (shuffled_x, ctx); pass ctx to
+    ``token_unshuffle`` to restore order. x: [T, H], same T on every rank."""
+    global _SHUFFLE_CALLS
+    _SHUFFLE_CALLS += 1
+    if group_size <= 1 or not dist.is_initialized():
+        return x, None
+    group, r = ps.get_token_shuffle_group(group_size)
+    T = x.size(0)
+    gen = torch.Generator().manual_seed(0x51F0 + _SHUFFLE_CALLS)
+    perm = torch.randperm(group_size * T, generator=gen)
+    inv = torch.empty_like(perm)
+    inv[perm] = torch.arange(group_size * T)
+    y = _route_tokens(x, inv.view(group_size, T), group, r)
+    return y, (perm.view(group_size, T), group, r)
+
+
+def token_unshuffle(y: torch.Tensor, ctx):
+    """Inverse of ``token_shuffle`` (returns tokens to their home rank and
+    original order)."""
+    if ctx is None:
+        return y
+    perm, group, r = ctx
+    return _route_tokens(y, perm, group, r)
 
 
 def load_balancing_loss_func(router_logits: torch.Tensor, num_experts: int,
@@ -179,16 +236,22 @@ class MoE(nn.Module):
     """
 
     def __init__(self, router: RouterTopK, experts: ExpertMLPs,
-                 capacity_factor: Optional[float] = None):
+                 capacity_factor: Optional[float] = None,
+                 token_shuffle_group_size: int = 1):
         super().__init__()
         self.router = router
         self.experts = experts
         self.capacity_factor = capacity_factor
+        self.token_shuffle_group_size = token_shuffle_group_size
         self.ep = ps.get_expert_model_parallel_world_size()
 
     def forward(self, x: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
         """x: [T, H] (caller flattens [s,b,h]). Returns (y [T, H],
-        router_logits [T, E])."""
+        router_logits [T, E] — computed on the shuffled token order when
+        token_shuffle_group_size > 1; the aux loss is order-invariant)."""
+        shuf_ctx = None
+        if self.token_shuffle_group_size > 1:
+            x, shuf_ctx = token_shuffle(x, self.token_shuffle_group_size)
         T, H = x.shape
         k = self.router.top_k
         E = self.router.num_experts
@@ -263,4 +326,6 @@ class MoE(nn.Module):
         # combine: scatter-add weighted expert outputs back to tokens
         y = torch.zeros_like(x)
         y.index_add_(0, sorted_tok, yexp * sorted_w.unsqueeze(-1))
+        if shuf_ctx is not None:
+            y = token_unshuffle(y, shuf_ctx)
         return y, logits

@@ -336,6 +336,32 @@ def get_tensor_model_parallel_replica_group(size: int):
     return _REPLICA_GROUPS[key]
 
 
+def get_token_shuffle_group(size: int):
+    """Sub-groups of `size` adjacent DP ranks for MoE token shuffle
+    (reference transformer.py:463 ``token_shuffle_group_size`` → NxD MoE).
+    Returns (group, rank_in_group). Lazily created; all ranks must reach
+    this with the same size in the same order (true: MoE layers are
+    constructed and stepped symmetrically across ranks)."""
+    s = _st()
+    key = ("tok_shuf", size)
+    if key not in _REPLICA_GROUPS:
+        grid = _build_grid(s.world_size, s.tp, s.cp, s.dp, s.pp)
+        my = None
+        my_r = 0
+        for p in range(s.pp):
+            for c in range(s.cp):
+                for t in range(s.tp):
+                    dp_ranks = grid[p, :, c, t].tolist()
+                    for start in range(0, len(dp_ranks), size):
+                        ranks = dp_ranks[start : start + size]
+                        g = dist.new_group(ranks)
+                        if s.rank in ranks:
+                            my = g
+                            my_r = ranks.index(s.rank)
+        _REPLICA_GROUPS[key] = (my, my_r)
+    return _REPLICA_GROUPS[key]
+
+
 # --- embedding (tied weights across first/last PP stage) ---
 def get_embedding_group():
     return _st().embedding_group

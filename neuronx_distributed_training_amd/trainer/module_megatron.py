@@ -62,6 +62,9 @@ class MegatronGPTModule(BaseModelModule):
             moe_frequency=int(moe.get("moe_frequency", 1)),
             moe_router_type=moe.get("router_type", "top_k"),
             moe_capacity_factor=moe.get("capacity_factor"),
+            token_shuffle_group_size=int(
+                self.cfg.get("distributed_strategy", {}).get(
+                    "token_shuffle_group_size", 1)),
             moe_aux_loss_coeff=float(moe.get("aux_loss_coef", 0.01)),
         )
         return GPTModel(cfg)

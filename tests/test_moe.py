@@ -126,3 +126,56 @@ def test_mixtral_trains():
 
 def test_mixtral_trains_ep2():
     run_distributed(_mixtral_train, 2, 2)
+
+
+def _token_shuffle_roundtrip(rank, world):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.modules.moe import (
+        token_shuffle, token_unshuffle,
+    )
+
+    ps.initialize_model_parallel()
+    torch.manual_seed(10 + rank)
+    x = torch.randn(16, 8, requires_grad=True)
+    y, ctx = token_shuffle(x, world)
+    assert y.shape == x.shape
+    if world > 1:
+        # shuffled content differs from local content (overwhelmingly likely)
+        assert not torch.equal(y, x)
+    back = token_unshuffle(y, ctx)
+    assert torch.allclose(back, x, atol=1e-6)
+    (back * 2).sum().backward()
+    assert torch.allclose(x.grad, torch.full_like(x, 2.0))
+    return 0.0
+
+
+@pytest.mark.parametrize("world", [1, 2])
+def test_token_shuffle_roundtrip(world):
+    run_distributed(_token_shuffle_roundtrip, world)
+
+
+def _moe_shuffled_vs_plain(rank, world):
+    """Dropless MoE is token-independent, so token shuffle must not change
+    any token's output."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.modules.moe import (
+        ExpertMLPs, MoE, RouterTopK,
+    )
+
+    ps.initialize_model_parallel()
+    torch.manual_seed(0)
+    router = RouterTopK(16, 4, 2, init_seed=1)
+    experts = ExpertMLPs(4, 16, 32, init_seed=2)
+    plain = MoE(router, experts)
+    shuf = MoE(router, experts, token_shuffle_group_size=world)
+    g = torch.Generator().manual_seed(100 + rank)
+    x = torch.randn(24, 16, generator=g)
+    y0, _ = plain(x)
+    y1, _ = shuf(x)
+    assert torch.allclose(y0, y1, atol=1e-5), (y0 - y1).abs().max()
+    return 0.0
+
+
+@pytest.mark.parametrize("world", [1, 2])
+def test_moe_token_shuffle_preserves_output(world):
+    run_distributed(_moe_shuffled_vs_plain, world)
