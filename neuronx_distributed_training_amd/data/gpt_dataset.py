@@ -135,6 +135,67 @@ class GPTDataset(torch.utils.data.Dataset):
         }
 
 
+def _blend_indices(weights, size: int):
+    """Deterministic proportional interleave: sample i of the blend comes
+    from the dataset whose next quota position (k+0.5)/w_j is earliest —
+    the merge-by-rate form of megatron's build_blending_indices."""
+    w = np.asarray(weights, dtype=np.float64)
+    w = w / w.sum()
+    parts = []
+    for j, wj in enumerate(w):
+        n = int(np.ceil(size * wj)) + 1
+        k = np.arange(n, dtype=np.float64)
+        t = (k + 0.5) / max(wj, 1e-12)
+        parts.append(np.stack([t, np.full(n, j), k], axis=1))
+    allp = np.concatenate(parts)
+    order = np.argsort(allp[:, 0], kind="stable")[:size]
+    sel = allp[order]
+    return sel[:, 1].astype(np.int64), sel[:, 2].astype(np.int64)
+
+
+class BlendableDataset(torch.utils.data.Dataset):
+    """Weighted mixture of datasets (reference NeMo BlendableDataset behind
+    the blended ``data_prefix`` list, data/megatron/data_module.py:89-130).
+    Sample i draws from dataset_index[i] at dataset_sample_index[i]; the
+    running composition tracks the normalized weights exactly."""
+
+    def __init__(self, datasets, weights, size: int):
+        assert len(datasets) == len(weights) and len(datasets) > 0
+        self.datasets = datasets
+        self.size = int(size)
+        self.dataset_index, self.dataset_sample_index = _blend_indices(
+            weights, self.size
+        )
+
+    def __len__(self):
+        return self.size
+
+    def __getitem__(self, idx):
+        j = self.dataset_index[idx]
+        ds = self.datasets[j]
+        return ds[int(self.dataset_sample_index[idx]) % len(ds)]
+
+
+def parse_data_prefix(data_prefix):
+    """NeMo data_prefix conventions → (prefixes, weights):
+    "path"                      → (["path"], [1.0])
+    ["path1", "path2"]          → equal weights
+    [w1, "path1", w2, "path2"]  → explicit weights
+    {"path1": w1, "path2": w2}  → explicit weights
+    """
+    if isinstance(data_prefix, str):
+        return [data_prefix], [1.0]
+    if isinstance(data_prefix, dict):
+        return list(data_prefix.keys()), [float(v) for v in data_prefix.values()]
+    items = list(data_prefix)
+    if items and not isinstance(items[0], str):
+        assert len(items) % 2 == 0, "expected [w1, prefix1, w2, prefix2, ...]"
+        return [str(items[i + 1]) for i in range(0, len(items), 2)], [
+            float(items[i]) for i in range(0, len(items), 2)
+        ]
+    return [str(p) for p in items], [1.0] * len(items)
+
+
 def build_train_valid_test_datasets(
     prefix: str,
     splits: str,
@@ -166,4 +227,46 @@ def build_train_valid_test_datasets(
         out.append(
             GPTDataset(indexed, docs, want, seq_length, seed, cache_dir, name)
         )
+    return tuple(out)
+
+
+def build_blended_train_valid_test_datasets(
+    data_prefix,
+    splits: str,
+    seq_length: int,
+    train_samples: int,
+    valid_samples: int,
+    test_samples: int,
+    seed: int = 1234,
+    cache_dir: Optional[str] = None,
+):
+    """Weighted multi-corpus build (reference blended data_prefix path,
+    data/megatron/data_module.py:89-130): each prefix contributes samples
+    proportional to its weight; splits apply per corpus."""
+    prefixes, weights = parse_data_prefix(data_prefix)
+    if len(prefixes) == 1:
+        return build_train_valid_test_datasets(
+            prefixes[0], splits, seq_length, train_samples, valid_samples,
+            test_samples, seed, cache_dir,
+        )
+    w = np.asarray(weights, dtype=np.float64)
+    w = w / w.sum()
+    per = []
+    for j, p in enumerate(prefixes):
+        # +margin so the blend never indexes past a member's sample count
+        per.append(
+            build_train_valid_test_datasets(
+                p, splits, seq_length,
+                int(train_samples * w[j]) + 2,
+                int(valid_samples * w[j]) + 2,
+                int(test_samples * w[j]) + 2,
+                seed + j, cache_dir,
+            )
+        )
+    out = []
+    for i, want in enumerate((train_samples, valid_samples, test_samples)):
+        members = [per[j][i] for j in range(len(prefixes)) if per[j][i] is not None]
+        mw = [weights[j] for j in range(len(prefixes)) if per[j][i] is not None]
+        out.append(BlendableDataset(members, mw, want) if members and want > 0
+                   else None)
     return tuple(out)

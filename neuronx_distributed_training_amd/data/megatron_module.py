@@ -9,7 +9,7 @@ from torch.utils.data import DataLoader
 
 from ..parallel import state as ps
 from .datamodule import BaseDataModule, default_collate
-from .gpt_dataset import build_train_valid_test_datasets
+from .gpt_dataset import build_blended_train_valid_test_datasets
 from .samplers import (
     MegatronPretrainingBatchSampler,
     MegatronPretrainingRandomBatchSampler,
@@ -23,8 +23,8 @@ class MegatronDataModule(BaseDataModule):
         eval_iters = int(self.cfg.get("trainer", {}).get("limit_val_batches", 8))
         train_samples = max_steps * self.global_batch_size
         valid_samples = max(eval_iters * self.global_batch_size, self.global_batch_size)
-        self.train_ds, self.val_ds, self.test_ds = build_train_valid_test_datasets(
-            prefix=d["data_prefix"],
+        self.train_ds, self.val_ds, self.test_ds = build_blended_train_valid_test_datasets(
+            data_prefix=d["data_prefix"],
             splits=d.get("splits_string", "969,30,1"),
             seq_length=self.seq_length,
             train_samples=train_samples,

@@ -207,3 +207,40 @@ def test_chunked_cross_attention():
     r1 = run_distributed(_cca, 1)
     r2 = run_distributed(_cca, 2)
     assert torch.allclose(r1[0], r2[0], atol=1e-3), (r1[0], r2[0])
+
+
+def test_blendable_dataset(tmp_path):
+    from neuronx_distributed_training_amd.data.indexed_dataset import (
+        MMapIndexedDatasetBuilder,
+    )
+    from neuronx_distributed_training_amd.data.gpt_dataset import (
+        BlendableDataset, build_blended_train_valid_test_datasets,
+        parse_data_prefix, _blend_indices,
+    )
+
+    # prefix conventions
+    assert parse_data_prefix("a") == (["a"], [1.0])
+    assert parse_data_prefix(["a", "b"]) == (["a", "b"], [1.0, 1.0])
+    assert parse_data_prefix([0.7, "a", 0.3, "b"]) == (["a", "b"], [0.7, 0.3])
+    assert parse_data_prefix({"a": 2, "b": 1}) == (["a", "b"], [2.0, 1.0])
+
+    # interleave tracks weights exactly
+    di, si = _blend_indices([0.75, 0.25], 400)
+    assert (di == 0).sum() == 300 and (di == 1).sum() == 100
+    # per-dataset sample indices are sequential
+    assert list(si[di == 1][:3]) == [0, 1, 2]
+
+    # two tiny corpora with distinguishable tokens
+    for name, tok in (("ca", 11), ("cb", 77)):
+        b = MMapIndexedDatasetBuilder(os.path.join(str(tmp_path), name))
+        for _ in range(40):
+            b.add_document([tok] * 64)
+        b.finalize()
+    tr, va, te = build_blended_train_valid_test_datasets(
+        [0.7, os.path.join(str(tmp_path), "ca"), 0.3, os.path.join(str(tmp_path), "cb")],
+        splits="100,0,0", seq_length=16, train_samples=40,
+        valid_samples=0, test_samples=0, cache_dir=str(tmp_path),
+    )
+    assert isinstance(tr, BlendableDataset) and len(tr) == 40 and va is None
+    toks = [int(tr[i]["input_ids"][0]) for i in range(40)]
+    assert toks.count(11) == 28 and toks.count(77) == 12
