@@ -26,6 +26,7 @@ import torch.nn.functional as F
 from torch.utils.checkpoint import checkpoint as _ckpt
 
 from ..parallel import state as ps
+from ..parallel.random import get_rng_tracker
 from ..parallel.layers import ColumnParallelLinear, ParallelEmbedding, RowParallelLinear
 from ..parallel.loss import parallel_cross_entropy
 from ..parallel.mappings import (
@@ -53,7 +54,7 @@ class GPTConfig:
     activation: str = "swiglu"  # swiglu | geglu | gelu
     normalization: str = "rmsnorm"  # rmsnorm | layernorm
     layernorm_epsilon: float = 1e-5
-    transformer_block_type: str = "pre_ln"  # pre_ln | post_ln
+    transformer_block_type: str = "pre_ln"  # pre_ln | post_ln | normformer
     hidden_dropout: float = 0.0
     attention_dropout: float = 0.0
     share_embeddings_and_output_weights: bool = True
@@ -254,27 +255,41 @@ class ParallelTransformerLayer(nn.Module):
         self.mlp = (
             NeuronSwitchMLP(cfg, layer_idx) if self.is_moe else ParallelMLP(cfg, layer_idx)
         )
+        if self.block_type == "normformer":
+            self.post_inner_layernorm = make_norm(cfg)
         self.dropout = nn.Dropout(cfg.hidden_dropout)
 
+    def _drop(self, t):
+        # SP shards the residual stream over TP ranks: dropout there must
+        # draw per-rank randomness (reference XLA RNG-tracker fork).
+        if self.cfg.sequence_parallel:
+            with get_rng_tracker().fork():
+                return self.dropout(t)
+        return self.dropout(t)
+
+    def _mlp(self, h):
+        if self.is_moe:
+            return self.mlp(h)
+        return self.mlp(h), None
+
     def forward(self, x, cos, sin, pos_offset=0):
-        logits = None
         if self.block_type == "post_ln":
             a = self.self_attention(x, cos, sin, pos_offset)
-            x = self.input_layernorm(x + self.dropout(a))
-            if self.is_moe:
-                m, logits = self.mlp(x)
-            else:
-                m = self.mlp(x)
-            x = self.post_attention_layernorm(x + self.dropout(m))
+            x = self.input_layernorm(x + self._drop(a))
+            m, logits = self._mlp(x)
+            x = self.post_attention_layernorm(x + self._drop(m))
+        elif self.block_type == "normformer":
+            # extra norm on the attention output before the residual add
+            # (reference transformer.py normformer block type)
+            a = self.self_attention(self.input_layernorm(x), cos, sin, pos_offset)
+            x = x + self._drop(self.post_inner_layernorm(a))
+            m, logits = self._mlp(self.post_attention_layernorm(x))
+            x = x + self._drop(m)
         else:  # pre_ln
             a = self.self_attention(self.input_layernorm(x), cos, sin, pos_offset)
-            x = x + self.dropout(a)
-            h = self.post_attention_layernorm(x)
-            if self.is_moe:
-                m, logits = self.mlp(h)
-            else:
-                m = self.mlp(h)
-            x = x + self.dropout(m)
+            x = x + self._drop(a)
+            m, logits = self._mlp(self.post_attention_layernorm(x))
+            x = x + self._drop(m)
         return x, logits
 
 
@@ -307,6 +322,10 @@ class Embedding(nn.Module):
         x = x.transpose(0, 1).contiguous()  # [s, b, h]
         if self.cfg.sequence_parallel:
             x = scatter_to_sequence_parallel_region(x)
+            # dropout on the sequence-sharded embedding draws per-TP-rank
+            # randomness (reference language_model.py:320 RNG-tracked)
+            with get_rng_tracker().fork():
+                return self.dropout(x)
         return self.dropout(x)
 
 

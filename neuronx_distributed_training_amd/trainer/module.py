@@ -50,6 +50,11 @@ class BaseModelModule:
         raise NotImplementedError
 
     def setup(self):
+        seed = self.cfg.get("seed")
+        if seed is not None:
+            from ..parallel.random import model_parallel_manual_seed
+
+            model_parallel_manual_seed(int(seed))
         self.model = self.build_model().to(self.device)
         self.model.train()
         self.pp_engine = None

@@ -244,3 +244,57 @@ def test_blendable_dataset(tmp_path):
     assert isinstance(tr, BlendableDataset) and len(tr) == 40 and va is None
     toks = [int(tr[i]["input_ids"][0]) for i in range(40)]
     assert toks.count(11) == 28 and toks.count(77) == 12
+
+
+def test_rng_tracker():
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.parallel.random import (
+        RNGStatesTracker, model_parallel_manual_seed,
+    )
+
+    ps.destroy_model_parallel()
+    tr = RNGStatesTracker()
+    tr.add("model-parallel-rng", 123)
+    torch.manual_seed(7)
+    base1 = torch.rand(4)
+    with tr.fork("model-parallel-rng"):
+        forked1 = torch.rand(4)
+    base2 = torch.rand(4)  # outside stream unaffected by the fork
+    torch.manual_seed(7)
+    base1b = torch.rand(4)
+    base2b = torch.rand(4)
+    assert torch.equal(base1, base1b) and torch.equal(base2, base2b)
+    with tr.fork("model-parallel-rng"):
+        forked2 = torch.rand(4)  # fork state advances across forks
+    assert not torch.equal(forked1, forked2)
+    # same seed → same fork stream
+    tr2 = RNGStatesTracker()
+    tr2.add("model-parallel-rng", 123)
+    with tr2.fork("model-parallel-rng"):
+        assert torch.equal(torch.rand(4), forked1)
+    # unknown name = no-op fork
+    with RNGStatesTracker().fork("nope"):
+        pass
+    s = model_parallel_manual_seed(42)
+    assert s == 42  # pp_rank 0
+
+
+def test_normformer_block():
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.megatron_gpt import (
+        GPTConfig, GPTModel,
+    )
+
+    ps.destroy_model_parallel()
+    torch.manual_seed(0)
+    cfg = GPTConfig(
+        vocab_size=64, hidden_size=32, ffn_hidden_size=64, num_layers=2,
+        num_attention_heads=4, max_position_embeddings=32,
+        transformer_block_type="normformer",
+    )
+    model = GPTModel(cfg)
+    assert hasattr(model.layers[0], "post_inner_layernorm")
+    ids = torch.randint(0, 64, (2, 16))
+    loss = model(ids, labels=ids.clone())
+    assert torch.isfinite(loss)
+    loss.backward()
