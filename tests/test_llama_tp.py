@@ -117,3 +117,54 @@ def test_fused_qkv_tp2_matches_tp1():
     l1 = run_distributed(_fused_qkv_loss, 1)[0]
     l2 = run_distributed(_fused_qkv_loss, 2)
     assert abs(l1 - l2[0]) < 5e-3, (l1, l2[0])
+
+
+def _bench_shape_step(rank, world):
+    """Mirror bench.py's training path (LlamaModule, TP=world, SP on,
+    ZeRO-1, 8B-like head layout: heads=TP, one KV head per rank) at tiny
+    dims — de-risks the driver's N=4/8 scaling runs."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 1, "seq_length": 64},
+        "distributed_strategy": {
+            "tensor_model_parallel_size": world,
+            "sequence_parallel": world > 1,
+            "zero1": True,
+        },
+        "model": {
+            "vocab_size": 512, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 2, "num_attention_heads": 8, "num_kv_heads": 8,
+            "rope_theta": 500000.0, "grad_clip": 1.0,
+            "optim": {"lr": 3e-4, "sched": {"warmup_steps": 2}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {"log_gradient_norm": True},
+    }
+    torch.manual_seed(1234)
+    mod = LlamaModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=4)
+    g = torch.Generator().manual_seed(5)
+    losses = []
+    for _ in range(2):
+        micros = [
+            {"input_ids": (ids := torch.randint(0, 512, (1, 64), generator=g)),
+             "labels": ids.clone()}
+            for _ in range(mod.num_microbatches)
+        ]
+        mod.optimizer.zero_grad()
+        m = mod.training_step(micros)
+        losses.append(m["reduced_train_loss"])
+    return losses[-1]
+
+
+@pytest.mark.parametrize("world", [4, 8])
+def test_bench_path_tp_wide(world):
+    """TP=4 and TP=8 (the driver's scaling widths) match single-rank loss."""
+    l1 = run_distributed(_bench_shape_step, 1)[0]
+    lw = run_distributed(_bench_shape_step, world)
+    assert max(abs(l - lw[0]) for l in lw) < 1e-5   # ranks agree
+    assert abs(l1 - lw[0]) < 5e-3, (l1, lw[0])
