@@ -175,3 +175,58 @@ def test_pp2_tp2_sp_matches_reference():
     ref = run_distributed(_ref_loss, 1, False)[0]
     res = run_distributed(_pp_sp_loss, 4)
     assert abs(res[0] - ref) < 5e-3, (res[0], ref)
+
+
+def _full_3d(rank, world):
+    """TP=2 × PP=2 × DP=2 on 8 ranks: full trainer-module training step
+    (ZeRO-1 over DP, 1F1B PP, SP on) vs the single-rank global batch."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+
+    tp = 2 if world == 8 else 1
+    pp = 2 if world == 8 else 1
+    ps.initialize_model_parallel(
+        tensor_model_parallel_size=tp, pipeline_model_parallel_size=pp
+    )
+    dp = ps.get_data_parallel_world_size()
+    cfg = {
+        "data": {"global_batch_size": 4, "micro_batch_size": 1, "seq_length": 32},
+        "distributed_strategy": {
+            "tensor_model_parallel_size": tp,
+            "pipeline_model_parallel_size": pp,
+            "sequence_parallel": tp > 1,
+            "zero1": True,
+        },
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 4, "num_attention_heads": 4, "num_kv_heads": 2,
+            "grad_clip": 1.0, "optim": {"lr": 1e-3, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(7)
+    mod = LlamaModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=4)
+    g = torch.Generator().manual_seed(5)
+    last = None
+    for _ in range(2):
+        glob = [
+            {"input_ids": (ids := torch.randint(0, 128, (1, 32), generator=g)),
+             "labels": ids.clone()}
+            for _ in range(4)
+        ]
+        per = 4 // dp
+        r = ps.get_data_parallel_rank()
+        micros = glob[r * per : (r + 1) * per]
+        m = mod.training_step(micros)
+        last = m["reduced_train_loss"]
+    return last
+
+
+def test_tp2_pp2_dp2_matches_single():
+    ref = run_distributed(_full_3d, 1)[0]
+    res = run_distributed(_full_3d, 8)
+    assert max(abs(x - res[0]) for x in res) < 1e-5, res
+    assert abs(ref - res[0]) < 5e-3, (ref, res[0])
