@@ -179,3 +179,42 @@ def _moe_shuffled_vs_plain(rank, world):
 @pytest.mark.parametrize("world", [1, 2])
 def test_moe_token_shuffle_preserves_output(world):
     run_distributed(_moe_shuffled_vs_plain, world)
+
+
+def _mixtral_tp_ep(rank, world):
+    """TP=2 × EP=2 on 4 ranks: attention TP-sharded, experts EP-sharded;
+    loss matches the single-rank model."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.mixtral import (
+        MixtralConfig, MixtralForCausalLM,
+    )
+
+    tp = 2 if world == 4 else 1
+    ep = 2 if world == 4 else 1
+    ps.initialize_model_parallel(
+        tensor_model_parallel_size=tp, expert_model_parallel_size=ep
+    )
+    torch.manual_seed(2)
+    cfg = MixtralConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=32, num_local_experts=4,
+        num_experts_per_tok=2,
+    )
+    model = MixtralForCausalLM(cfg)
+    g = torch.Generator().manual_seed(9)
+    ids = torch.randint(0, 128, (2, 32), generator=g)
+    loss = model(ids, labels=ids)
+    loss.backward()
+    gsum = sum(
+        float(p.grad.abs().sum()) for p in model.parameters() if p.grad is not None
+    )
+    assert gsum > 0
+    return float(loss)
+
+
+def test_mixtral_tp2_ep2_matches_single():
+    ref = run_distributed(_mixtral_tp_ep, 1)[0]
+    res = run_distributed(_mixtral_tp_ep, 4)
+    assert max(abs(x - res[0]) for x in res) < 1e-5, res
+    assert abs(ref - res[0]) < 5e-3, (ref, res[0])
