@@ -169,3 +169,44 @@ def _fit_async_ckpt(rank, world, tmpdir):
 
 def test_async_checkpointing(tmp_path):
     assert run_distributed(_fit_async_ckpt, 1, str(tmp_path))[0] >= 1
+
+
+def _fit_3d(rank, world, tmpdir, max_steps, resume):
+    """Trainer fit + checkpoint resume under TP2×PP2 (4 ranks): sharded
+    save/load round-trips every (tp, pp) shard and the ZeRO state."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.trainer import Trainer
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+    from neuronx_distributed_training_amd.trainer.checkpoint import (
+        find_latest_checkpoint,
+    )
+    from neuronx_distributed_training_amd.data.datamodule import build_datamodule
+
+    ps.initialize_model_parallel(
+        tensor_model_parallel_size=2, pipeline_model_parallel_size=2
+    )
+    torch.manual_seed(0)
+    cfg = _cfg(tmpdir, max_steps)
+    cfg["distributed_strategy"] = {
+        "tensor_model_parallel_size": 2,
+        "pipeline_model_parallel_size": 2,
+        "sequence_parallel": True,
+    }
+    cfg["model"]["num_layers"] = 4
+    tr = Trainer(cfg)
+    tr.ckpt_dir = os.path.join(tmpdir, "checkpoints")
+    os.makedirs(tr.ckpt_dir, exist_ok=True)
+    module = LlamaModule(cfg)
+    dm = build_datamodule(cfg)
+    ckpt = find_latest_checkpoint(tr.ckpt_dir) if resume else None
+    tr.fit(module, dm, ckpt_path=ckpt)
+    w = next(module.model.parameters()).detach()
+    return tr.global_step, float(w.abs().sum())
+
+
+def test_fit_and_resume_tp2_pp2(tmp_path):
+    d = str(tmp_path)
+    res = run_distributed(_fit_3d, 4, d, 2, False)
+    assert all(s == 2 for s, _ in res)
+    res2 = run_distributed(_fit_3d, 4, d, 4, True)
+    assert all(s == 4 for s, _ in res2)
