@@ -67,7 +67,12 @@ class GPTConfig:
     moe_top_k: int = 2
     moe_frequency: int = 1
     moe_router_type: str = "top_k"
+    moe_sinkhorn_iterations: int = 3
     moe_capacity_factor: Optional[float] = None
+    moe_router_activation: str = "softmax"   # softmax | sigmoid
+    moe_sinkhorn_tol: Optional[float] = None
+    normalize_top_k_affinities: bool = True
+    moe_dropout: float = 0.0
     token_shuffle_group_size: int = 1
     moe_aux_loss_coeff: float = 0.01
 
@@ -140,15 +145,29 @@ class NeuronSwitchMLP(nn.Module):
 
     def __init__(self, cfg: GPTConfig, layer_idx: int):
         super().__init__()
-        router_cls = RouterSinkhorn if cfg.moe_router_type == "sinkhorn" else RouterTopK
         self.sequence_parallel = cfg.sequence_parallel
+        rkw = dict(
+            dtype=cfg.torch_dtype, init_seed=4000 + layer_idx,
+            act_fn=cfg.moe_router_activation,
+            normalize_top_k_affinities=cfg.normalize_top_k_affinities,
+        )
+        if cfg.moe_router_type == "sinkhorn":
+            router = RouterSinkhorn(
+                cfg.hidden_size, cfg.num_moe_experts, cfg.moe_top_k,
+                n_iter=cfg.moe_sinkhorn_iterations, tol=cfg.moe_sinkhorn_tol,
+                **rkw,
+            )
+        else:
+            router = RouterTopK(
+                cfg.hidden_size, cfg.num_moe_experts, cfg.moe_top_k, **rkw
+            )
         self.moe = MoE(
-            router_cls(cfg.hidden_size, cfg.num_moe_experts, cfg.moe_top_k,
-                       dtype=cfg.torch_dtype, init_seed=4000 + layer_idx),
+            router,
             ExpertMLPs(cfg.num_moe_experts, cfg.hidden_size, cfg.ffn_size,
                        dtype=cfg.torch_dtype, init_seed=4001 + layer_idx),
             capacity_factor=cfg.moe_capacity_factor,
             token_shuffle_group_size=cfg.token_shuffle_group_size,
+            moe_dropout=cfg.moe_dropout,
         )
 
     def forward(self, x):

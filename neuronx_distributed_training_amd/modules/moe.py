@@ -123,11 +123,19 @@ def load_balancing_loss_func(router_logits: torch.Tensor, num_experts: int,
 
 
 class RouterTopK(nn.Module):
+    """Top-k router (reference RouterTopK contract incl. the
+    ``act_fn`` softmax/sigmoid and ``normalize_top_k_affinities`` knobs,
+    transformer.py:402-431)."""
+
     def __init__(self, hidden_size: int, num_experts: int, top_k: int,
-                 dtype: torch.dtype = torch.float32, init_seed: Optional[int] = None):
+                 dtype: torch.dtype = torch.float32, init_seed: Optional[int] = None,
+                 act_fn: str = "softmax", normalize_top_k_affinities: bool = True):
         super().__init__()
         self.num_experts = num_experts
         self.top_k = top_k
+        assert act_fn in ("softmax", "sigmoid"), act_fn
+        self.act_fn = act_fn
+        self.normalize_top_k_affinities = normalize_top_k_affinities
         if init_seed is not None:
             st = torch.random.get_rng_state()
             torch.manual_seed(init_seed)
@@ -140,18 +148,23 @@ class RouterTopK(nn.Module):
     def forward(self, x: torch.Tensor):
         """x: [T, H] → (weights [T, k], indices [T, k], logits [T, E])."""
         logits = F.linear(x, self.weight).float()
-        probs = torch.softmax(logits, dim=-1)
+        if self.act_fn == "sigmoid":
+            probs = torch.sigmoid(logits)
+        else:
+            probs = torch.softmax(logits, dim=-1)
         topw, topi = probs.topk(self.top_k, dim=-1)
-        topw = topw / topw.sum(dim=-1, keepdim=True)  # renormalize
+        if self.normalize_top_k_affinities:
+            topw = topw / topw.sum(dim=-1, keepdim=True)
         return topw.to(x.dtype), topi, logits
 
 
 class RouterSinkhorn(RouterTopK):
     """Sinkhorn-balanced routing (reference RouterSinkhorn contract)."""
 
-    def __init__(self, *args, n_iter: int = 3, **kw):
+    def __init__(self, *args, n_iter: int = 3, tol: Optional[float] = None, **kw):
         super().__init__(*args, **kw)
         self.n_iter = n_iter
+        self.tol = tol  # early-exit threshold (reference moe_sinkhorn_tol)
 
     def forward(self, x: torch.Tensor):
         logits = F.linear(x, self.weight).float()
@@ -161,8 +174,13 @@ class RouterSinkhorn(RouterTopK):
             d1 = torch.ones(cost.size(1), device=cost.device)
             eps = 1e-8
             for _ in range(self.n_iter):
+                d1_old = d1
                 d0 = 1.0 / (cost @ d1 + eps)
                 d1 = cost.size(0) / cost.size(1) / (cost.t() @ d0 + eps)
+                if self.tol is not None and float(
+                    (d1 - d1_old).abs().max()
+                ) < self.tol:
+                    break
             balanced = cost * d0.unsqueeze(1) * d1.unsqueeze(0)
             _, topi = balanced.topk(self.top_k, dim=-1)
         # gradients flow through softmax of raw logits at chosen experts
@@ -237,12 +255,14 @@ class MoE(nn.Module):
 
     def __init__(self, router: RouterTopK, experts: ExpertMLPs,
                  capacity_factor: Optional[float] = None,
-                 token_shuffle_group_size: int = 1):
+                 token_shuffle_group_size: int = 1,
+                 moe_dropout: float = 0.0):
         super().__init__()
         self.router = router
         self.experts = experts
         self.capacity_factor = capacity_factor
         self.token_shuffle_group_size = token_shuffle_group_size
+        self.dropout = nn.Dropout(moe_dropout) if moe_dropout else None
         self.ep = ps.get_expert_model_parallel_world_size()
 
     def forward(self, x: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
@@ -326,6 +346,8 @@ class MoE(nn.Module):
         # combine: scatter-add weighted expert outputs back to tokens
         y = torch.zeros_like(x)
         y.index_add_(0, sorted_tok, yexp * sorted_w.unsqueeze(-1))
+        if self.dropout is not None:
+            y = self.dropout(y)
         if shuf_ctx is not None:
             y = token_unshuffle(y, shuf_ctx)
         return y, logits

@@ -62,6 +62,12 @@ class MegatronGPTModule(BaseModelModule):
             moe_frequency=int(moe.get("moe_frequency", 1)),
             moe_router_type=moe.get("router_type", "top_k"),
             moe_capacity_factor=moe.get("capacity_factor"),
+            moe_router_activation=moe.get("router_activation", "softmax"),
+            moe_sinkhorn_iterations=int(moe.get("sinkhorn_iterations", 3)),
+            moe_sinkhorn_tol=moe.get("sinkhorn_tol"),
+            normalize_top_k_affinities=bool(
+                moe.get("normalize_top_k_affinities", True)),
+            moe_dropout=float(moe.get("moe_dropout", 0.0)),
             token_shuffle_group_size=int(
                 self.cfg.get("distributed_strategy", {}).get(
                     "token_shuffle_group_size", 1)),

@@ -218,3 +218,35 @@ def test_mixtral_tp2_ep2_matches_single():
     res = run_distributed(_mixtral_tp_ep, 4)
     assert max(abs(x - res[0]) for x in res) < 1e-5, res
     assert abs(ref - res[0]) < 5e-3, (ref, res[0])
+
+
+def _router_variants(rank, world):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.modules.moe import (
+        RouterSinkhorn, RouterTopK,
+    )
+
+    ps.initialize_model_parallel()
+    torch.manual_seed(0)
+    x = torch.randn(32, 16)
+    # sigmoid activation, unnormalized affinities
+    r = RouterTopK(16, 4, 2, init_seed=1, act_fn="sigmoid",
+                   normalize_top_k_affinities=False)
+    w, i, logits = r(x)
+    assert w.shape == (32, 2) and (w <= 1.0).all() and (w >= 0.0).all()
+    # without renorm, pair sums need not be 1
+    assert not torch.allclose(w.sum(-1), torch.ones(32))
+    rn = RouterTopK(16, 4, 2, init_seed=1)
+    wn, _, _ = rn(x)
+    assert torch.allclose(wn.sum(-1), torch.ones(32), atol=1e-5)
+    # sinkhorn early-exit tolerance: loose tol gives same top-1 choices
+    rs = RouterSinkhorn(16, 4, 1, init_seed=2, n_iter=50, tol=1e-7)
+    _, i1, _ = rs(x)
+    rs2 = RouterSinkhorn(16, 4, 1, init_seed=2, n_iter=50)
+    _, i2, _ = rs2(x)
+    assert (i1 == i2).float().mean() > 0.9
+    return 0.0
+
+
+def test_router_variants():
+    run_distributed(_router_variants, 1)
