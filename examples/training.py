@@ -78,7 +78,10 @@ def train(cfg):
     datamodule = build_datamodule(cfg)
 
     ckpt_path = None
-    if cfg.get("exp_manager", {}).get("resume_if_exists") and trainer_obj.ckpt_dir:
+    manual = cfg.get("exp_manager", {}).get("resume_from_checkpoint")
+    if manual:
+        ckpt_path = manual
+    elif cfg.get("exp_manager", {}).get("resume_if_exists") and trainer_obj.ckpt_dir:
         ckpt_path = find_latest_checkpoint(trainer_obj.ckpt_dir)
         if ckpt_path:
             print(f"resuming from {ckpt_path}")

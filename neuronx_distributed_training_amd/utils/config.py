@@ -23,9 +23,27 @@ def load_config(path: str, overrides: Optional[List[str]] = None) -> Dict:
             raise ValueError(f"override must be key.path=value: {ov}")
         key, val = ov.split("=", 1)
         _set_dotted(cfg, key, _parse_scalar(val))
+    _normalize_reference_keys(cfg)
     validate_config(cfg)
     derive_env(cfg)
     return cfg
+
+
+def _normalize_reference_keys(cfg: Dict):
+    """Accept reference-style YAML keys as aliases so configs migrate
+    without edits (reference examples/conf/hf_llama3_8B_config.yaml):
+    trainer.gradient_clip_val → model.grad_clip,
+    model.activations_checkpoint_granularity → model.activation_checkpoint,
+    model.encoder_seq_length → data.seq_length."""
+    tr = cfg.get("trainer", {}) or {}
+    m = cfg.setdefault("model", {})
+    d = cfg.setdefault("data", {})
+    if "gradient_clip_val" in tr and "grad_clip" not in m:
+        m["grad_clip"] = tr["gradient_clip_val"]
+    if "activations_checkpoint_granularity" in m and "activation_checkpoint" not in m:
+        m["activation_checkpoint"] = m["activations_checkpoint_granularity"]
+    if "encoder_seq_length" in m and "seq_length" not in d:
+        d["seq_length"] = m["encoder_seq_length"]
 
 
 def _parse_scalar(v: str) -> Any:
