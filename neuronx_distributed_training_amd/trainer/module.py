@@ -9,6 +9,7 @@ XLA machinery (mark_step, step closures) is gone — eager HIP streams.
 
 from __future__ import annotations
 
+import contextlib
 import time
 from typing import Dict, Optional
 
@@ -44,6 +45,19 @@ class BaseModelModule:
         self.throughput = Throughput(window=10)
         self.log_param_norm = bool(cfg.get("exp_manager", {}).get("log_parameter_norm", False))
         self.log_grad_norm = bool(cfg.get("exp_manager", {}).get("log_gradient_norm", True))
+        # reference precision mode "autocast": fp32 weights, bf16 autocast
+        # region around forward/loss (vs "mixed_precision" = bf16 weights +
+        # fp32 ZeRO masters)
+        self.autocast_dtype = (
+            torch.bfloat16
+            if str(cfg.get("precision", {}).get("type", "")) == "autocast"
+            else None
+        )
+
+    def _autocast(self):
+        if self.autocast_dtype is None:
+            return contextlib.nullcontext()
+        return torch.autocast(self.device.type, dtype=self.autocast_dtype)
 
     # -- to override --
     def build_model(self) -> torch.nn.Module:
@@ -143,7 +157,8 @@ class BaseModelModule:
             # microbatch runs (reference data/base.py:58-64 semantics)
             mbs = [self.get_batch_on_this_context_parallel_rank(
                        {k: v for k, v in b.items()}) for b in microbatches]
-            loss = self.pp_engine.run_train(mbs).float()
+            with self._autocast():
+                loss = self.pp_engine.run_train(mbs).float()
             # loss lives on the last stage only; SUM over PP broadcasts it
             dist.all_reduce(loss, group=ps.get_pipeline_model_parallel_group())
             running = loss
@@ -163,7 +178,8 @@ class BaseModelModule:
                 for k, v in batch.items()
             }
             batch = self.get_batch_on_this_context_parallel_rank(batch)
-            loss = self.model_fwd_calc_loss(batch)
+            with self._autocast():
+                loss = self.model_fwd_calc_loss(batch)
             if bi == len(batches) - 1 and hasattr(self.optimizer, "enable_grad_sync"):
                 # final microbatch: overlap the DP grad reduce with backward
                 self.optimizer.enable_grad_sync()

@@ -210,3 +210,35 @@ def test_fit_and_resume_tp2_pp2(tmp_path):
     assert all(s == 2 for s, _ in res)
     res2 = run_distributed(_fit_3d, 4, d, 4, True)
     assert all(s == 4 for s, _ in res2)
+
+
+def _autocast_step(rank, world):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+
+    ps.initialize_model_parallel()
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 2, "seq_length": 32},
+        "distributed_strategy": {},
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 2, "num_attention_heads": 4, "num_kv_heads": 2,
+            "grad_clip": 1.0, "optim": {"lr": 1e-3, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "autocast"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(3)
+    mod = LlamaModule(cfg)
+    mod.setup()
+    # weights stay fp32 under autocast mode
+    assert next(mod.model.parameters()).dtype == torch.float32
+    mod.configure_optimizers(max_steps=4)
+    ids = torch.randint(0, 128, (2, 32), generator=torch.Generator().manual_seed(5))
+    m = mod.training_step([{"input_ids": ids, "labels": ids.clone()}])
+    assert m["reduced_train_loss"] == m["reduced_train_loss"]
+    return m["reduced_train_loss"]
+
+
+def test_autocast_precision_mode():
+    run_distributed(_autocast_step, 1)
