@@ -15,6 +15,7 @@ def generate(
     eos_token_id: int | None = None,
     temperature: float = 0.0,
     top_k: int = 0,
+    top_p: float = 0.0,
 ) -> torch.Tensor:
     """input_ids: [b, s]. Returns [b, s + new]. Full-recompute decode (no
     KV cache) — intended for offline evaluation, not serving."""
@@ -28,6 +29,14 @@ def generate(
             if top_k:
                 v, _ = torch.topk(nxt, top_k)
                 nxt[nxt < v[:, [-1]]] = float("-inf")
+            if top_p and top_p < 1.0:
+                sorted_logits, sorted_idx = torch.sort(nxt, descending=True)
+                cum = torch.softmax(sorted_logits, dim=-1).cumsum(-1)
+                cut = cum > top_p
+                cut[:, 1:] = cut[:, :-1].clone()  # keep first token over p
+                cut[:, 0] = False
+                mask = torch.zeros_like(cut).scatter_(1, sorted_idx, cut)
+                nxt = nxt.masked_fill(mask, float("-inf"))
             probs = torch.softmax(nxt, dim=-1)
             tok = torch.multinomial(probs, 1)
         else:

@@ -103,3 +103,29 @@ def test_pad_vocab_size():
     assert pad_vocab_size(32000, 8, 8) == 32000
     assert pad_vocab_size(32003, 8, 8) == 32064        # padded up
     assert pad_vocab_size(50257, 128, 1) == 50304      # classic GPT-2 pad
+
+
+def test_generation_sampling_options():
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.llama import (
+        LlamaConfig, LlamaForCausalLM,
+    )
+    from neuronx_distributed_training_amd.utils.generation import generate
+
+    ps.destroy_model_parallel()
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(
+        LlamaConfig(vocab_size=64, hidden_size=32, intermediate_size=64,
+                    num_hidden_layers=1, num_attention_heads=2,
+                    num_key_value_heads=1, max_position_embeddings=64)
+    )
+    ids = torch.randint(0, 64, (2, 8))
+    torch.manual_seed(1)
+    out = generate(model, ids, max_new_tokens=4, temperature=0.8,
+                   top_k=10, top_p=0.9)
+    assert out.shape == (2, 12)
+    # top_p=tiny → only the argmax survives → equals greedy
+    torch.manual_seed(1)
+    out_p = generate(model, ids, max_new_tokens=4, temperature=1.0, top_p=1e-9)
+    out_g = generate(model, ids, max_new_tokens=4)
+    assert torch.equal(out_p, out_g)
