@@ -216,7 +216,8 @@ class LlamaAttention(nn.Module):
             window=getattr(self.cfg, "sliding_window", None),
         )
 
-    def forward(self, x, cos, sin, pos_offset: int = 0):
+    def forward(self, x, cos, sin, pos_offset: int = 0, cache=None,
+                layer_idx: int = 0):
         # x: [s(, /tp if SP), b, h]
         s_dim, b = x.size(0), x.size(1)
         d = self.head_dim
@@ -227,12 +228,43 @@ class LlamaAttention(nn.Module):
         v = v.view(s_full, b, self.n_kv_local, d).permute(1, 2, 0, 3)
         q = apply_rotary_pos_emb(q, cos, sin, pos_offset)
         k = apply_rotary_pos_emb(k, cos, sin, pos_offset)
-        if self.cfg.activation_checkpoint == "selective" and self.training:
+        if cache is not None:
+            k, v = cache.update(layer_idx, k, v)
+            if q.size(2) < k.size(2):
+                # incremental decode: new queries attend to the whole cache
+                rep = self.n_heads_local // self.n_kv_local
+                o = torch.nn.functional.scaled_dot_product_attention(
+                    q, k.repeat_interleave(rep, 1),
+                    v.repeat_interleave(rep, 1),
+                    is_causal=False, scale=self.scale,
+                )
+            else:
+                o = self.core_attention(q, k, v)  # prefill
+        elif self.cfg.activation_checkpoint == "selective" and self.training:
             o = _ckpt(self.core_attention, q, k, v, use_reentrant=False)
         else:
             o = self.core_attention(q, k, v)
         o = o.permute(2, 0, 1, 3).reshape(s_full, b, self.n_heads_local * d)
         return self.o_proj(o)
+
+
+class KVCache:
+    """Per-layer K/V cache for incremental decode ([b, h, s, d], dim 2).
+    Training never uses this — it serves the eval/generation path
+    (utils/generation.py). Requires SP off and CP == 1."""
+
+    def __init__(self, num_layers: int):
+        self.k = [None] * num_layers
+        self.v = [None] * num_layers
+        self.seq_len = 0
+
+    def update(self, i: int, k: torch.Tensor, v: torch.Tensor):
+        if self.k[i] is None:
+            self.k[i], self.v[i] = k, v
+        else:
+            self.k[i] = torch.cat([self.k[i], k], dim=2)
+            self.v[i] = torch.cat([self.v[i], v], dim=2)
+        return self.k[i], self.v[i]
 
 
 class LlamaDecoderLayer(nn.Module):
@@ -244,8 +276,10 @@ class LlamaDecoderLayer(nn.Module):
         self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, dtype=dt)
         self.mlp = LlamaMLP(cfg, layer_idx)
 
-    def forward(self, x, cos, sin, pos_offset: int = 0):
-        x = x + self.self_attn(self.input_layernorm(x), cos, sin, pos_offset)
+    def forward(self, x, cos, sin, pos_offset: int = 0, cache=None,
+                layer_idx: int = 0):
+        x = x + self.self_attn(self.input_layernorm(x), cos, sin, pos_offset,
+                               cache=cache, layer_idx=layer_idx)
         x = x + self.mlp(self.post_attention_layernorm(x))
         return x
 
@@ -271,21 +305,29 @@ class LlamaModel(nn.Module):
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
 
-    def forward(self, input_ids):
-        # input_ids: [b, s(, /cp)] — CP split done by the trainer.
-        cp_rank = ps.get_context_model_parallel_rank()
-        pos_offset = cp_rank * input_ids.size(1)
+    def forward(self, input_ids, kv_cache=None):
+        # input_ids: [b, s(, /cp)] — CP split done by the trainer. With
+        # kv_cache, input_ids are the NEW tokens only (decode path).
+        if kv_cache is not None:
+            assert not self.cfg.sequence_parallel, "kv cache requires SP off"
+            pos_offset = kv_cache.seq_len
+        else:
+            cp_rank = ps.get_context_model_parallel_rank()
+            pos_offset = cp_rank * input_ids.size(1)
         x = self.embed_tokens(input_ids)  # [b, s, h]
         x = x.transpose(0, 1).contiguous()  # [s, b, h]
         if self.cfg.sequence_parallel:
             x = scatter_to_sequence_parallel_region(x)
         full_ckpt = self.cfg.activation_checkpoint == "full" and self.training
-        for layer in self.layers:
+        for li, layer in enumerate(self.layers):
             if full_ckpt:
                 x = _ckpt(layer, x, self.rope_cos, self.rope_sin, pos_offset,
                           use_reentrant=False)
             else:
-                x = layer(x, self.rope_cos, self.rope_sin, pos_offset)
+                x = layer(x, self.rope_cos, self.rope_sin, pos_offset,
+                          cache=kv_cache, layer_idx=li)
+        if kv_cache is not None:
+            kv_cache.seq_len += input_ids.size(1)
         x = self.norm(x)
         if self.cfg.sequence_parallel:
             x = gather_from_sequence_parallel_region(x)
@@ -305,8 +347,10 @@ class LlamaForCausalLM(nn.Module):
         if cfg.tie_word_embeddings:
             self.lm_head.weight = self.model.embed_tokens.weight
 
-    def forward(self, input_ids, labels=None, loss_mask=None):
-        hidden = self.model(input_ids)
+    supports_kv_cache = True
+
+    def forward(self, input_ids, labels=None, loss_mask=None, kv_cache=None):
+        hidden = self.model(input_ids, kv_cache=kv_cache)
         logits = self.lm_head(hidden)  # [b, s, v/tp]
         if labels is None:
             return gather_from_tensor_model_parallel_region(logits)

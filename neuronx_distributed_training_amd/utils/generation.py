@@ -16,13 +16,23 @@ def generate(
     temperature: float = 0.0,
     top_k: int = 0,
     top_p: float = 0.0,
+    use_cache: bool = True,
 ) -> torch.Tensor:
-    """input_ids: [b, s]. Returns [b, s + new]. Full-recompute decode (no
-    KV cache) — intended for offline evaluation, not serving."""
+    """input_ids: [b, s]. Returns [b, s + new]. Incremental decode with a
+    KV cache when the model supports it (LlamaForCausalLM); full-recompute
+    fallback otherwise."""
     model.eval()
     ids = input_ids
+    cache = None
+    if use_cache and getattr(model, "supports_kv_cache", False) \
+            and not getattr(model.cfg, "sequence_parallel", False):
+        from ..models.llama import KVCache
+
+        cache = KVCache(len(model.model.layers))
+    step_in = ids
     for _ in range(max_new_tokens):
-        logits = model(ids)  # [b, s, V] (gathered over TP by the model)
+        logits = model(step_in, kv_cache=cache) if cache is not None \
+            else model(ids)  # [b, s, V] (gathered over TP by the model)
         nxt = logits[:, -1].float()
         if temperature and temperature > 0:
             nxt = nxt / temperature
@@ -42,6 +52,7 @@ def generate(
         else:
             tok = nxt.argmax(-1, keepdim=True)
         ids = torch.cat([ids, tok], dim=1)
+        step_in = tok  # cache path feeds only the new token next step
         if eos_token_id is not None and bool((tok == eos_token_id).all()):
             break
     return ids

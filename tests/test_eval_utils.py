@@ -129,3 +129,32 @@ def test_generation_sampling_options():
     out_p = generate(model, ids, max_new_tokens=4, temperature=1.0, top_p=1e-9)
     out_g = generate(model, ids, max_new_tokens=4)
     assert torch.equal(out_p, out_g)
+
+
+def test_kv_cache_decode_matches_recompute():
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.llama import (
+        KVCache, LlamaConfig, LlamaForCausalLM,
+    )
+    from neuronx_distributed_training_amd.utils.generation import generate
+
+    ps.destroy_model_parallel()
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(
+        LlamaConfig(vocab_size=64, hidden_size=32, intermediate_size=64,
+                    num_hidden_layers=2, num_attention_heads=4,
+                    num_key_value_heads=2, max_position_embeddings=64)
+    ).eval()
+    ids = torch.randint(0, 64, (2, 8))
+    # logits equivalence: prefill+decode == full forward
+    with torch.no_grad():
+        full = model(ids)
+        cache = KVCache(2)
+        pre = model(ids[:, :-1], kv_cache=cache)
+        last = model(ids[:, -1:], kv_cache=cache)
+    assert torch.allclose(pre, full[:, :-1], atol=1e-5)
+    assert torch.allclose(last, full[:, -1:], atol=1e-5)
+    # greedy generation identical with and without the cache
+    out_c = generate(model, ids, max_new_tokens=6, use_cache=True)
+    out_r = generate(model, ids, max_new_tokens=6, use_cache=False)
+    assert torch.equal(out_c, out_r)
