@@ -60,8 +60,11 @@ def _cpu_copy(obj):
 
 
 class CheckpointIO:
-    def __init__(self, async_save: bool = False):
+    def __init__(self, async_save: bool = False, save_bf16: bool = False):
         self.async_save = async_save
+        # reference exp_manager `save_bf16`: cast fp32 model tensors to
+        # bf16 on save (halves shard size; optimizer masters stay fp32)
+        self.save_bf16 = save_bf16
         self._pending: List[threading.Thread] = []
 
     # ---- save ----
@@ -80,9 +83,16 @@ class CheckpointIO:
         dp, tp, pp = _rank_tags()
         work = []
         if dp == 0 and ps.get_context_model_parallel_rank() == 0:
+            msd = _cpu_copy(module.model.state_dict())
+            if self.save_bf16:
+                msd = {
+                    k: (v.to(torch.bfloat16)
+                        if torch.is_tensor(v) and v.dtype == torch.float32
+                        else v)
+                    for k, v in msd.items()
+                }
             work.append(
-                (os.path.join(root, "model", _model_shard_name()),
-                 _cpu_copy(module.model.state_dict()))
+                (os.path.join(root, "model", _model_shard_name()), msd)
             )
         if module.optimizer is not None and ps.get_context_model_parallel_rank() == 0:
             work.append(

@@ -48,7 +48,10 @@ class Trainer:
         self.ckpt_every = int(ck.get("every_n_train_steps", 0) or 0)
         self.save_top_k = int(ck.get("save_top_k", 1))
         self.async_save = bool(em.get("async_checkpointing", False))
-        self.ckpt_io = CheckpointIO(async_save=self.async_save)
+        self.ckpt_io = CheckpointIO(
+            async_save=self.async_save,
+            save_bf16=bool(em.get("save_bf16", False)),
+        )
         self.global_step = 0
 
     def _log(self, metrics: Dict, step: int):

@@ -242,3 +242,40 @@ def _autocast_step(rank, world):
 
 def test_autocast_precision_mode():
     run_distributed(_autocast_step, 1)
+
+
+def _save_bf16(rank, world, tmpdir):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.checkpoint import CheckpointIO
+    from neuronx_distributed_training_amd.models.llama import (
+        LlamaConfig, LlamaForCausalLM,
+    )
+
+    ps.initialize_model_parallel()
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(
+        LlamaConfig(vocab_size=64, hidden_size=32, intermediate_size=64,
+                    num_hidden_layers=1, num_attention_heads=2,
+                    num_key_value_heads=1, max_position_embeddings=32)
+    )
+    holder = type("M", (), {})()
+    holder.model = model
+    holder.optimizer = None
+    holder.scheduler = None
+    CheckpointIO(save_bf16=True).save(tmpdir, "t", holder, {})
+    sd = torch.load(
+        os.path.join(tmpdir, "t.ckpt", "model",
+                     "dp_rank_00_tp_rank_00_pp_rank_00.pt"),
+        weights_only=True,
+    )
+    assert all(v.dtype == torch.bfloat16 for v in sd.values()
+               if torch.is_tensor(v) and v.is_floating_point())
+    # bf16 shards load back into an fp32 model
+    holder.model.load_state_dict(
+        {k: v.to(torch.float32) for k, v in sd.items()}
+    )
+    return 0.0
+
+
+def test_save_bf16_checkpoint(tmp_path):
+    run_distributed(_save_bf16, 1, str(tmp_path))
