@@ -217,7 +217,7 @@ class LlamaAttention(nn.Module):
         )
 
     def forward(self, x, cos, sin, pos_offset: int = 0, cache=None,
-                layer_idx: int = 0):
+                layer_idx: int = 0, pad_mask=None):
         # x: [s(, /tp if SP), b, h]
         s_dim, b = x.size(0), x.size(1)
         d = self.head_dim
@@ -228,7 +228,22 @@ class LlamaAttention(nn.Module):
         v = v.view(s_full, b, self.n_kv_local, d).permute(1, 2, 0, 3)
         q = apply_rotary_pos_emb(q, cos, sin, pos_offset)
         k = apply_rotary_pos_emb(k, cos, sin, pos_offset)
-        if cache is not None:
+        if pad_mask is not None:
+            # padded batch (e.g. left-padded DPO prompts): eager SDPA with
+            # the combined causal+padding mask (reference CoreAttention
+            # masked_fill, modeling_llama.py:226-251). −1e4 additive, not
+            # −inf, so fully-masked pad-query rows don't NaN.
+            sq = q.size(2)
+            rep = self.n_heads_local // self.n_kv_local
+            causal = torch.ones(sq, sq, dtype=torch.bool,
+                                device=q.device).triu(1)
+            bad = causal.unsqueeze(0) | pad_mask[:, None, :]
+            am = bad.unsqueeze(1).to(q.dtype) * -1e4
+            o = torch.nn.functional.scaled_dot_product_attention(
+                q, k.repeat_interleave(rep, 1), v.repeat_interleave(rep, 1),
+                attn_mask=am, scale=self.scale,
+            )
+        elif cache is not None:
             k, v = cache.update(layer_idx, k, v)
             if q.size(2) < k.size(2):
                 # incremental decode: new queries attend to the whole cache
@@ -277,9 +292,10 @@ class LlamaDecoderLayer(nn.Module):
         self.mlp = LlamaMLP(cfg, layer_idx)
 
     def forward(self, x, cos, sin, pos_offset: int = 0, cache=None,
-                layer_idx: int = 0):
+                layer_idx: int = 0, pad_mask=None):
         x = x + self.self_attn(self.input_layernorm(x), cos, sin, pos_offset,
-                               cache=cache, layer_idx=layer_idx)
+                               cache=cache, layer_idx=layer_idx,
+                               pad_mask=pad_mask)
         x = x + self.mlp(self.post_attention_layernorm(x))
         return x
 
@@ -305,7 +321,7 @@ class LlamaModel(nn.Module):
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
 
-    def forward(self, input_ids, kv_cache=None):
+    def forward(self, input_ids, kv_cache=None, attention_mask=None):
         # input_ids: [b, s(, /cp)] — CP split done by the trainer. With
         # kv_cache, input_ids are the NEW tokens only (decode path).
         if kv_cache is not None:
@@ -314,6 +330,11 @@ class LlamaModel(nn.Module):
         else:
             cp_rank = ps.get_context_model_parallel_rank()
             pos_offset = cp_rank * input_ids.size(1)
+        pad_mask = None
+        if attention_mask is not None and bool((attention_mask == 0).any()):
+            assert ps.get_context_model_parallel_world_size() == 1, \
+                "padding masks are not supported under context parallelism"
+            pad_mask = attention_mask == 0  # [b, s] True at pad keys
         x = self.embed_tokens(input_ids)  # [b, s, h]
         x = x.transpose(0, 1).contiguous()  # [s, b, h]
         if self.cfg.sequence_parallel:
@@ -322,10 +343,10 @@ class LlamaModel(nn.Module):
         for li, layer in enumerate(self.layers):
             if full_ckpt:
                 x = _ckpt(layer, x, self.rope_cos, self.rope_sin, pos_offset,
-                          use_reentrant=False)
+                          use_reentrant=False, pad_mask=pad_mask)
             else:
                 x = layer(x, self.rope_cos, self.rope_sin, pos_offset,
-                          cache=kv_cache, layer_idx=li)
+                          cache=kv_cache, layer_idx=li, pad_mask=pad_mask)
         if kv_cache is not None:
             kv_cache.seq_len += input_ids.size(1)
         x = self.norm(x)
@@ -349,8 +370,10 @@ class LlamaForCausalLM(nn.Module):
 
     supports_kv_cache = True
 
-    def forward(self, input_ids, labels=None, loss_mask=None, kv_cache=None):
-        hidden = self.model(input_ids, kv_cache=kv_cache)
+    def forward(self, input_ids, labels=None, loss_mask=None, kv_cache=None,
+                attention_mask=None):
+        hidden = self.model(input_ids, kv_cache=kv_cache,
+                            attention_mask=attention_mask)
         logits = self.lm_head(hidden)  # [b, s, v/tp]
         if labels is None:
             return gather_from_tensor_model_parallel_region(logits)

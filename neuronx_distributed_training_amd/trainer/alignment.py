@@ -49,8 +49,9 @@ class DPOBaseModel(LlamaModule):
         self.kl_beta = float(dcfg.get("kl_beta", dcfg.get("beta", 0.1)))
 
     # -- log-prob of the labeled (response) tokens of each sample --
-    def _sequence_logps(self, input_ids, labels, loss_mask, average=False):
-        hidden = self.model.model(input_ids)  # [b, s, h]
+    def _sequence_logps(self, input_ids, labels, loss_mask, average=False,
+                        attention_mask=None):
+        hidden = self.model.model(input_ids, attention_mask=attention_mask)
         logits = self.model.lm_head(hidden)   # [b, s, v/tp]
         safe_labels = labels.clamp(min=0)
         lp = from_parallel_logits_to_logprobs(logits, safe_labels)  # [b, s-1]
@@ -77,10 +78,12 @@ class DPOBaseModel(LlamaModule):
                 c = self._sequence_logps(
                     batch["chosen_input_ids"], batch["chosen_labels"],
                     batch["chosen_loss_mask"],
+                    attention_mask=batch.get("chosen_attention_mask"),
                 )
                 r = self._sequence_logps(
                     batch["rejected_input_ids"], batch["rejected_labels"],
                     batch["rejected_loss_mask"],
+                    attention_mask=batch.get("rejected_attention_mask"),
                 )
                 ref_c.append(c.float().cpu())
                 ref_r.append(r.float().cpu())
@@ -106,7 +109,12 @@ class DPOBaseModel(LlamaModule):
         ids = torch.cat([batch["chosen_input_ids"], batch["rejected_input_ids"]])
         labels = torch.cat([batch["chosen_labels"], batch["rejected_labels"]])
         mask = torch.cat([batch["chosen_loss_mask"], batch["rejected_loss_mask"]])
-        logps = self._sequence_logps(ids, labels, mask)
+        am = None
+        if batch.get("chosen_attention_mask") is not None:
+            am = torch.cat(
+                [batch["chosen_attention_mask"], batch["rejected_attention_mask"]]
+            )
+        logps = self._sequence_logps(ids, labels, mask, attention_mask=am)
         b = batch["chosen_input_ids"].size(0)
         pol_c, pol_r = logps[:b], logps[b:]
         ref_c = batch["ref_chosen_logps"].to(pol_c.device).to(pol_c.dtype)
@@ -131,7 +139,13 @@ class ORPOBaseModel(DPOBaseModel):
         ids = torch.cat([batch["chosen_input_ids"], batch["rejected_input_ids"]])
         labels = torch.cat([batch["chosen_labels"], batch["rejected_labels"]])
         mask = torch.cat([batch["chosen_loss_mask"], batch["rejected_loss_mask"]])
-        avg_logps = self._sequence_logps(ids, labels, mask, average=True)
+        am = None
+        if batch.get("chosen_attention_mask") is not None:
+            am = torch.cat(
+                [batch["chosen_attention_mask"], batch["rejected_attention_mask"]]
+            )
+        avg_logps = self._sequence_logps(ids, labels, mask, average=True,
+                                         attention_mask=am)
         b = batch["chosen_input_ids"].size(0)
         lc, lr = avg_logps[:b], avg_logps[b:]
         # log odds: log(p/(1-p)) with p = exp(avg_logp)

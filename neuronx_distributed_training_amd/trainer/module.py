@@ -114,10 +114,14 @@ class BaseModelModule:
         )
 
     def model_fwd_calc_loss(self, batch: Dict[str, torch.Tensor]) -> torch.Tensor:
+        kw = {}
+        if batch.get("attention_mask") is not None:
+            kw["attention_mask"] = batch["attention_mask"]
         return self.model(
             batch["input_ids"],
             labels=batch.get("labels", batch["input_ids"]),
             loss_mask=batch.get("loss_mask"),
+            **kw,
         )
 
     def get_batch_on_this_context_parallel_rank(self, batch):
