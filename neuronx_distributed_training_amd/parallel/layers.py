@@ -37,6 +37,168 @@ from .mappings import (
     scatter_to_tensor_model_parallel_region,
 )
 
+# --- SP comm/GEMM overlap (dark: NXDT_SP_OVERLAP=<chunks>, default off) ---
+#
+# Pipelines the SP collectives against the adjacent GEMM in chunks along
+# the sequence dim: Column fwd overlaps the all-gather with the GEMM
+# (chunk c computes while chunk c+1 gathers), Column bwd and Row fwd
+# overlap the reduce-scatter behind the next chunk's GEMM. Numerically
+# identical to the plain path (CPU/gloo tests assert bit-level equality);
+# scheduled for A/B measurement on the 8-GPU xGMI clique (ROADMAP §3).
+import os as _os
+
+_SP_OVERLAP_CHUNKS = int(_os.environ.get("NXDT_SP_OVERLAP", "0") or 0)
+
+
+def _sp_group():
+    import torch.distributed as dist
+
+    return dist, ps.get_tensor_model_parallel_group()
+
+
+class _SPOverlapColumnLinear(torch.autograd.Function):
+    """SP ColumnParallel: chunked all-gather(x) pipelined with the GEMM in
+    forward; chunked dgrad GEMM pipelined with reduce-scatter in backward."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, chunks):
+        dist, group = _sp_group()
+        world = ps.get_tensor_model_parallel_world_size()
+        s_loc = x.size(0)
+        cs = s_loc // chunks
+        x = x.contiguous()
+        rest = tuple(x.shape[1:])
+        temps, handles = [], []
+        for c in range(chunks):
+            xc = x[c * cs : (c + 1) * cs].contiguous()
+            buf = torch.empty((world * cs,) + rest, dtype=x.dtype, device=x.device)
+            h = dist.all_gather_into_tensor(buf, xc, group=group, async_op=True)
+            temps.append(buf)
+            handles.append(h)
+        x_full = torch.empty((world * s_loc,) + rest, dtype=x.dtype, device=x.device)
+        y = torch.empty(
+            (world * s_loc,) + rest[:-1] + (weight.size(0),),
+            dtype=x.dtype, device=x.device,
+        )
+        for c in range(chunks):
+            handles[c].wait()
+            piece = temps[c]                     # [world*cs, ...] rank-major
+            yc = F.linear(piece, weight, bias)   # GEMM while later gathers run
+            pv = piece.view((world, cs) + rest)
+            yv = yc.view((world, cs) + rest[:-1] + (weight.size(0),))
+            for r in range(world):
+                lo = r * s_loc + c * cs
+                x_full[lo : lo + cs] = pv[r]
+                y[lo : lo + cs] = yv[r]
+        ctx.save_for_backward(x_full, weight)
+        ctx.has_bias = bias is not None
+        ctx.chunks = chunks
+        ctx.s_loc = s_loc
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        dist, group = _sp_group()
+        world = ps.get_tensor_model_parallel_world_size()
+        x_full, weight = ctx.saved_tensors
+        chunks, s_loc = ctx.chunks, ctx.s_loc
+        cs = s_loc // chunks
+        dy = dy.contiguous()
+        rest = tuple(x_full.shape[1:])
+        dx = torch.empty((s_loc,) + rest, dtype=dy.dtype, device=dy.device)
+        handles = []
+        for c in range(chunks):
+            # contiguous (rank, chunk-c) rows → dgrad GEMM → async RS;
+            # the next chunk's GEMM runs while this RS is in flight
+            buf = torch.empty((world * cs,) + rest[:-1] + (dy.size(-1),),
+                              dtype=dy.dtype, device=dy.device)
+            bv = buf.view((world, cs) + rest[:-1] + (dy.size(-1),))
+            for r in range(world):
+                lo = r * s_loc + c * cs
+                bv[r] = dy[lo : lo + cs]
+            dxc_full = buf @ weight            # [world*cs, ..., in]
+            h = dist.reduce_scatter_tensor(
+                dx[c * cs : (c + 1) * cs], dxc_full.contiguous(),
+                group=group, async_op=True,
+            )
+            handles.append(h)
+        dw = dy.reshape(-1, dy.size(-1)).T @ x_full.reshape(-1, x_full.size(-1))
+        db = dy.reshape(-1, dy.size(-1)).sum(0) if ctx.has_bias else None
+        for h in handles:
+            h.wait()
+        return dx, dw, db, None
+
+
+class _SPOverlapRowLinear(torch.autograd.Function):
+    """SP RowParallel: GEMM chunks pipelined with the output
+    reduce-scatter in forward; chunked all-gather + GEMM in backward."""
+
+    @staticmethod
+    def forward(ctx, x, weight, chunks):
+        dist, group = _sp_group()
+        world = ps.get_tensor_model_parallel_world_size()
+        s_full = x.size(0)
+        s_loc = s_full // world
+        cs = s_loc // chunks
+        x = x.contiguous()
+        rest = tuple(x.shape[1:])
+        out = torch.empty((s_loc,) + rest[:-1] + (weight.size(0),),
+                          dtype=x.dtype, device=x.device)
+        handles = []
+        for c in range(chunks):
+            buf = torch.empty((world * cs,) + rest[:-1] + (weight.size(0),),
+                              dtype=x.dtype, device=x.device)
+            bv = buf.view((world, cs) + rest[:-1] + (weight.size(0),))
+            for r in range(world):
+                lo = r * s_loc + c * cs
+                bv[r] = F.linear(x[lo : lo + cs], weight)
+            h = dist.reduce_scatter_tensor(
+                out[c * cs : (c + 1) * cs], buf.contiguous(),
+                group=group, async_op=True,
+            )
+            handles.append(h)
+        for h in handles:
+            h.wait()
+        ctx.save_for_backward(x, weight)
+        ctx.chunks = chunks
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        dist, group = _sp_group()
+        world = ps.get_tensor_model_parallel_world_size()
+        x, weight = ctx.saved_tensors
+        chunks = ctx.chunks
+        s_loc = dy.size(0)
+        cs = s_loc // chunks
+        dy = dy.contiguous()
+        rest = tuple(dy.shape[1:])
+        temps, handles = [], []
+        for c in range(chunks):
+            buf = torch.empty((world * cs,) + rest, dtype=dy.dtype, device=dy.device)
+            h = dist.all_gather_into_tensor(
+                buf, dy[c * cs : (c + 1) * cs].contiguous(),
+                group=group, async_op=True,
+            )
+            temps.append(buf)
+            handles.append(h)
+        dx = torch.empty_like(x)
+        dy_full = torch.empty((world * s_loc,) + rest, dtype=dy.dtype,
+                              device=dy.device)
+        for c in range(chunks):
+            handles[c].wait()
+            piece = temps[c]
+            dxc = piece @ weight
+            pv = piece.view((world, cs) + rest)
+            xv = dxc.view((world, cs) + tuple(x.shape[1:])[:-1] + (weight.size(1),))
+            for r in range(world):
+                lo = r * s_loc + c * cs
+                dy_full[lo : lo + cs] = pv[r]
+                dx[lo : lo + cs] = xv[r]
+        dw = dy_full.reshape(-1, dy_full.size(-1)).T @ x.reshape(-1, x.size(-1))
+        return dx, dw, None
+
+
 __all__ = [
     "ColumnParallelLinear",
     "RowParallelLinear",
@@ -134,6 +296,15 @@ class ColumnParallelLinear(nn.Module):
         reduces exactly once through the caller's mapping."""
         if not pre_mapped:
             if self.sequence_parallel:
+                world = ps.get_tensor_model_parallel_world_size()
+                if (
+                    _SP_OVERLAP_CHUNKS > 0 and world > 1
+                    and not self.gather_output
+                    and x.size(0) % _SP_OVERLAP_CHUNKS == 0
+                ):
+                    return _SPOverlapColumnLinear.apply(
+                        x, self.weight, self.bias, _SP_OVERLAP_CHUNKS
+                    )
                 x = gather_from_sequence_parallel_region(x)
             else:
                 x = copy_to_tensor_model_parallel_region(x)
@@ -183,6 +354,15 @@ class RowParallelLinear(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if not self.input_is_parallel:
             x = scatter_to_tensor_model_parallel_region(x)
+        world = ps.get_tensor_model_parallel_world_size()
+        if (
+            self.sequence_parallel and _SP_OVERLAP_CHUNKS > 0 and world > 1
+            and (x.size(0) // world) % _SP_OVERLAP_CHUNKS == 0
+        ):
+            out = _SPOverlapRowLinear.apply(x, self.weight, _SP_OVERLAP_CHUNKS)
+            if self.bias is not None:
+                out = out + self.bias
+            return out
         out = F.linear(x, self.weight)
         if self.sequence_parallel:
             out = reduce_scatter_to_sequence_parallel_region(out)

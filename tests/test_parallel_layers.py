@@ -146,3 +146,47 @@ def _sp_roundtrip(rank, world):
 @pytest.mark.parametrize("world", [1, 2])
 def test_sequence_parallel_mappings(world):
     run_distributed(_sp_roundtrip, world)
+
+
+def _sp_overlap_parity(rank, world):
+    """NXDT_SP_OVERLAP chunked comm/GEMM pipeline == plain SP path exactly
+    (fwd values and all grads)."""
+    import os
+    import neuronx_distributed_training_amd.parallel.layers as L
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.parallel.layers import (
+        ColumnParallelLinear, RowParallelLinear,
+    )
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(0)
+    col = ColumnParallelLinear(32, 64, sequence_parallel=True, init_seed=1)
+    row = RowParallelLinear(64, 32, sequence_parallel=True, init_seed=2)
+    g = torch.Generator().manual_seed(7 + rank)
+    x = torch.randn(16 // world, 2, 32, generator=g)  # [s/tp, b, h]
+
+    def run():
+        xi = x.clone().requires_grad_(True)
+        y = row(torch.nn.functional.gelu(col(xi)))
+        y.square().sum().backward()
+        return (y.detach().clone(), xi.grad.clone(),
+                col.weight.grad.clone(), row.weight.grad.clone())
+
+    L._SP_OVERLAP_CHUNKS = 0
+    ref = run()
+    for p in (col.weight, row.weight, col.bias):
+        if p is not None and p.grad is not None:
+            p.grad = None
+    L._SP_OVERLAP_CHUNKS = 4
+    try:
+        out = run()
+    finally:
+        L._SP_OVERLAP_CHUNKS = 0
+    for a, b in zip(ref, out):
+        assert torch.allclose(a, b, atol=1e-5), (a - b).abs().max()
+    return 0.0
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_sp_overlap_matches_plain(world):
+    run_distributed(_sp_overlap_parity, world)
