@@ -246,12 +246,20 @@ class LlamaAttention(nn.Module):
         elif cache is not None:
             k, v = cache.update(layer_idx, k, v)
             if q.size(2) < k.size(2):
-                # incremental decode: new queries attend to the whole cache
+                # incremental decode: query at global position past+i sees
+                # keys 0..past+i (causal within the new chunk too)
                 rep = self.n_heads_local // self.n_kv_local
+                sq, sk = q.size(2), k.size(2)
+                past = sk - sq
+                am = None
+                if sq > 1:
+                    bad = torch.ones(sq, sk, dtype=torch.bool,
+                                     device=q.device).triu(past + 1)
+                    am = bad.to(q.dtype) * -1e4
                 o = torch.nn.functional.scaled_dot_product_attention(
                     q, k.repeat_interleave(rep, 1),
                     v.repeat_interleave(rep, 1),
-                    is_causal=False, scale=self.scale,
+                    attn_mask=am, is_causal=False, scale=self.scale,
                 )
             else:
                 o = self.core_attention(q, k, v)  # prefill

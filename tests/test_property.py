@@ -80,3 +80,71 @@ def test_sampler_partition(total, mbs, dp, consumed_batches):
                 assert consumed <= idx < total
                 assert idx not in seen
                 seen.add(idx)
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    vocab=st.integers(min_value=1, max_value=300000),
+    div=st.integers(min_value=1, max_value=256),
+    tp=st.sampled_from([1, 2, 4, 8, 16, 32]),
+)
+def test_pad_vocab_properties(vocab, div, tp):
+    from neuronx_distributed_training_amd.data.datamodule import pad_vocab_size
+
+    p = pad_vocab_size(vocab, div, tp)
+    assert p >= vocab                   # never shrinks
+    assert p % (div * tp) == 0          # aligned
+    assert p - vocab < div * tp         # minimal padding
+    assert pad_vocab_size(p, div, tp) == p  # idempotent
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    weights=st.lists(st.floats(min_value=0.01, max_value=10.0),
+                     min_size=1, max_size=6),
+    size=st.integers(min_value=1, max_value=500),
+)
+def test_blend_indices_properties(weights, size):
+    import numpy as np
+    from neuronx_distributed_training_amd.data.gpt_dataset import _blend_indices
+
+    di, si = _blend_indices(weights, size)
+    assert len(di) == size and len(si) == size
+    w = np.asarray(weights) / sum(weights)
+    for j in range(len(weights)):
+        got = (di == j).sum()
+        # composition within ±1 of the exact proportional count at the end
+        assert abs(got - round(size * w[j])) <= 1 + size * 0.01
+        # per-dataset sample indices are 0..k-1 in order
+        sj = si[di == j]
+        assert (sj == np.arange(len(sj))).all()
+
+
+@settings(max_examples=25, deadline=None)
+@given(
+    s=st.integers(min_value=2, max_value=12),
+    split=st.integers(min_value=1, max_value=11),
+)
+def test_kv_cache_split_invariance(s, split):
+    """Decoding with any prefill/decode split matches the full forward."""
+    import torch
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.llama import (
+        KVCache, LlamaConfig, LlamaForCausalLM,
+    )
+
+    split = min(split, s - 1)
+    ps.destroy_model_parallel()
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(
+        LlamaConfig(vocab_size=32, hidden_size=16, intermediate_size=32,
+                    num_hidden_layers=1, num_attention_heads=2,
+                    num_key_value_heads=1, max_position_embeddings=16)
+    ).eval()
+    ids = torch.randint(0, 32, (1, s), generator=torch.Generator().manual_seed(s))
+    with torch.no_grad():
+        full = model(ids)
+        cache = KVCache(1)
+        a = model(ids[:, :split], kv_cache=cache)
+        b = model(ids[:, split:], kv_cache=cache)
+    assert torch.allclose(torch.cat([a, b], 1), full, atol=1e-5)
