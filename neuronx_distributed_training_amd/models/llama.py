@@ -328,6 +328,10 @@ class LlamaModel(nn.Module):
         )
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
+        if cfg.sequence_parallel:
+            from ..parallel.layers import tag_sequence_parallel_params
+
+            tag_sequence_parallel_params(self)
 
     def forward(self, input_ids, kv_cache=None, attention_mask=None):
         # input_ids: [b, s(, /cp)] — CP split done by the trainer. With
@@ -382,7 +386,10 @@ class LlamaForCausalLM(nn.Module):
                 attention_mask=None):
         hidden = self.model(input_ids, kv_cache=kv_cache,
                             attention_mask=attention_mask)
-        logits = self.lm_head(hidden)  # [b, s, v/tp]
+        # under SP the model's output gather already provides the TP input
+        # mapping (backward reduce-scatter); pre_mapped skips the copy so
+        # the reduction isn't applied twice
+        logits = self.lm_head(hidden, pre_mapped=self.cfg.sequence_parallel)
         if labels is None:
             return gather_from_tensor_model_parallel_region(logits)
         cp = ps.get_context_model_parallel_world_size()

@@ -54,6 +54,10 @@ class LlamaChunk(nn.Module):
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
         self._batch = {}
+        if cfg.sequence_parallel:
+            from ..parallel.layers import tag_sequence_parallel_params
+
+            tag_sequence_parallel_params(self)
 
     def set_batch(self, batch):
         dev = next(self.parameters()).device
@@ -84,7 +88,7 @@ class LlamaChunk(nn.Module):
 
             x = gather_from_sequence_parallel_region(x)
         x = self.norm(x)
-        logits = self.lm_head(x).transpose(0, 1)
+        logits = self.lm_head(x, pre_mapped=self.cfg.sequence_parallel).transpose(0, 1)
         labels = self._batch.get("labels", self._batch["input_ids"])
         loss_mask = self._batch.get("loss_mask")
         logits = logits[:, :-1]
@@ -170,6 +174,10 @@ class LlamaStage(nn.Module):
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
         self._batch: Dict[str, torch.Tensor] = {}
+        if cfg.sequence_parallel:
+            from ..parallel.layers import tag_sequence_parallel_params
+
+            tag_sequence_parallel_params(self)
 
     @property
     def tied_embedding_weight(self):
@@ -211,7 +219,7 @@ class LlamaStage(nn.Module):
 
             x = gather_from_sequence_parallel_region(x)
         x = self.norm(x)
-        logits = self.lm_head(x).transpose(0, 1)  # [b, s, v/tp]
+        logits = self.lm_head(x, pre_mapped=self.cfg.sequence_parallel).transpose(0, 1)  # [b, s, v/tp]
         labels = self._batch.get("labels", self._batch["input_ids"])
         loss_mask = self._batch.get("loss_mask")
         cp = ps.get_context_model_parallel_world_size()

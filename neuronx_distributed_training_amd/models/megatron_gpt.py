@@ -374,6 +374,10 @@ class GPTModel(nn.Module):
             self.register_buffer("rope_sin", sin, persistent=False)
         else:
             self.rope_cos = self.rope_sin = None
+        if cfg.sequence_parallel:
+            from ..parallel.layers import tag_sequence_parallel_params
+
+            tag_sequence_parallel_params(self)
 
     def forward(self, input_ids, position_ids=None, labels=None, loss_mask=None):
         cp_rank = ps.get_context_model_parallel_rank()
@@ -393,9 +397,17 @@ class GPTModel(nn.Module):
         if self.cfg.sequence_parallel:
             x = gather_from_sequence_parallel_region(x)
         if self.cfg.share_embeddings_and_output_weights:
+            if not self.cfg.sequence_parallel:
+                # replicate-input mapping: backward all-reduces the partial
+                # dX from each vocab shard (the SP gather handles it
+                # otherwise via reduce-scatter)
+                from ..parallel.mappings import (
+                    copy_to_tensor_model_parallel_region,
+                )
+                x = copy_to_tensor_model_parallel_region(x)
             logits = F.linear(x, self.embedding.word_embeddings.weight)
         else:
-            logits = self.output_layer(x)
+            logits = self.output_layer(x, pre_mapped=self.cfg.sequence_parallel)
         logits = logits.transpose(0, 1)  # [b, s, v/tp]
         if labels is None:
             from ..parallel.mappings import gather_from_tensor_model_parallel_region

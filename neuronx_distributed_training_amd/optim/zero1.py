@@ -124,6 +124,17 @@ class ZeRO1AdamW:
             if not is_tp and tp_rank != 0:
                 tp_once_mask[o : o + nel] = False
         self.wd_shard = wd_mask[self.shard_start : self.shard_start + self.shard_size].to(dev)
+
+        # shard-relative ranges of sequence-parallel-tagged params (norm
+        # weights etc. that see sequence-SHARDED activations): their grads
+        # are partial per TP rank and get SUMMED over TP in step()
+        self._sp_ranges = []
+        lo, hi = self.shard_start, self.shard_start + self.shard_size
+        for (n, p), o in zip(self.named_params, offsets):
+            if getattr(p, "sequence_parallel_enabled", False):
+                a, b = max(o, lo), min(o + p.numel(), hi)
+                if a < b:
+                    self._sp_ranges.append((a - lo, b - lo))
         nm = tp_once_mask[self.shard_start : self.shard_start + self.shard_size]
         self.normmask_all = bool(nm.all())
         self.normmask_shard = nm.to(dev).to(torch.float32)
@@ -292,6 +303,17 @@ class ZeRO1AdamW:
         if ps.get_context_model_parallel_world_size() > 1:
             dist.all_reduce(shard, group=cp_group)
             shard.div_(ps.get_context_model_parallel_world_size())
+
+        # 1c) SUM sequence-parallel-tagged grads over TP (each rank's grad
+        # covers only its sequence shard — reference sequence_parallel_enabled
+        # tag semantics); single fused all-reduce over the packed segments
+        if self._sp_ranges and ps.get_tensor_model_parallel_world_size() > 1:
+            buf = torch.cat([shard[a:b] for a, b in self._sp_ranges])
+            dist.all_reduce(buf, group=ps.get_tensor_model_parallel_group())
+            off = 0
+            for a, b in self._sp_ranges:
+                shard[a:b] = buf[off : off + (b - a)]
+                off += b - a
 
         # 1b) expert grads: average over the expert-DP group
         expert_grads = []

@@ -199,6 +199,36 @@ class _SPOverlapRowLinear(torch.autograd.Function):
         return dx, dw, None
 
 
+def tag_sequence_parallel_params(module):
+    """Mark replicated parameters that operate on sequence-SHARDED
+    activations (norm weights/biases, RowParallel biases): each TP rank's
+    autograd grad is only that rank's sequence-shard contribution, so the
+    optimizer must SUM them over the TP group before stepping (reference
+    ``sequence_parallel_enabled`` weight tag, modeling_llama.py:151).
+    Call only when the model runs with sequence_parallel=True."""
+    from ..ops.rmsnorm import RMSNorm
+
+    for m in module.modules():
+        if isinstance(m, (RMSNorm, nn.LayerNorm)):
+            for p in m.parameters(recurse=False):
+                p.sequence_parallel_enabled = True
+        elif isinstance(m, RowParallelLinear) and m.sequence_parallel \
+                and m.bias is not None:
+            m.bias.sequence_parallel_enabled = True
+
+
+def allreduce_sequence_parallel_grads(module):
+    """SUM tagged params' .grad over the TP group (for raw-autograd users;
+    ZeRO1AdamW does this internally on its flat buffer)."""
+    import torch.distributed as dist
+
+    if ps.get_tensor_model_parallel_world_size() == 1:
+        return
+    for p in module.parameters():
+        if getattr(p, "sequence_parallel_enabled", False) and p.grad is not None:
+            dist.all_reduce(p.grad, group=ps.get_tensor_model_parallel_group())
+
+
 __all__ = [
     "ColumnParallelLinear",
     "RowParallelLinear",

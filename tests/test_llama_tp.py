@@ -211,3 +211,73 @@ def test_padded_attention_mask(world):
     r = run_distributed(_padded_attention, world)
     if world == 2:
         assert abs(r[0] - r[1]) < 1e-6
+
+
+def _sp_grad_exact(rank, world):
+    """SP backward grads are EXACT vs TP=1 (regression: the LM head's copy
+    mapping double-reduced below-head grads by ×tp under SP; SP-region norm
+    grads lacked the TP sum)."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.parallel.layers import (
+        allreduce_sequence_parallel_grads,
+    )
+    from neuronx_distributed_training_amd.models.llama import (
+        LlamaConfig, LlamaForCausalLM,
+    )
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(7)
+    cfg = LlamaConfig(**TINY, sequence_parallel=world > 1)
+    m = LlamaForCausalLM(cfg)
+    ids = torch.randint(0, 128, (2, 32), generator=torch.Generator().manual_seed(9))
+    m(ids, labels=ids).backward()
+    allreduce_sequence_parallel_grads(m)
+    tp_r = ps.get_tensor_model_parallel_rank()
+    og = m.model.layers[0].self_attn.o_proj.weight.grad[:, :8]
+    ng = m.model.layers[0].input_layernorm.weight.grad
+    return (og.detach().clone(), ng.detach().clone()) if tp_r == 0 else None
+
+
+def test_sp_grads_exact_vs_tp1():
+    og1, ng1 = run_distributed(_sp_grad_exact, 1)[0]
+    og2, ng2 = [r for r in run_distributed(_sp_grad_exact, 2) if r is not None][0]
+    assert torch.allclose(og1, og2, atol=1e-5), (og2 / og1).median()
+    assert torch.allclose(ng1, ng2, atol=1e-5), (ng2 / ng1).median()
+
+
+def _sp_train_weights(rank, world):
+    """Optimizer path: after 3 ZeRO-1 steps under TP2+SP the replicated
+    norm weights match the TP=1 run (validates the in-optimizer TP sum of
+    sequence-parallel-tagged grads)."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 2, "seq_length": 32},
+        "distributed_strategy": {"tensor_model_parallel_size": world,
+                                 "sequence_parallel": world > 1},
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 2, "num_attention_heads": 4, "num_kv_heads": 2,
+            "grad_clip": 1.0, "optim": {"lr": 1e-2, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(3)
+    mod = LlamaModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=6)
+    g = torch.Generator().manual_seed(5)
+    for _ in range(3):
+        ids = torch.randint(0, 128, (2, 32), generator=g)
+        mod.training_step([{"input_ids": ids, "labels": ids.clone()}])
+    w = mod.model.model.layers[0].input_layernorm.weight.detach().clone()
+    return w if ps.get_tensor_model_parallel_rank() == 0 else None
+
+
+def test_sp_norm_weights_match_after_training():
+    w1 = run_distributed(_sp_train_weights, 1)[0]
+    w2 = [w for w in run_distributed(_sp_train_weights, 2) if w is not None][0]
+    assert torch.allclose(w1, w2, atol=1e-4), (w1 - w2).abs().max()
