@@ -174,10 +174,15 @@ class MixtralForCausalLM(nn.Module):
         )
         if cfg.tie_word_embeddings:
             self.lm_head.weight = self.model.embed_tokens.weight
+        if cfg.sequence_parallel:
+            from ..parallel.layers import tag_sequence_parallel_params
+
+            tag_sequence_parallel_params(self)
 
     def forward(self, input_ids, labels=None, loss_mask=None):
         hidden, router_logits = self.model(input_ids)
-        logits = self.lm_head(hidden)
+        # SP gather in the model already provides the TP input mapping
+        logits = self.lm_head(hidden, pre_mapped=self.cfg.sequence_parallel)
         if labels is None:
             return gather_from_tensor_model_parallel_region(logits)
         cp = ps.get_context_model_parallel_world_size()

@@ -52,7 +52,9 @@ class DPOBaseModel(LlamaModule):
     def _sequence_logps(self, input_ids, labels, loss_mask, average=False,
                         attention_mask=None):
         hidden = self.model.model(input_ids, attention_mask=attention_mask)
-        logits = self.model.lm_head(hidden)   # [b, s, v/tp]
+        logits = self.model.lm_head(
+            hidden, pre_mapped=self.model.cfg.sequence_parallel
+        )  # [b, s, v/tp]
         safe_labels = labels.clamp(min=0)
         lp = from_parallel_logits_to_logprobs(logits, safe_labels)  # [b, s-1]
         m = loss_mask[:, 1:].to(lp.dtype)
