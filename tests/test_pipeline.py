@@ -230,3 +230,56 @@ def test_tp2_pp2_dp2_matches_single():
     res = run_distributed(_full_3d, 8)
     assert max(abs(x - res[0]) for x in res) < 1e-5, res
     assert abs(ref - res[0]) < 5e-3, (ref, res[0])
+
+
+def _pp_tp_sp_weights(rank, world):
+    """2 training steps under PP2×TP2×SP: layer-0 norm weight matches the
+    single-rank run (optimizer-path SP grad sync through the PP engine)."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+
+    tp = 2 if world == 4 else 1
+    pp = 2 if world == 4 else 1
+    ps.initialize_model_parallel(
+        tensor_model_parallel_size=tp, pipeline_model_parallel_size=pp
+    )
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 1, "seq_length": 32},
+        "distributed_strategy": {
+            "tensor_model_parallel_size": tp,
+            "pipeline_model_parallel_size": pp,
+            "sequence_parallel": tp > 1,
+        },
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 4, "num_attention_heads": 4, "num_kv_heads": 2,
+            "grad_clip": 1.0, "optim": {"lr": 1e-2, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(7)
+    mod = LlamaModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=4)
+    g = torch.Generator().manual_seed(5)
+    for _ in range(2):
+        micros = [
+            {"input_ids": (ids := torch.randint(0, 128, (1, 32), generator=g)),
+             "labels": ids.clone()}
+            for _ in range(2)
+        ]
+        mod.training_step(micros)
+    if ps.get_pipeline_model_parallel_rank() == 0 \
+            and ps.get_tensor_model_parallel_rank() == 0:
+        stage = mod.model
+        layer = stage.layers[0] if hasattr(stage, "layers") \
+            else stage.model.layers[0]
+        return layer.input_layernorm.weight.detach().clone()
+    return None
+
+
+def test_pp2_tp2_sp_weights_match():
+    w1 = run_distributed(_pp_tp_sp_weights, 1)[0]
+    w4 = [w for w in run_distributed(_pp_tp_sp_weights, 4) if w is not None][0]
+    assert torch.allclose(w1, w4, atol=1e-4), (w1 - w4).abs().max()
