@@ -383,7 +383,7 @@ class LlamaForCausalLM(nn.Module):
     supports_kv_cache = True
 
     def forward(self, input_ids, labels=None, loss_mask=None, kv_cache=None,
-                attention_mask=None):
+                attention_mask=None, loss_denominator=None):
         hidden = self.model(input_ids, kv_cache=kv_cache,
                             attention_mask=attention_mask)
         # under SP the model's output gather already provides the TP input
@@ -402,7 +402,9 @@ class LlamaForCausalLM(nn.Module):
         per_tok = parallel_cross_entropy(logits, labels)
         if loss_mask is not None:
             m = loss_mask.to(per_tok.dtype)
-            loss = (per_tok * m).sum() / m.sum().clamp(min=1)
+            denom = (loss_denominator if loss_denominator is not None
+                     else m.sum()).clamp(min=1)
+            loss = (per_tok * m).sum() / denom
         else:
             loss = per_tok.mean()
         return loss

@@ -379,7 +379,8 @@ class GPTModel(nn.Module):
 
             tag_sequence_parallel_params(self)
 
-    def forward(self, input_ids, position_ids=None, labels=None, loss_mask=None):
+    def forward(self, input_ids, position_ids=None, labels=None, loss_mask=None,
+                loss_denominator=None):
         cp_rank = ps.get_context_model_parallel_rank()
         pos_offset = cp_rank * input_ids.size(1)
         x = self.embedding(input_ids, position_ids)
@@ -416,7 +417,9 @@ class GPTModel(nn.Module):
         per_tok = parallel_cross_entropy(logits, labels)
         if loss_mask is not None:
             m = loss_mask.to(per_tok.dtype)
-            loss = (per_tok * m).sum() / m.sum().clamp(min=1)
+            denom = (loss_denominator if loss_denominator is not None
+                     else m.sum()).clamp(min=1)
+            loss = (per_tok * m).sum() / denom
         else:
             loss = per_tok.mean()
         if router_logits:

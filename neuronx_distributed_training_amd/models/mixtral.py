@@ -179,7 +179,8 @@ class MixtralForCausalLM(nn.Module):
 
             tag_sequence_parallel_params(self)
 
-    def forward(self, input_ids, labels=None, loss_mask=None):
+    def forward(self, input_ids, labels=None, loss_mask=None,
+                loss_denominator=None):
         hidden, router_logits = self.model(input_ids)
         # SP gather in the model already provides the TP input mapping
         logits = self.lm_head(hidden, pre_mapped=self.cfg.sequence_parallel)
@@ -193,7 +194,9 @@ class MixtralForCausalLM(nn.Module):
         per_tok = parallel_cross_entropy(logits, labels)
         if loss_mask is not None:
             m = loss_mask.to(per_tok.dtype)
-            loss = (per_tok * m).sum() / m.sum().clamp(min=1)
+            denom = (loss_denominator if loss_denominator is not None
+                     else m.sum()).clamp(min=1)
+            loss = (per_tok * m).sum() / denom
         else:
             loss = per_tok.mean()
         if router_logits:

@@ -117,6 +117,8 @@ class BaseModelModule:
         kw = {}
         if batch.get("attention_mask") is not None:
             kw["attention_mask"] = batch["attention_mask"]
+        if batch.get("loss_denominator") is not None:
+            kw["loss_denominator"] = batch["loss_denominator"]
         return self.model(
             batch["input_ids"],
             labels=batch.get("labels", batch["input_ids"]),
@@ -146,6 +148,12 @@ class BaseModelModule:
             mask[:, -1] = 0.0  # no label for the final position
             batch["labels"] = shifted
             batch["loss_mask"] = mask
+            # exact global-mean loss under CP: every rank divides its
+            # masked sum by (global mask count / cp) so the CP loss/grad
+            # average reproduces sum/global exactly even when chunk mask
+            # counts differ (the reference's mean-of-local-means is only
+            # approximate there)
+            batch["loss_denominator"] = mask.sum() / cp
         out = {}
         for k, v in batch.items():
             if torch.is_tensor(v) and v.dim() >= 2 and v.size(1) == self.seq_length:

@@ -82,4 +82,5 @@ class MegatronGPTModule(BaseModelModule):
             position_ids=batch.get("position_ids"),
             labels=batch.get("labels", batch["input_ids"]),
             loss_mask=batch.get("loss_mask"),
+            loss_denominator=batch.get("loss_denominator"),
         )

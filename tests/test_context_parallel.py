@@ -120,3 +120,39 @@ def test_cp2_tp2_loss_matches_single():
     l4 = run_distributed(_cp_tp_train, 4)
     assert max(abs(l - l4[0]) for l in l4) < 1e-6  # consistent across ranks
     assert abs(l1 - l4[0]) < 0.05, (l1, l4[0])
+
+
+def _cp_weights(rank, world):
+    """2 training steps under CP=2: weights match single-rank (CP grad
+    averaging over the whole flat shard)."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+
+    ps.initialize_model_parallel(context_parallel_size=world)
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 2, "seq_length": 32},
+        "distributed_strategy": {"context_parallel_size": world},
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 2, "num_attention_heads": 4, "num_kv_heads": 2,
+            "grad_clip": 1.0, "optim": {"lr": 1e-2, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(3)
+    mod = LlamaModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=4)
+    g = torch.Generator().manual_seed(5)
+    for _ in range(2):
+        ids = torch.randint(0, 128, (2, 32), generator=g)
+        mod.training_step([{"input_ids": ids, "labels": ids.clone()}])
+    return mod.model.model.layers[0].input_layernorm.weight.detach().clone()
+
+
+def test_cp2_weights_match_cp1():
+    w1 = run_distributed(_cp_weights, 1)[0]
+    w2 = run_distributed(_cp_weights, 2)
+    assert torch.allclose(w2[0], w2[1], atol=1e-6)
+    assert torch.allclose(w1, w2[0], atol=1e-4), (w1 - w2[0]).abs().max()
