@@ -283,3 +283,50 @@ def test_pp2_tp2_sp_weights_match():
     w1 = run_distributed(_pp_tp_sp_weights, 1)[0]
     w4 = [w for w in run_distributed(_pp_tp_sp_weights, 4) if w is not None][0]
     assert torch.allclose(w1, w4, atol=1e-4), (w1 - w4).abs().max()
+
+
+def _vp_weights(rank, world):
+    """2 optimizer steps under PP2×VP2 (interleaved): chunk-0 layer-0 norm
+    weight matches the single-rank run."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+
+    pp = 2 if world == 2 else 1
+    ps.initialize_model_parallel(pipeline_model_parallel_size=pp)
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 1, "seq_length": 32},
+        "distributed_strategy": {
+            "pipeline_model_parallel_size": pp,
+            "virtual_pipeline_model_parallel_size": 2 if pp > 1 else 1,
+        },
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 4, "num_attention_heads": 4, "num_kv_heads": 2,
+            "grad_clip": 1.0, "optim": {"lr": 1e-2, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(7)
+    mod = LlamaModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=4)
+    g = torch.Generator().manual_seed(5)
+    for _ in range(2):
+        micros = [
+            {"input_ids": (ids := torch.randint(0, 128, (1, 32), generator=g)),
+             "labels": ids.clone()}
+            for _ in range(2)
+        ]
+        mod.training_step(micros)
+    if pp == 1:
+        return mod.model.model.layers[0].input_layernorm.weight.detach().clone()
+    if ps.get_pipeline_model_parallel_rank() == 0:
+        return mod.model[0].layers[0].input_layernorm.weight.detach().clone()
+    return None
+
+
+def test_vp2_weights_match_single():
+    w1 = run_distributed(_vp_weights, 1)[0]
+    w2 = [w for w in run_distributed(_vp_weights, 2) if w is not None][0]
+    assert torch.allclose(w1, w2, atol=1e-4), (w1 - w2).abs().max()
