@@ -6,6 +6,8 @@ from __future__ import annotations
 
 import torch
 
+from ..parallel import state as ps
+
 from .module import BaseModelModule
 from ..models.megatron_gpt import GPTConfig, GPTModel
 
@@ -73,6 +75,12 @@ class MegatronGPTModule(BaseModelModule):
                     "token_shuffle_group_size", 1)),
             moe_aux_loss_coeff=float(moe.get("aux_loss_coef", 0.01)),
         )
+        if ps.get_pipeline_model_parallel_world_size() > 1:
+            vp = int(dstr.get("virtual_pipeline_model_parallel_size", 1) or 1)
+            assert vp == 1, "interleaved VP not wired for megatron GPT yet"
+            from ..models.megatron_pipeline import GPTStage
+
+            return GPTStage(cfg, pipeline_cuts=mcfg.get("pipeline_cuts"))
         return GPTModel(cfg)
 
     def model_fwd_calc_loss(self, batch):

@@ -333,3 +333,56 @@ def test_megatron_head_grads_exact(tied, sp):
     g2 = [g for g in run_distributed(_mega_head_grads, 2, tied, sp)
           if g is not None][0]
     assert torch.allclose(g1, g2, atol=1e-5), (g1 - g2).abs().max()
+
+
+def _mega_pp(rank, world, tp):
+    """Megatron GPT under PP2(×TP2/SP): training-step loss matches the
+    single-rank GPTModel."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module_megatron import (
+        MegatronGPTModule,
+    )
+
+    pp = world // tp if world > 1 else 1
+    ps.initialize_model_parallel(
+        tensor_model_parallel_size=tp, pipeline_model_parallel_size=pp
+    )
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 1, "seq_length": 32},
+        "distributed_strategy": {
+            "tensor_model_parallel_size": tp,
+            "pipeline_model_parallel_size": pp,
+            "sequence_parallel": tp > 1,
+            "zero1": True,
+        },
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "ffn_hidden_size": 128,
+            "num_layers": 4, "num_attention_heads": 4,
+            "grad_clip": 1.0, "optim": {"lr": 1e-3, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(7)
+    mod = MegatronGPTModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=4)
+    g = torch.Generator().manual_seed(5)
+    last = None
+    for _ in range(2):
+        micros = [
+            {"input_ids": (ids := torch.randint(0, 128, (1, 32), generator=g)),
+             "labels": ids.clone()}
+            for _ in range(2)
+        ]
+        m = mod.training_step(micros)
+        last = m["reduced_train_loss"]
+    return last
+
+
+@pytest.mark.parametrize("world,tp", [(2, 1), (4, 2)])
+def test_megatron_pp2(world, tp):
+    ref = run_distributed(_mega_pp, 1, 1)[0]
+    res = run_distributed(_mega_pp, world, tp)
+    assert max(abs(x - res[0]) for x in res) < 1e-5, res
+    assert abs(ref - res[0]) < 5e-3, (ref, res[0])
