@@ -110,6 +110,38 @@ def token_unshuffle(y: torch.Tensor, ctx):
     return _route_tokens(y, perm, group, r)
 
 
+# backward scale for attached aux losses: the PP engine divides the main
+# loss by num_microbatches before backward, so attached aux grads must
+# carry the same factor (Megatron MoEAuxLossAutoScaler.set_loss_scale)
+_AUX_LOSS_SCALE = 1.0
+
+
+def set_aux_loss_scale(scale: float):
+    global _AUX_LOSS_SCALE
+    _AUX_LOSS_SCALE = float(scale)
+
+
+class _AttachAuxLoss(torch.autograd.Function):
+    """Route an auxiliary scalar's gradient through the activation chain
+    (Megatron MoEAuxLossAutoScaler pattern): forward passes x through
+    unchanged; backward hands the aux loss a unit gradient scaled by
+    `coeff`, so stages before the pipeline's loss stage still train their
+    routers under 1F1B."""
+
+    @staticmethod
+    def forward(ctx, x, aux, coeff):
+        ctx.coeff = coeff
+        return x
+
+    @staticmethod
+    def backward(ctx, dy):
+        return dy, dy.new_tensor(ctx.coeff * _AUX_LOSS_SCALE), None
+
+
+def attach_aux_loss(x: torch.Tensor, aux: torch.Tensor, coeff: float):
+    return _AttachAuxLoss.apply(x, aux, coeff)
+
+
 def load_balancing_loss_func(router_logits: torch.Tensor, num_experts: int,
                              top_k: int) -> torch.Tensor:
     """Switch-style aux loss: num_experts * sum_e f_e * P_e

@@ -4,6 +4,8 @@ from __future__ import annotations
 
 import torch
 
+from ..parallel import state as ps
+
 from .module import BaseModelModule
 from ..models.mixtral import MixtralConfig, MixtralForCausalLM
 
@@ -39,4 +41,8 @@ class MixtralModule(BaseModelModule):
                 self.cfg.get("distributed_strategy", {}).get(
                     "token_shuffle_group_size", 1)),
         )
+        if ps.get_pipeline_model_parallel_world_size() > 1:
+            from ..models.mixtral_pipeline import MixtralStage
+
+            return MixtralStage(cfg, pipeline_cuts=mcfg.get("pipeline_cuts"))
         return MixtralForCausalLM(cfg)

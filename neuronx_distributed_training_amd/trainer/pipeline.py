@@ -72,6 +72,9 @@ class PipelineEngine:
     def run_train(self, microbatches: List[Dict[str, torch.Tensor]]) -> torch.Tensor:
         """Full 1F1B over the microbatch list; returns mean loss (last
         stage; zeros elsewhere — caller broadcasts for logging)."""
+        from ..modules.moe import set_aux_loss_scale
+
+        set_aux_loss_scale(1.0 / max(len(microbatches), 1))
         M = len(microbatches)
         num_warmup = min(self.world - self.rank - 1, M)
         num_steady = M - num_warmup
@@ -187,6 +190,9 @@ class InterleavedPipelineEngine:
         self.pp_ranks = s.pp_ranks or [s.rank]
 
     def run_train(self, microbatches):
+        from ..modules.moe import set_aux_loss_scale
+
+        set_aux_loss_scale(1.0 / max(len(microbatches), 1))
         M = len(microbatches)
         pp, vp, rank = self.world, self.vp, self.rank
         assert M % pp == 0, f"num_microbatches {M} must divide by pp {pp}"
