@@ -3,9 +3,10 @@
 engine via transformer_layer_cls=ParallelTransformerLayer,
 megatron_gpt_model.py:67-77 + model/base.py:148-156).
 
-MoE layers with PP are not supported yet (the load-balancing aux loss
-needs router logits from every stage); dense megatron GPT pipelines
-fully.
+MoE layers ride PP via ``attach_aux_loss``: each stage's router aux
+losses enter backward through the activation chain (per-layer aux, vs
+the non-PP model's concatenated-logits aux — same scale, slightly
+different cross-layer coupling).
 """
 
 from __future__ import annotations
@@ -19,6 +20,7 @@ from ..parallel import state as ps
 from ..parallel.layers import ColumnParallelLinear
 from ..parallel.loss import parallel_cross_entropy
 from .llama_pipeline import partition_layers
+from ..modules.moe import attach_aux_loss, load_balancing_loss_func
 from .megatron_gpt import (
     Embedding, GPTConfig, ParallelTransformerLayer, _init, make_norm,
 )
@@ -33,8 +35,11 @@ class GPTStage(nn.Module):
 
     def __init__(self, cfg: GPTConfig, pipeline_cuts=None):
         super().__init__()
-        assert cfg.num_moe_experts == 0, "MoE + PP not supported yet"
         self.cfg = cfg
+        self.n_moe_total = (
+            sum(1 for i in range(cfg.num_layers) if i % cfg.moe_frequency == 0)
+            if cfg.num_moe_experts > 0 else 0
+        )
         pp = ps.get_pipeline_model_parallel_world_size()
         rank = ps.get_pipeline_model_parallel_rank()
         self.is_first = rank == 0
@@ -98,8 +103,15 @@ class GPTStage(nn.Module):
             x = self.embedding(
                 self._batch["input_ids"], self._batch.get("position_ids")
             )
+        coeff = (self.cfg.moe_aux_loss_coeff / self.n_moe_total
+                 if self.n_moe_total else 0.0)
         for layer in self.layers:
-            x, _ = layer(x, self.rope_cos, self.rope_sin, pos_offset)
+            x, logits = layer(x, self.rope_cos, self.rope_sin, pos_offset)
+            if logits is not None and self.training and coeff:
+                aux = load_balancing_loss_func(
+                    logits, self.cfg.num_moe_experts, self.cfg.moe_top_k
+                )
+                x = attach_aux_loss(x, aux, coeff)
         if not self.is_last:
             return x
         x = self.final_layernorm(x)

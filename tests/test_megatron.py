@@ -386,3 +386,52 @@ def test_megatron_pp2(world, tp):
     res = run_distributed(_mega_pp, world, tp)
     assert max(abs(x - res[0]) for x in res) < 1e-5, res
     assert abs(ref - res[0]) < 5e-3, (ref, res[0])
+
+
+def _mega_moe_pp(rank, world):
+    """Megatron MoE under PP=2: steps run, stage-0 routers get gradients."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module_megatron import (
+        MegatronGPTModule,
+    )
+
+    pp = 2 if world == 2 else 1
+    ps.initialize_model_parallel(pipeline_model_parallel_size=pp)
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 1, "seq_length": 32},
+        "distributed_strategy": {"pipeline_model_parallel_size": pp},
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "ffn_hidden_size": 128,
+            "num_layers": 4, "num_attention_heads": 4,
+            "grad_clip": 1.0, "optim": {"lr": 1e-3, "sched": {"warmup_steps": 1}},
+            "moe": {"num_experts": 4, "top_k": 2, "moe_frequency": 2},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(7)
+    mod = MegatronGPTModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=4)
+    g = torch.Generator().manual_seed(5)
+    micros = [
+        {"input_ids": (ids := torch.randint(0, 128, (1, 32), generator=g)),
+         "labels": ids.clone()}
+        for _ in range(2)
+    ]
+    m = mod.training_step(micros)
+    if pp > 1 and ps.get_pipeline_model_parallel_rank() == 0:
+        gsum = sum(
+            float(p.grad.abs().sum())
+            for n, p in mod.model.named_parameters()
+            if "router" in n and p.grad is not None
+        )
+        assert gsum > 0, "stage-0 megatron router got no grad"
+    return m["reduced_train_loss"]
+
+
+def test_megatron_moe_pp2():
+    ref = run_distributed(_mega_moe_pp, 1)[0]
+    res = run_distributed(_mega_moe_pp, 2)
+    assert max(abs(x - res[0]) for x in res) < 1e-5
+    assert abs(ref - res[0]) < 5e-2, (ref, res[0])
