@@ -33,19 +33,23 @@ class GPTStage(nn.Module):
     Tied embeddings register the shared weight on both end stages
     (grad all-reduce by PipelineEngine._sync_tied_embeddings)."""
 
-    def __init__(self, cfg: GPTConfig, pipeline_cuts=None):
+    def __init__(self, cfg: GPTConfig, pipeline_cuts=None, override=None):
         super().__init__()
         self.cfg = cfg
         self.n_moe_total = (
             sum(1 for i in range(cfg.num_layers) if i % cfg.moe_frequency == 0)
             if cfg.num_moe_experts > 0 else 0
         )
-        pp = ps.get_pipeline_model_parallel_world_size()
-        rank = ps.get_pipeline_model_parallel_rank()
-        self.is_first = rank == 0
-        self.is_last = rank == pp - 1
+        if override is not None:
+            # virtual-pipeline chunk: explicit (is_first, is_last, start, end)
+            self.is_first, self.is_last, start, end = override
+        else:
+            pp = ps.get_pipeline_model_parallel_world_size()
+            rank = ps.get_pipeline_model_parallel_rank()
+            self.is_first = rank == 0
+            self.is_last = rank == pp - 1
+            start, end = partition_layers(cfg.num_layers, pp, pipeline_cuts)[rank]
         self.dtype = cfg.torch_dtype
-        start, end = partition_layers(cfg.num_layers, pp, pipeline_cuts)[rank]
         self.layer_range = (start, end)
 
         if self.is_first or (self.is_last and cfg.share_embeddings_and_output_weights):
@@ -141,3 +145,21 @@ class GPTStage(nn.Module):
             denom = (denom if denom is not None else m.sum()).clamp(min=1)
             return (per_tok * m).sum() / denom
         return per_tok.mean()
+
+
+def build_virtual_chunks_gpt(cfg: GPTConfig, vp: int):
+    """This rank's vp GPTStage chunks for the interleaved schedule."""
+    assert not cfg.share_embeddings_and_output_weights, \
+        "tied embeddings + interleaved VP not supported (untie the head)"
+    pp = ps.get_pipeline_model_parallel_world_size()
+    rank = ps.get_pipeline_model_parallel_rank()
+    n_virtual = pp * vp
+    ranges = partition_layers(cfg.num_layers, n_virtual)
+    chunks = nn.ModuleList()
+    for c in range(vp):
+        v = c * pp + rank
+        start, end = ranges[v]
+        chunks.append(
+            GPTStage(cfg, override=(v == 0, v == n_virtual - 1, start, end))
+        )
+    return chunks

@@ -29,16 +29,21 @@ from .mixtral import MixtralConfig, MixtralDecoderLayer
 
 
 class MixtralStage(nn.Module):
-    def __init__(self, cfg: MixtralConfig, pipeline_cuts=None):
+    def __init__(self, cfg: MixtralConfig, pipeline_cuts=None, override=None):
         super().__init__()
         self.cfg = cfg
-        pp = ps.get_pipeline_model_parallel_world_size()
-        rank = ps.get_pipeline_model_parallel_rank()
-        self.is_first = rank == 0
-        self.is_last = rank == pp - 1
+        if override is not None:
+            self.is_first, self.is_last, start, end = override
+        else:
+            pp = ps.get_pipeline_model_parallel_world_size()
+            rank = ps.get_pipeline_model_parallel_rank()
+            self.is_first = rank == 0
+            self.is_last = rank == pp - 1
+            start, end = partition_layers(
+                cfg.num_hidden_layers, pp, pipeline_cuts
+            )[rank]
         dt = cfg.torch_dtype
         self.dtype = dt
-        start, end = partition_layers(cfg.num_hidden_layers, pp, pipeline_cuts)[rank]
         self.layer_range = (start, end)
         self.n_layers_total = cfg.num_hidden_layers
 
@@ -128,3 +133,21 @@ class MixtralStage(nn.Module):
             denom = (denom if denom is not None else m.sum()).clamp(min=1)
             return (per_tok * m).sum() / denom
         return per_tok.mean()
+
+
+def build_virtual_chunks_mixtral(cfg: MixtralConfig, vp: int):
+    """This rank's vp MixtralStage chunks for the interleaved schedule."""
+    assert not cfg.tie_word_embeddings, \
+        "tied embeddings + interleaved VP not supported"
+    pp = ps.get_pipeline_model_parallel_world_size()
+    rank = ps.get_pipeline_model_parallel_rank()
+    n_virtual = pp * vp
+    ranges = partition_layers(cfg.num_hidden_layers, n_virtual)
+    chunks = nn.ModuleList()
+    for c in range(vp):
+        v = c * pp + rank
+        start, end = ranges[v]
+        chunks.append(
+            MixtralStage(cfg, override=(v == 0, v == n_virtual - 1, start, end))
+        )
+    return chunks

@@ -77,9 +77,12 @@ class MegatronGPTModule(BaseModelModule):
         )
         if ps.get_pipeline_model_parallel_world_size() > 1:
             vp = int(dstr.get("virtual_pipeline_model_parallel_size", 1) or 1)
-            assert vp == 1, "interleaved VP not wired for megatron GPT yet"
-            from ..models.megatron_pipeline import GPTStage
+            from ..models.megatron_pipeline import (
+                GPTStage, build_virtual_chunks_gpt,
+            )
 
+            if vp > 1:
+                return build_virtual_chunks_gpt(cfg, vp)
             return GPTStage(cfg, pipeline_cuts=mcfg.get("pipeline_cuts"))
         return GPTModel(cfg)
 

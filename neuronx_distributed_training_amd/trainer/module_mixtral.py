@@ -42,7 +42,13 @@ class MixtralModule(BaseModelModule):
                     "token_shuffle_group_size", 1)),
         )
         if ps.get_pipeline_model_parallel_world_size() > 1:
-            from ..models.mixtral_pipeline import MixtralStage
+            dstr = self.cfg.get("distributed_strategy", {})
+            vp = int(dstr.get("virtual_pipeline_model_parallel_size", 1) or 1)
+            from ..models.mixtral_pipeline import (
+                MixtralStage, build_virtual_chunks_mixtral,
+            )
 
+            if vp > 1:
+                return build_virtual_chunks_mixtral(cfg, vp)
             return MixtralStage(cfg, pipeline_cuts=mcfg.get("pipeline_cuts"))
         return MixtralForCausalLM(cfg)

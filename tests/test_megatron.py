@@ -435,3 +435,48 @@ def test_megatron_moe_pp2():
     res = run_distributed(_mega_moe_pp, 2)
     assert max(abs(x - res[0]) for x in res) < 1e-5
     assert abs(ref - res[0]) < 5e-2, (ref, res[0])
+
+
+def _mega_vp(rank, world):
+    """Megatron GPT under PP2×VP2 (untied head): loss matches single-rank."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module_megatron import (
+        MegatronGPTModule,
+    )
+
+    pp = 2 if world == 2 else 1
+    ps.initialize_model_parallel(pipeline_model_parallel_size=pp)
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 1, "seq_length": 32},
+        "distributed_strategy": {
+            "pipeline_model_parallel_size": pp,
+            "virtual_pipeline_model_parallel_size": 2 if pp > 1 else 1,
+        },
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "ffn_hidden_size": 128,
+            "num_layers": 4, "num_attention_heads": 4,
+            "share_embeddings_and_output_weights": False,
+            "grad_clip": 1.0, "optim": {"lr": 1e-3, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(7)
+    mod = MegatronGPTModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=4)
+    g = torch.Generator().manual_seed(5)
+    micros = [
+        {"input_ids": (ids := torch.randint(0, 128, (1, 32), generator=g)),
+         "labels": ids.clone()}
+        for _ in range(2)
+    ]
+    m = mod.training_step(micros)
+    return m["reduced_train_loss"]
+
+
+def test_megatron_vp2():
+    ref = run_distributed(_mega_vp, 1)[0]
+    res = run_distributed(_mega_vp, 2)
+    assert max(abs(x - res[0]) for x in res) < 1e-5
+    assert abs(ref - res[0]) < 5e-3, (ref, res[0])
