@@ -103,6 +103,8 @@ class LlamaChunk(nn.Module):
 
 def build_virtual_chunks(cfg: LlamaConfig, vp: int):
     """ModuleList of this rank's vp model chunks (interleaved VP)."""
+    assert not cfg.tie_word_embeddings, \
+        "tied embeddings + interleaved VP not supported (untie the head)"
     pp = ps.get_pipeline_model_parallel_world_size()
     rank = ps.get_pipeline_model_parallel_rank()
     n_virtual = pp * vp
