@@ -44,6 +44,13 @@ class DPOBaseModel(LlamaModule):
 
     def __init__(self, cfg: Dict):
         super().__init__(cfg)
+        dstr = cfg.get("distributed_strategy", {})
+        if int(dstr.get("pipeline_model_parallel_size", 1) or 1) > 1 or \
+                int(dstr.get("context_parallel_size", 1) or 1) > 1:
+            raise ValueError(
+                "DPO/ORPO run with TP/DP only (sequence logprobs need the "
+                "full model on each PP stage and the full sequence per rank)"
+            )
         align = cfg.get("model_alignment_strategy", {})
         dcfg = align.get("dpo") or align.get("orpo") or {}
         self.kl_beta = float(dcfg.get("kl_beta", dcfg.get("beta", 0.1)))

@@ -148,3 +148,28 @@ def test_kv_cache_split_invariance(s, split):
         a = model(ids[:, :split], kv_cache=cache)
         b = model(ids[:, split:], kv_cache=cache)
     assert torch.allclose(torch.cat([a, b], 1), full, atol=1e-5)
+
+
+@settings(max_examples=20, deadline=None)
+@given(
+    s_loc=st.integers(min_value=2, max_value=24),
+    chunks=st.integers(min_value=1, max_value=6),
+)
+def test_sp_overlap_divisibility_gate(s_loc, chunks):
+    """The overlap path only engages when the shard divides by the chunk
+    count; otherwise layers silently fall back to the plain path (single
+    process: world==1 also gates it off). Forward must always succeed."""
+    import torch
+    import neuronx_distributed_training_amd.parallel.layers as L
+    from neuronx_distributed_training_amd.parallel import state as ps
+
+    ps.destroy_model_parallel()
+    old = L._SP_OVERLAP_CHUNKS
+    L._SP_OVERLAP_CHUNKS = chunks
+    try:
+        col = L.ColumnParallelLinear(8, 16, sequence_parallel=True, init_seed=1)
+        x = torch.randn(s_loc, 2, 8)
+        y = col(x)
+        assert y.shape == (s_loc, 2, 16)
+    finally:
+        L._SP_OVERLAP_CHUNKS = old
