@@ -139,3 +139,56 @@ def test_zero1_overlap_matches_plain():
     o2 = run_distributed(_zero1_overlap, 2)
     assert torch.allclose(o1[0], o2[0], atol=1e-5), (o1[0] - o2[0]).abs().max()
     assert torch.allclose(o2[0], o2[1])
+
+
+def _overlap_tp_dp(rank, world, overlap):
+    """DP2×TP2+SP with overlap_grad_reduce: weights after 2 steps match the
+    non-overlapped path (tagged SP-norm TP sum composes with the bucketed
+    DP all-reduce)."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=2)
+    cfg = {
+        "data": {"global_batch_size": 4, "micro_batch_size": 1, "seq_length": 32},
+        "distributed_strategy": {
+            "tensor_model_parallel_size": 2,
+            "sequence_parallel": True,
+            "zero1": True,
+            "overlap_grad_reduce": overlap,
+        },
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 2, "num_attention_heads": 4, "num_kv_heads": 2,
+            "grad_clip": 1.0, "optim": {"lr": 1e-2, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(3)
+    mod = LlamaModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=4)
+    dp = ps.get_data_parallel_world_size()
+    r = ps.get_data_parallel_rank()
+    g = torch.Generator().manual_seed(5)
+    for _ in range(2):
+        glob = [
+            {"input_ids": (ids := torch.randint(0, 128, (1, 32), generator=g)),
+             "labels": ids.clone()}
+            for _ in range(4)
+        ]
+        per = 4 // dp
+        mod.training_step(glob[r * per : (r + 1) * per])
+    w = mod.model.model.layers[0].input_layernorm.weight.detach().clone()
+    q = mod.model.model.layers[0].self_attn.o_proj.weight.detach()[:, :4].clone()
+    if ps.get_tensor_model_parallel_rank() == 0 and r == 0:
+        return w, q
+    return None
+
+
+def test_overlap_grad_reduce_tp2_dp2():
+    plain = [x for x in run_distributed(_overlap_tp_dp, 4, False) if x is not None][0]
+    over = [x for x in run_distributed(_overlap_tp_dp, 4, True) if x is not None][0]
+    assert torch.allclose(plain[0], over[0], atol=1e-5), (plain[0] - over[0]).abs().max()
+    assert torch.allclose(plain[1], over[1], atol=1e-5), (plain[1] - over[1]).abs().max()
