@@ -279,3 +279,26 @@ def _save_bf16(rank, world, tmpdir):
 
 def test_save_bf16_checkpoint(tmp_path):
     run_distributed(_save_bf16, 1, str(tmp_path))
+
+
+def _fit_max_time(rank, world, tmpdir):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.trainer import Trainer
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+    from neuronx_distributed_training_amd.data.datamodule import build_datamodule
+
+    ps.initialize_model_parallel()
+    torch.manual_seed(0)
+    cfg = _cfg(tmpdir, 10000)
+    cfg["trainer"]["max_time"] = "00:00:00:01"  # DD:HH:MM:SS = 1 second
+    tr = Trainer(cfg)
+    assert tr.max_time_s == 1.0
+    module = LlamaModule(cfg)
+    dm = build_datamodule(cfg)
+    tr.fit(module, dm)
+    return tr.global_step
+
+
+def test_max_time_budget(tmp_path):
+    steps = run_distributed(_fit_max_time, 1, str(tmp_path))[0]
+    assert 0 < steps < 10000
