@@ -103,12 +103,15 @@ class BaseDataModule:
     def val_dataloader(self):
         return self._loader(self.val_ds, shuffle=False) if self.val_ds else None
 
-    def microbatch_iterator(self, loader_iter) -> Iterator:
-        """Yield num_microbatches microbatches = one global batch."""
+    def microbatch_iterator(self, loader_iter):
+        """num_microbatches microbatches = one global batch. A plain
+        function (NOT a generator) so the StopIteration of an exhausted
+        epoch propagates to the trainer's epoch-wrap handler instead of
+        becoming PEP-479 RuntimeError."""
         n_micro = self.per_rank_batch // self.micro_batch_size
-        for _ in range(n_micro):
-            yield next(loader_iter)
+        out = [next(loader_iter) for _ in range(n_micro)]
         self.consumed_samples += self.global_batch_size
+        return out
 
 
 class SyntheticDataModule(BaseDataModule):
