@@ -91,10 +91,15 @@ class Trainer:
 
         loader = datamodule.train_dataloader()
         it = iter(loader)
+        epoch = 0
         while self.global_step < self.max_steps:
             try:
                 micro = list(datamodule.microbatch_iterator(it))
             except StopIteration:
+                epoch += 1
+                sampler = getattr(loader, "sampler", None)
+                if hasattr(sampler, "set_epoch"):
+                    sampler.set_epoch(epoch)  # fresh DP shuffle each epoch
                 it = iter(loader)
                 micro = list(datamodule.microbatch_iterator(it))
             metrics = module.training_step(micro)
