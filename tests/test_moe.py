@@ -349,3 +349,34 @@ def test_mixtral_pp2():
     # PP reports the CE loss only (aux enters grads via attach_aux_loss);
     # single-rank reports CE + 0.02*aux — difference is exactly the aux term
     assert abs(ref - res[0]) < 4e-2, (ref, res[0])
+
+
+def test_attach_aux_loss_gradient_semantics():
+    """attach_aux_loss(x, aux, c) must make backward equivalent to adding
+    c·aux·(scale) to the loss (scale = the PP engine's 1/num_microbatches)."""
+    from neuronx_distributed_training_amd.modules.moe import (
+        attach_aux_loss, set_aux_loss_scale,
+    )
+
+    torch.manual_seed(0)
+    w = torch.randn(4, 4, requires_grad=True)
+    x0 = torch.randn(3, 4)
+
+    def path(attach, scale):
+        wl = w.detach().clone().requires_grad_(True)
+        h = x0 @ wl
+        aux = h.square().mean()
+        if attach:
+            set_aux_loss_scale(scale)
+            h = attach_aux_loss(h, aux, 0.5)
+            loss = h.sum() * scale
+        else:
+            loss = (h.sum() + 0.5 * aux) * scale
+        loss.backward()
+        set_aux_loss_scale(1.0)
+        return wl.grad.clone()
+
+    for scale in (1.0, 0.25):
+        ga = path(True, scale)
+        gr = path(False, scale)
+        assert torch.allclose(ga, gr, atol=1e-6), (ga - gr).abs().max()
