@@ -480,3 +480,44 @@ def test_megatron_vp2():
     res = run_distributed(_mega_vp, 2)
     assert max(abs(x - res[0]) for x in res) < 1e-5
     assert abs(ref - res[0]) < 5e-3, (ref, res[0])
+
+
+def _variant_grads(rank, world, block, pos, act):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.parallel.layers import (
+        allreduce_sequence_parallel_grads,
+    )
+    from neuronx_distributed_training_amd.models.megatron_gpt import (
+        GPTConfig, GPTModel,
+    )
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(0)
+    cfg = GPTConfig(vocab_size=64, hidden_size=32, ffn_hidden_size=64,
+                    num_layers=2, num_attention_heads=4,
+                    max_position_embeddings=32,
+                    transformer_block_type=block,
+                    position_embedding_type=pos, activation=act,
+                    normalization="layernorm" if block == "post_ln" else "rmsnorm",
+                    sequence_parallel=world > 1)
+    m = GPTModel(cfg)
+    ids = torch.randint(0, 64, (2, 16), generator=torch.Generator().manual_seed(1))
+    m(ids, labels=ids.clone()).backward()
+    allreduce_sequence_parallel_grads(m)
+    g = m.final_layernorm.weight.grad.detach().clone()
+    g2 = m.layers[0].mlp.dense_4h_to_h.weight.grad[:, :4].detach().clone()
+    return (g, g2) if ps.get_tensor_model_parallel_rank() == 0 else None
+
+
+@pytest.mark.parametrize("block,pos,act", [
+    ("post_ln", "learned_absolute", "gelu"),
+    ("normformer", "rope", "geglu"),
+])
+def test_block_variant_sp_grads_exact(block, pos, act):
+    """Every block type / position type / activation keeps exact grads
+    under TP2+SP (LayerNorm params tag correctly, not just RMSNorm)."""
+    a1 = run_distributed(_variant_grads, 1, block, pos, act)[0]
+    a2 = [r for r in run_distributed(_variant_grads, 2, block, pos, act)
+          if r is not None][0]
+    for x, y in zip(a1, a2):
+        assert torch.allclose(x, y, atol=1e-4), (x - y).abs().max()
