@@ -2,6 +2,8 @@
 
 import os
 
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
 import torch
 import pytest
 
@@ -49,3 +51,29 @@ def test_hf_datamodule_dp2(tmp_path):
     path = os.path.join(str(tmp_path), "ds")
     _make_ds(path)
     run_distributed(_hf_loader, 2, path)
+
+
+def test_preprocess_data_cli(tmp_path):
+    """tools/preprocess_data.py: jsonl → MMIDIDX round trip."""
+    import json
+    import subprocess
+    import sys
+
+    src = tmp_path / "corpus.jsonl"
+    with open(src, "w") as f:
+        for i in range(5):
+            f.write(json.dumps({"text": f"document number {i} " * 3}) + "\n")
+    prefix = str(tmp_path / "corpus_text_document")
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tools", "preprocess_data.py"),
+         "--input", str(src), "--output-prefix", prefix, "--append-eod"],
+        capture_output=True, text=True, timeout=300, cwd=REPO,
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    from neuronx_distributed_training_amd.data.indexed_dataset import (
+        MMapIndexedDataset,
+    )
+
+    ds = MMapIndexedDataset(prefix)
+    assert len(ds) == 5
+    assert len(ds[0]) > 5
