@@ -67,9 +67,13 @@ class LoRAColumnParallelLinear(nn.Module):
         if isinstance(base, ColumnParallelLinear):
             self.lora_B.tensor_model_parallel = True
             self.lora_B.partition_dim = 0
+            # A is replicated but each rank's grad only carries its B
+            # shard's contribution → optimizer must SUM over TP
+            self.lora_A.tensor_parallel_grad_sum = True
         else:
             self.lora_A.tensor_model_parallel = True
             self.lora_A.partition_dim = 1
+            self.lora_B.tensor_parallel_grad_sum = True
 
     @property
     def sequence_parallel(self):
@@ -77,7 +81,6 @@ class LoRAColumnParallelLinear(nn.Module):
 
     def forward(self, x, pre_mapped: bool = False):
         b = self.base
-        delta_in = self.dropout(x)
         if isinstance(b, ColumnParallelLinear):
             if pre_mapped:
                 xg = x
@@ -86,8 +89,9 @@ class LoRAColumnParallelLinear(nn.Module):
             else:
                 xg = copy_to_tensor_model_parallel_region(x)
             out = F.linear(xg, b.weight, b.bias)
-            if not pre_mapped and b.sequence_parallel:
-                delta_in = gather_from_sequence_parallel_region(delta_in)
+            # delta rides the SAME mapped input so its dx contribution is
+            # reduced over TP exactly once in backward
+            delta_in = self.dropout(xg)
             out = out + F.linear(F.linear(delta_in, self.lora_A), self.lora_B) * self.scaling
             if b.gather_output:
                 from ..parallel.mappings import gather_from_tensor_model_parallel_region
@@ -95,7 +99,7 @@ class LoRAColumnParallelLinear(nn.Module):
             return out
         # RowParallel: add the low-rank partial before the reduction
         out = F.linear(x, b.weight)
-        out = out + F.linear(F.linear(delta_in, self.lora_A), self.lora_B) * self.scaling
+        out = out + F.linear(F.linear(self.dropout(x), self.lora_A), self.lora_B) * self.scaling
         if b.sequence_parallel:
             out = reduce_scatter_to_sequence_parallel_region(out)
         else:

@@ -131,7 +131,9 @@ class ZeRO1AdamW:
         self._sp_ranges = []
         lo, hi = self.shard_start, self.shard_start + self.shard_size
         for (n, p), o in zip(self.named_params, offsets):
-            if getattr(p, "sequence_parallel_enabled", False):
+            if getattr(p, "sequence_parallel_enabled", False) or getattr(
+                p, "tensor_parallel_grad_sum", False
+            ):
                 a, b = max(o, lo), min(o + p.numel(), hi)
                 if a < b:
                     self._sp_ranges.append((a - lo, b - lo))

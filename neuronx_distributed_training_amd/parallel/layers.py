@@ -225,7 +225,10 @@ def allreduce_sequence_parallel_grads(module):
     if ps.get_tensor_model_parallel_world_size() == 1:
         return
     for p in module.parameters():
-        if getattr(p, "sequence_parallel_enabled", False) and p.grad is not None:
+        if (
+            getattr(p, "sequence_parallel_enabled", False)
+            or getattr(p, "tensor_parallel_grad_sum", False)
+        ) and p.grad is not None:
             dist.all_reduce(p.grad, group=ps.get_tensor_model_parallel_group())
 
 

@@ -199,3 +199,63 @@ def test_dpo_tp2(tmp_path):
     l1 = run_distributed(_dpo_tp_run, 1, str(tmp_path), "dpo")[0]
     l2 = run_distributed(_dpo_tp_run, 2, str(tmp_path), "dpo")
     assert abs(l1 - l2[0]) < 5e-3, (l1, l2[0])
+
+
+def _lora_grad_exact(rank, world, sp):
+    """LoRA factor grads are EXACT vs TP=1 (regression: replicated A of
+    Column wraps / B of Row wraps lacked the TP grad sum; the Column delta
+    path bypassed the input mapping)."""
+    import zlib
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.parallel.layers import (
+        allreduce_sequence_parallel_grads,
+    )
+    from neuronx_distributed_training_amd.models.llama import (
+        LlamaConfig, LlamaForCausalLM,
+    )
+    from neuronx_distributed_training_amd.modules.lora import (
+        LoraConfig, apply_lora,
+    )
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(3)
+    cfg = LlamaConfig(vocab_size=128, hidden_size=64, intermediate_size=128,
+                      num_hidden_layers=2, num_attention_heads=4,
+                      num_key_value_heads=2, max_position_embeddings=32,
+                      sequence_parallel=sp and world > 1)
+    m = LlamaForCausalLM(cfg)
+    apply_lora(m, LoraConfig(lora_rank=4, lora_dropout=0.0,
+                             target_modules=["q_proj", "o_proj"]))
+    # deterministic A/B from shared full matrices (B starts zero, which
+    # would mask the bugs; shard-slice so TP2 == TP1 model identity)
+    r0 = ps.get_tensor_model_parallel_rank()
+    for name, mod in m.named_modules():
+        if not hasattr(mod, "lora_B"):
+            continue
+        gb = torch.Generator().manual_seed(zlib.crc32((name + "/B").encode()) % 10**8)
+        ga = torch.Generator().manual_seed(zlib.crc32((name + "/A").encode()) % 10**8)
+        base = mod.base
+        if hasattr(base, "output_size_per_partition"):  # Column wrap
+            fullB = torch.randn(base.output_size, mod.lora_B.shape[1], generator=gb)
+            sh = base.output_size_per_partition
+            mod.lora_B.data.copy_(fullB[r0 * sh:(r0 + 1) * sh] * 0.05)
+            mod.lora_A.data.copy_(torch.randn(*mod.lora_A.shape, generator=ga) * 0.1)
+        else:  # Row wrap
+            mod.lora_B.data.copy_(torch.randn(*mod.lora_B.shape, generator=gb) * 0.05)
+            fullA = torch.randn(mod.lora_A.shape[0], base.input_size, generator=ga)
+            shi = base.input_size_per_partition
+            mod.lora_A.data.copy_(fullA[:, r0 * shi:(r0 + 1) * shi] * 0.1)
+    ids = torch.randint(0, 128, (2, 32), generator=torch.Generator().manual_seed(1))
+    m(ids, labels=ids).backward()
+    allreduce_sequence_parallel_grads(m)
+    qa = m.model.layers[0].self_attn.q_proj.lora_A.grad
+    ob = m.model.layers[0].self_attn.o_proj.lora_B.grad
+    return (qa.detach().clone(), ob.detach().clone()) if r0 == 0 else None
+
+
+@pytest.mark.parametrize("sp", [False, True])
+def test_lora_grads_exact_vs_tp1(sp):
+    a1 = run_distributed(_lora_grad_exact, 1, sp)[0]
+    a2 = [r for r in run_distributed(_lora_grad_exact, 2, sp) if r is not None][0]
+    assert torch.allclose(a1[0], a2[0], atol=1e-5), (a1[0] - a2[0]).abs().max()
+    assert torch.allclose(a1[1], a2[1], atol=1e-5), (a1[1] - a2[1]).abs().max()
