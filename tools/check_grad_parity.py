@@ -64,7 +64,8 @@ def run(rank: int, world: int, tp: int = 1, sp: bool = False, qkv: bool = False)
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--layout", default="tp2_sp", choices=sorted(LAYOUTS))
+    ap.add_argument("--layout", default="tp2_sp",
+                choices=sorted(LAYOUTS) + ["lora_tp2_sp"])
     ap.add_argument("--dist", action="store_true",
                     help="already inside torch.distributed.run")
     args = ap.parse_args()
@@ -90,6 +91,20 @@ def main():
 
     # CPU path: spawn both runs via the test helper
     from tests.distutils import run_distributed
+
+    if args.layout == "lora_tp2_sp":
+        from tests.test_alignment import _lora_grad_exact
+
+        a1 = run_distributed(_lora_grad_exact, 1, True)[0]
+        a2 = [r for r in run_distributed(_lora_grad_exact, 2, True)
+              if r is not None][0]
+        ok = True
+        for name, x, y in (("lora_A", a1[0], a2[0]), ("lora_B", a1[1], a2[1])):
+            d = float((x - y).abs().max())
+            ok &= d < 1e-4
+            print(f"lora_tp2_sp {name}: maxdiff={d:.6f}",
+                  "OK" if d < 1e-4 else "MISMATCH")
+        sys.exit(0 if ok else 1)
 
     ref = run_distributed(run, 1, 1, False, lay.get("qkv", False))[0]
     world = lay["tp"]
