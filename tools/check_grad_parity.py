@@ -69,7 +69,7 @@ def main():
     ap.add_argument("--dist", action="store_true",
                     help="already inside torch.distributed.run")
     args = ap.parse_args()
-    lay = LAYOUTS[args.layout]
+    lay = LAYOUTS.get(args.layout, {})
 
     if args.dist:
         import torch.distributed as dist
