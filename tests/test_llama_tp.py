@@ -281,3 +281,40 @@ def test_sp_norm_weights_match_after_training():
     w1 = run_distributed(_sp_train_weights, 1)[0]
     w2 = [w for w in run_distributed(_sp_train_weights, 2) if w is not None][0]
     assert torch.allclose(w1, w2, atol=1e-4), (w1 - w2).abs().max()
+
+
+def _dpo_logprob_grads(rank, world):
+    """Grads through from_parallel_logits_to_logprobs under TP2+SP are
+    exact (the DPO/ORPO backward path)."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.parallel.layers import (
+        allreduce_sequence_parallel_grads,
+    )
+    from neuronx_distributed_training_amd.parallel.loss import (
+        from_parallel_logits_to_logprobs,
+    )
+    from neuronx_distributed_training_amd.models.llama import (
+        LlamaConfig, LlamaForCausalLM,
+    )
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(3)
+    cfg = LlamaConfig(**TINY, sequence_parallel=world > 1)
+    m = LlamaForCausalLM(cfg)
+    ids = torch.randint(0, 128, (2, 32), generator=torch.Generator().manual_seed(1))
+    hidden = m.model(ids)
+    logits = m.lm_head(hidden, pre_mapped=cfg.sequence_parallel)
+    from_parallel_logits_to_logprobs(logits, ids.clamp(min=0)).sum().backward()
+    allreduce_sequence_parallel_grads(m)
+    og = m.model.layers[0].self_attn.o_proj.weight.grad[:, :8]
+    ng = m.model.norm.weight.grad
+    if ps.get_tensor_model_parallel_rank() == 0:
+        return og.detach().clone(), ng.detach().clone()
+    return None
+
+
+def test_dpo_logprob_grads_exact():
+    a1 = run_distributed(_dpo_logprob_grads, 1)[0]
+    a2 = [r for r in run_distributed(_dpo_logprob_grads, 2) if r is not None][0]
+    for x, y in zip(a1, a2):
+        assert torch.allclose(x, y, atol=1e-4), (x - y).abs().max()
