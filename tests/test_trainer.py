@@ -302,3 +302,37 @@ def _fit_max_time(rank, world, tmpdir):
 def test_max_time_budget(tmp_path):
     steps = run_distributed(_fit_max_time, 1, str(tmp_path))[0]
     assert 0 < steps < 10000
+
+
+def _autocast_tp2(rank, world):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 2, "seq_length": 32},
+        "distributed_strategy": {"tensor_model_parallel_size": world,
+                                 "sequence_parallel": world > 1},
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 2, "num_attention_heads": 4, "num_kv_heads": 2,
+            "grad_clip": 1.0, "optim": {"lr": 1e-3, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "autocast"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(3)
+    mod = LlamaModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=4)
+    ids = torch.randint(0, 128, (2, 32), generator=torch.Generator().manual_seed(5))
+    m = mod.training_step([{"input_ids": ids, "labels": ids.clone()}])
+    return m["reduced_train_loss"]
+
+
+def test_autocast_tp2_sp():
+    """bf16 autocast composes with TP2+SP collectives."""
+    l1 = run_distributed(_autocast_tp2, 1)[0]
+    l2 = run_distributed(_autocast_tp2, 2)
+    assert abs(l2[0] - l2[1]) < 1e-5
+    assert abs(l1 - l2[0]) < 0.05, (l1, l2[0])
