@@ -55,6 +55,7 @@ class LlamaConfig:
     rms_norm_eps: float = 1e-5
     rope_theta: float = 500000.0
     rope_scaling: Optional[dict] = None
+    sliding_window: Optional[int] = None   # Mistral-style windowed attention
     tie_word_embeddings: bool = False
     initializer_range: float = 0.02
     # framework knobs (reference hf_llama3_8B_config.yaml:46-83)

@@ -320,6 +320,7 @@ class LlamaModule(BaseModelModule):
             kv_replicator=int(mcfg.get("kv_replicator", 1)),
             fuse_qkv=bool(mcfg.get("fuse_qkv", True)),
             activation_checkpoint=mcfg.get("activation_checkpoint"),
+            sliding_window=mcfg.get("sliding_window"),
             dtype=dtype,
         )
         if ps.get_pipeline_model_parallel_world_size() > 1:
