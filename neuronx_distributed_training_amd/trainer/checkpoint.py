@@ -59,6 +59,32 @@ def _cpu_copy(obj):
     return obj
 
 
+def _remote_fs(path: str):
+    """fsspec filesystem for scheme-qualified paths (s3://, gcs://,
+    memory:// in tests) — the reference's "dump a checkpoint to S3"
+    capability (features.rst checkpointing list). None for local paths.
+    Remote checkpoints skip atomic-rename and top-k pruning."""
+    if "://" not in str(path) or str(path).startswith("file://"):
+        return None
+    import fsspec
+
+    return fsspec.filesystem(str(path).split("://", 1)[0])
+
+
+def _join(root: str, fs, *parts: str) -> str:
+    if fs is not None:
+        return "/".join([root.rstrip("/"), *parts])
+    return os.path.join(root, *parts)
+
+
+def _touch_done(root: str, fs):
+    if fs is not None:
+        with fs.open(_join(root, fs, "done"), "wb") as f:
+            f.write(b"")
+    else:
+        open(os.path.join(root, "done"), "w").close()
+
+
 class CheckpointIO:
     def __init__(self, async_save: bool = False, save_bf16: bool = False):
         self.async_save = async_save
@@ -76,9 +102,15 @@ class CheckpointIO:
         user_content: Dict,
         keep_top_k: int = 0,
     ):
-        root = os.path.join(ckpt_dir, f"{tag}.ckpt")
-        os.makedirs(os.path.join(root, "model"), exist_ok=True)
-        os.makedirs(os.path.join(root, "optim"), exist_ok=True)
+        fs = _remote_fs(ckpt_dir)
+        if fs is not None:
+            root = f"{ckpt_dir.rstrip('/')}/{tag}.ckpt"
+            fs.makedirs(f"{root}/model", exist_ok=True)
+            fs.makedirs(f"{root}/optim", exist_ok=True)
+        else:
+            root = os.path.join(ckpt_dir, f"{tag}.ckpt")
+            os.makedirs(os.path.join(root, "model"), exist_ok=True)
+            os.makedirs(os.path.join(root, "optim"), exist_ok=True)
 
         dp, tp, pp = _rank_tags()
         work = []
@@ -92,11 +124,11 @@ class CheckpointIO:
                     for k, v in msd.items()
                 }
             work.append(
-                (os.path.join(root, "model", _model_shard_name()), msd)
+                (_join(root, fs, "model", _model_shard_name()), msd)
             )
         if module.optimizer is not None and ps.get_context_model_parallel_rank() == 0:
             work.append(
-                (os.path.join(root, "optim", _optim_shard_name()),
+                (_join(root, fs, "optim", _optim_shard_name()),
                  _cpu_copy(module.optimizer.state_dict()))
             )
         is_global_zero = (not dist.is_initialized()) or dist.get_rank() == 0
@@ -104,13 +136,17 @@ class CheckpointIO:
             uc = dict(user_content)
             if module.scheduler is not None:
                 uc["scheduler"] = module.scheduler.state_dict()
-            work.append((os.path.join(root, "user_content.pt"), uc))
+            work.append((_join(root, fs, "user_content.pt"), uc))
 
         def _write():
             for path, obj in work:
-                tmp = path + ".tmp"
-                torch.save(obj, tmp)
-                os.replace(tmp, path)
+                if fs is not None:
+                    with fs.open(path, "wb") as f:
+                        torch.save(obj, f)
+                else:
+                    tmp = path + ".tmp"
+                    torch.save(obj, tmp)
+                    os.replace(tmp, path)
 
         if self.async_save:
             t = threading.Thread(target=_write, daemon=False)
@@ -122,15 +158,15 @@ class CheckpointIO:
         if dist.is_initialized():
             dist.barrier()
         if is_global_zero and not self.async_save:
-            open(os.path.join(root, "done"), "w").close()
-            if keep_top_k:
+            _touch_done(root, fs)
+            if keep_top_k and fs is None:
                 self._prune(ckpt_dir, keep_top_k)
         elif is_global_zero and self.async_save:
             def _commit(threads=list(self._pending), root=root, k=keep_top_k, d=ckpt_dir):
                 for t in threads:
                     t.join()
-                open(os.path.join(root, "done"), "w").close()
-                if k:
+                _touch_done(root, fs)
+                if k and fs is None:
                     self._prune(d, k)
             tc = threading.Thread(target=_commit, daemon=False)
             tc.start()
@@ -154,18 +190,27 @@ class CheckpointIO:
     # ---- load ----
     def load(self, path: str, module, weight_init_only: bool = False) -> Dict:
         """path: .../<tag>.ckpt directory. Returns user_content."""
-        model_path = os.path.join(path, "model", _model_shard_name())
-        sd = torch.load(model_path, map_location="cpu", weights_only=False)
+        fs = _remote_fs(path)
+
+        def _ld(p):
+            if fs is not None:
+                with fs.open(p, "rb") as f:
+                    return torch.load(f, map_location="cpu", weights_only=False)
+            return torch.load(p, map_location="cpu", weights_only=False)
+
+        def _exists(p):
+            return fs.exists(p) if fs is not None else os.path.exists(p)
+
+        sd = _ld(_join(path, fs, "model", _model_shard_name()))
         module.model.load_state_dict(sd)
         if not weight_init_only and module.optimizer is not None:
-            opath = os.path.join(path, "optim", _optim_shard_name())
-            if os.path.exists(opath):
-                osd = torch.load(opath, map_location="cpu", weights_only=False)
-                module.optimizer.load_state_dict(osd)
-        uc_path = os.path.join(path, "user_content.pt")
+            opath = _join(path, fs, "optim", _optim_shard_name())
+            if _exists(opath):
+                module.optimizer.load_state_dict(_ld(opath))
+        uc_path = _join(path, fs, "user_content.pt")
         uc = {}
-        if os.path.exists(uc_path):
-            uc = torch.load(uc_path, map_location="cpu", weights_only=False)
+        if _exists(uc_path):
+            uc = _ld(uc_path)
             if uc.get("scheduler") and module.scheduler is not None and not weight_init_only:
                 module.scheduler.load_state_dict(uc["scheduler"])
         return uc
