@@ -85,7 +85,12 @@ class Trainer:
             module.on_train_start(datamodule)
 
         if ckpt_path:
-            user = self.ckpt_io.load(ckpt_path, module)
+            user = self.ckpt_io.load(
+                ckpt_path, module,
+                broadcast_over_dp=bool(
+                    self.cfg.get("exp_manager", {}).get(
+                        "broadcast_checkpoint_load", False)),
+            )
             self.global_step = int(user.get("global_step", 0))
             datamodule.consumed_samples = int(user.get("consumed_samples", 0))
 
