@@ -328,7 +328,12 @@ class ZeRO1AdamW:
             )
             if self.expert_dp_world > 1 and self.expert_dp_group is not None:
                 dist.all_reduce(g, group=self.expert_dp_group)
-                g.div_(self.expert_dp_world)
+            # per-rank losses are microbatch MEANS, so the DP convention
+            # divides by the FULL dp world — the expert-DP sum collects
+            # disjoint token contributions from all dp ranks (÷expert_dp
+            # alone left expert grads ep× too large)
+            if self.dp_world > 1:
+                g.div_(self.dp_world)
             expert_grads.append(g)
 
         # 2) global grad norm: count TP-sharded params on all ranks,

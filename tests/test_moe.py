@@ -380,3 +380,42 @@ def test_attach_aux_loss_gradient_semantics():
         ga = path(True, scale)
         gr = path(False, scale)
         assert torch.allclose(ga, gr, atol=1e-6), (ga - gr).abs().max()
+
+
+def _ep_dp_weights(rank, world):
+    """EP2 inside DP4: expert-0 weights after one ZeRO-1 step match the
+    single-rank run over the concatenated global batch (regression: expert
+    grads were divided by expert_dp only — ep× too large)."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.modules.moe import (
+        ExpertMLPs, MoE, RouterTopK,
+    )
+    from neuronx_distributed_training_amd.optim.zero1 import ZeRO1AdamW
+
+    ep = 2 if world == 4 else 1
+    ps.initialize_model_parallel(expert_model_parallel_size=ep)
+    torch.manual_seed(0)
+    router = RouterTopK(16, 4, 2, init_seed=1)
+    experts = ExpertMLPs(4, 16, 32, init_seed=2)
+    moe = MoE(router, experts)
+    opt = ZeRO1AdamW(
+        [("router.weight", router.weight),
+         ("experts.gate_up", experts.gate_up),
+         ("experts.down", experts.down)], lr=1e-2, grad_clip=1.0)
+    dp_r = ps.get_data_parallel_rank()
+    xs = [torch.randn(12, 16, generator=torch.Generator().manual_seed(50 + i))
+          for i in range(4)]
+    x = xs[dp_r] if world == 4 else torch.cat(xs)
+    opt.zero_grad()
+    y, _ = moe(x)
+    y.square().mean().backward()
+    opt.step()
+    if ps.get_expert_model_parallel_rank() == 0 and dp_r == 0:
+        return experts.gate_up.detach()[0, :6, :6].clone()
+    return None
+
+
+def test_ep2_dp2_expert_weights_match_single():
+    a1 = run_distributed(_ep_dp_weights, 1)[0]
+    a2 = [r for r in run_distributed(_ep_dp_weights, 4) if r is not None][0]
+    assert torch.allclose(a1, a2, atol=1e-4), (a1 - a2).abs().max()
