@@ -169,6 +169,10 @@ class LlamaStage(nn.Module):
             )
             if cfg.tie_word_embeddings:
                 self.lm_head.weight = self.embed_tokens.weight
+                if not self.is_first:
+                    # duplicate copy of the stage-0 embedding: keep it out
+                    # of the PP-summed grad norm (counted on stage 0)
+                    self.embed_tokens.weight.norm_duplicate = True
         cos, sin = build_rope_cache(
             cfg.max_position_embeddings, cfg.head_dim, cfg.rope_theta,
             rope_scaling=cfg.rope_scaling,

@@ -54,6 +54,8 @@ class GPTStage(nn.Module):
 
         if self.is_first or (self.is_last and cfg.share_embeddings_and_output_weights):
             self.embedding = Embedding(cfg)
+            if not self.is_first:
+                self.embedding.word_embeddings.weight.norm_duplicate = True
         self.layers = nn.ModuleList(
             [ParallelTransformerLayer(cfg, i) for i in range(start, end)]
         )

@@ -65,6 +65,8 @@ class MixtralStage(nn.Module):
             )
             if cfg.tie_word_embeddings:
                 self.lm_head.weight = self.embed_tokens.weight
+                if not self.is_first:
+                    self.embed_tokens.weight.norm_duplicate = True
         cos, sin = build_rope_cache(
             cfg.max_position_embeddings, cfg.head_dim, cfg.rope_theta
         )

@@ -123,6 +123,10 @@ class ZeRO1AdamW:
             is_tp = getattr(p, "tensor_model_parallel", False)
             if not is_tp and tp_rank != 0:
                 tp_once_mask[o : o + nel] = False
+            if getattr(p, "norm_duplicate", False):
+                # PP-replicated copy (tied embedding on the last stage):
+                # its sq is counted on the owning stage only
+                tp_once_mask[o : o + nel] = False
         self.wd_shard = wd_mask[self.shard_start : self.shard_start + self.shard_size].to(dev)
 
         # shard-relative ranges of sequence-parallel-tagged params (norm

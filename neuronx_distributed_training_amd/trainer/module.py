@@ -321,6 +321,7 @@ class LlamaModule(BaseModelModule):
             fuse_qkv=bool(mcfg.get("fuse_qkv", True)),
             activation_checkpoint=mcfg.get("activation_checkpoint"),
             sliding_window=mcfg.get("sliding_window"),
+            tie_word_embeddings=bool(mcfg.get("tie_word_embeddings", False)),
             dtype=dtype,
         )
         if ps.get_pipeline_model_parallel_world_size() > 1:

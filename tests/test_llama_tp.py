@@ -318,3 +318,41 @@ def test_dpo_logprob_grads_exact():
     a2 = [r for r in run_distributed(_dpo_logprob_grads, 2) if r is not None][0]
     for x, y in zip(a1, a2):
         assert torch.allclose(x, y, atol=1e-4), (x - y).abs().max()
+
+
+def _kv_replica_lockstep(rank, world):
+    """TP4 with kv_replicator=2: kv replica pairs hold identical weights
+    after 3 optimizer steps (grad-SUM hooks + ZeRO shard updates)."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 2, "seq_length": 32},
+        "distributed_strategy": {"tensor_model_parallel_size": world,
+                                 "sequence_parallel": world > 1},
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 2, "num_attention_heads": 4, "num_kv_heads": 2,
+            "qkv_linear": True, "kv_replicator": 2,
+            "grad_clip": 1.0, "optim": {"lr": 1e-2, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(3)
+    mod = LlamaModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=5)
+    g = torch.Generator().manual_seed(5)
+    for _ in range(3):
+        ids = torch.randint(0, 128, (2, 32), generator=g)
+        mod.training_step([{"input_ids": ids, "labels": ids.clone()}])
+    kv = mod.model.model.layers[0].self_attn.qkv_proj
+    return kv.weight_k.detach().flatten()[:64].clone()
+
+
+def test_kv_replica_lockstep_tp4():
+    res = run_distributed(_kv_replica_lockstep, 4)
+    assert torch.allclose(res[0], res[1], atol=1e-6)
+    assert torch.allclose(res[2], res[3], atol=1e-6)

@@ -330,3 +330,48 @@ def test_vp2_weights_match_single():
     w1 = run_distributed(_vp_weights, 1)[0]
     w2 = [w for w in run_distributed(_vp_weights, 2) if w is not None][0]
     assert torch.allclose(w1, w2, atol=1e-4), (w1 - w2).abs().max()
+
+
+def _tied_pp_weights(rank, world):
+    """PP2 tied embeddings: stage replicas stay in lockstep AND match the
+    single-rank weights after 2 ZeRO steps (regression: the tied weight's
+    sq was counted on both stages in the PP-summed grad norm)."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+
+    pp = 2 if world == 2 else 1
+    ps.initialize_model_parallel(pipeline_model_parallel_size=pp)
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 1, "seq_length": 32},
+        "distributed_strategy": {"pipeline_model_parallel_size": pp},
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 4, "num_attention_heads": 4, "num_kv_heads": 2,
+            "tie_word_embeddings": True,
+            "grad_clip": 1.0, "optim": {"lr": 1e-2, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(7)
+    mod = LlamaModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=4)
+    g = torch.Generator().manual_seed(5)
+    for _ in range(2):
+        micros = [
+            {"input_ids": (ids := torch.randint(0, 128, (1, 32), generator=g)),
+             "labels": ids.clone()}
+            for _ in range(2)
+        ]
+        mod.training_step(micros)
+    m = mod.model
+    emb = m.embed_tokens if hasattr(m, "embed_tokens") else m.model.embed_tokens
+    return emb.weight.detach()[:16, :8].clone()
+
+
+def test_tied_pp2_weights_match_single():
+    ref = run_distributed(_tied_pp_weights, 1)[0]
+    res = run_distributed(_tied_pp_weights, 2)
+    assert torch.allclose(res[0], res[1], atol=1e-6)
+    assert torch.allclose(ref, res[0], atol=1e-4), (ref - res[0]).abs().max()
