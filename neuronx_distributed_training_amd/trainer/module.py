@@ -258,6 +258,8 @@ class BaseModelModule:
         tp_rank = ps.get_tensor_model_parallel_rank()
         sq = torch.zeros((), dtype=torch.float32, device=self.device)
         for p in self.model.parameters():
+            if getattr(p, "norm_duplicate", False):
+                continue  # PP-replicated tied-embedding copy: counted once
             if getattr(p, "tensor_model_parallel", False) or tp_rank == 0:
                 sq += p.float().pow(2).sum()
         if ps.get_tensor_model_parallel_world_size() > 1:
