@@ -30,7 +30,14 @@ def _worker(rank, world_size, port, fn, args, q):
             dist.destroy_process_group()
 
 
-_PORT = [29701]
+def _free_port() -> int:
+    # bind-to-0 gives an OS-assigned free port (safe under parallel test
+    # runners; the tiny close→gloo-bind race is acceptable for CI)
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
 
 
 def run_distributed(fn, world_size: int, *args):
@@ -38,8 +45,7 @@ def run_distributed(fn, world_size: int, *args):
     Returns list of per-rank results ordered by rank. Raises on any failure."""
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    _PORT[0] += 1
-    port = _PORT[0] + os.getpid() % 500
+    port = _free_port()
     procs = [
         ctx.Process(target=_worker, args=(r, world_size, port, fn, args, q))
         for r in range(world_size)
