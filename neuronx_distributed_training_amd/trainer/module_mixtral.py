@@ -37,6 +37,7 @@ class MixtralModule(BaseModelModule):
             router_aux_loss_coef=float(moe.get("aux_loss_coef", 0.02)),
             router_type=str(moe.get("router_type", "top_k")),
             capacity_factor=moe.get("capacity_factor"),
+            tie_word_embeddings=bool(mcfg.get("tie_word_embeddings", False)),
             token_shuffle_group_size=int(
                 self.cfg.get("distributed_strategy", {}).get(
                     "token_shuffle_group_size", 1)),

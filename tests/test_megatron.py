@@ -521,3 +521,21 @@ def test_block_variant_sp_grads_exact(block, pos, act):
           if r is not None][0]
     for x, y in zip(a1, a2):
         assert torch.allclose(x, y, atol=1e-4), (x - y).abs().max()
+
+
+def test_megatron_generation():
+    """utils.generate works over GPTModel (full-recompute path)."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.megatron_gpt import (
+        GPTConfig, GPTModel,
+    )
+    from neuronx_distributed_training_amd.utils.generation import generate
+
+    ps.destroy_model_parallel()
+    torch.manual_seed(0)
+    m = GPTModel(GPTConfig(vocab_size=64, hidden_size=32, ffn_hidden_size=64,
+                           num_layers=1, num_attention_heads=2,
+                           max_position_embeddings=32))
+    ids = torch.randint(0, 64, (2, 8))
+    out = generate(m, ids, max_new_tokens=4)
+    assert out.shape == (2, 12)
