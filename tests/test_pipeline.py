@@ -375,3 +375,58 @@ def test_tied_pp2_weights_match_single():
     res = run_distributed(_tied_pp_weights, 2)
     assert torch.allclose(res[0], res[1], atol=1e-6)
     assert torch.allclose(ref, res[0], atol=1e-4), (ref - res[0]).abs().max()
+
+
+def _vp4_loss(rank, world):
+    """VP=4 on PP=2 (8 virtual stages, M=8): interleaved schedule parity."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.llama import LlamaConfig
+    from neuronx_distributed_training_amd.models.llama_pipeline import (
+        build_virtual_chunks,
+    )
+    from neuronx_distributed_training_amd.trainer.pipeline import (
+        InterleavedPipelineEngine,
+    )
+    import torch.distributed as dist
+
+    ps.initialize_model_parallel(pipeline_model_parallel_size=world)
+    torch.manual_seed(7)
+    cfg = LlamaConfig(vocab_size=128, hidden_size=64, intermediate_size=128,
+                      num_hidden_layers=8, num_attention_heads=4,
+                      num_key_value_heads=2, max_position_embeddings=32)
+    g = torch.Generator().manual_seed(99)
+    micro = []
+    for _ in range(8):
+        ids = torch.randint(0, 128, (1, 32), generator=g)
+        micro.append({"input_ids": ids, "labels": ids.clone()})
+    eng = InterleavedPipelineEngine(build_virtual_chunks(cfg, 4))
+    loss = eng.run_train(micro).float()
+    dist.all_reduce(loss, group=ps.get_pipeline_model_parallel_group())
+    return float(loss)
+
+
+def test_vp4_matches_reference():
+    ref = run_distributed(_ref8_loss, 1)[0]
+    res = run_distributed(_vp4_loss, 2)
+    assert abs(res[0] - res[1]) < 1e-6
+    assert abs(ref - res[0]) < 5e-3, (ref, res[0])
+
+
+def _ref8_loss(rank, world):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.llama import (
+        LlamaConfig, LlamaForCausalLM,
+    )
+
+    ps.initialize_model_parallel()
+    torch.manual_seed(7)
+    cfg = LlamaConfig(vocab_size=128, hidden_size=64, intermediate_size=128,
+                      num_hidden_layers=8, num_attention_heads=4,
+                      num_key_value_heads=2, max_position_embeddings=32)
+    m = LlamaForCausalLM(cfg)
+    g = torch.Generator().manual_seed(99)
+    tot = 0.0
+    for _ in range(8):
+        ids = torch.randint(0, 128, (1, 32), generator=g)
+        tot += float(m(ids, labels=ids))
+    return tot / 8
