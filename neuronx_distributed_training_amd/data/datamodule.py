@@ -9,7 +9,7 @@ DistributedSampler over the DP group, consumed-samples bookkeeping.
 
 from __future__ import annotations
 
-from typing import Dict, Iterator, List, Optional
+from typing import Dict, List, Optional
 
 import torch
 from torch.utils.data import DataLoader, Dataset, DistributedSampler

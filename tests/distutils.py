@@ -3,7 +3,6 @@
 from __future__ import annotations
 
 import os
-import pickle
 import traceback
 
 import torch.distributed as dist
