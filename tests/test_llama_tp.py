@@ -356,3 +356,32 @@ def test_kv_replica_lockstep_tp4():
     res = run_distributed(_kv_replica_lockstep, 4)
     assert torch.allclose(res[0], res[1], atol=1e-6)
     assert torch.allclose(res[2], res[3], atol=1e-6)
+
+
+def _ckpt_grads(rank, world, mode):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.llama import (
+        LlamaConfig, LlamaForCausalLM,
+    )
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(7)
+    cfg = LlamaConfig(**TINY, sequence_parallel=world > 1,
+                      activation_checkpoint=mode)
+    m = LlamaForCausalLM(cfg)
+    m.train()
+    ids = torch.randint(0, 128, (2, 32), generator=torch.Generator().manual_seed(9))
+    m(ids, labels=ids).backward()
+    return (m.model.layers[0].self_attn.o_proj.weight.grad[:, :8].clone(),
+            m.model.embed_tokens.weight.grad[:8, :8].clone())
+
+
+@pytest.mark.parametrize("mode", ["selective", "full"])
+def test_activation_checkpoint_grads_exact(mode):
+    """Recompute (selective/full) reproduces the no-checkpoint grads
+    exactly, with and without TP+SP."""
+    for world in (1, 2):
+        base = run_distributed(_ckpt_grads, world, None)[0]
+        got = run_distributed(_ckpt_grads, world, mode)[0]
+        for a, b in zip(base, got):
+            assert torch.allclose(a, b, atol=1e-6), (a - b).abs().max()
