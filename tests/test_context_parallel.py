@@ -156,3 +156,45 @@ def test_cp2_weights_match_cp1():
     w2 = run_distributed(_cp_weights, 2)
     assert torch.allclose(w2[0], w2[1], atol=1e-6)
     assert torch.allclose(w1, w2[0], atol=1e-4), (w1 - w2[0]).abs().max()
+
+
+def _dp_cp_weights(rank, world):
+    """DP2×CP2 on 4 ranks: weights after 2 ZeRO steps match single-rank."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+
+    cp = 2 if world == 4 else 1
+    ps.initialize_model_parallel(context_parallel_size=cp)
+    dp = ps.get_data_parallel_world_size()
+    cfg = {
+        "data": {"global_batch_size": 4, "micro_batch_size": 2, "seq_length": 32},
+        "distributed_strategy": {"context_parallel_size": cp},
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 2, "num_attention_heads": 4, "num_kv_heads": 2,
+            "grad_clip": 1.0, "optim": {"lr": 1e-2, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(3)
+    mod = LlamaModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=4)
+    g = torch.Generator().manual_seed(5)
+    r = ps.get_data_parallel_rank()
+    for _ in range(2):
+        glob = [torch.randint(0, 128, (2, 32), generator=g) for _ in range(2)]
+        if dp == 2:
+            micros = [{"input_ids": glob[r], "labels": glob[r].clone()}]
+        else:
+            micros = [{"input_ids": x, "labels": x.clone()} for x in glob]
+        mod.training_step(micros)
+    return mod.model.model.layers[0].input_layernorm.weight.detach().clone()
+
+
+def test_dp2_cp2_weights_match_single():
+    ref = run_distributed(_dp_cp_weights, 1)[0]
+    res = run_distributed(_dp_cp_weights, 4)
+    assert max(float((res[0] - r).abs().max()) for r in res) < 1e-6
+    assert torch.allclose(ref, res[0], atol=1e-4), (ref - res[0]).abs().max()
