@@ -91,7 +91,15 @@ def create_app(cfg: dict):
 
     @app.post("/v1/completions")
     def complete(req: dict = Body(...)):
-        ids = torch.tensor([tokenizer.encode(str(req["prompt"]))])
+        prompts = req["prompt"]
+        batched = isinstance(prompts, list)
+        if not batched:
+            prompts = [prompts]
+        enc = [tokenizer.encode(str(p)) for p in prompts]
+        # left-pad to a rectangle so positions align at the right edge
+        width = max(len(e) for e in enc)
+        pad = getattr(tokenizer, "pad_token_id", 0) or 0
+        ids = torch.tensor([[pad] * (width - len(e)) + e for e in enc])
         if torch.cuda.is_available():
             ids = ids.cuda()
         out = generate(
@@ -103,8 +111,11 @@ def create_app(cfg: dict):
             top_p=float(req.get("top_p", 0.0)),
             eos_token_id=getattr(tokenizer, "eos_token_id", None),
         )
-        new = out[0, ids.size(1):].tolist()
-        return {"text": tokenizer.decode(new), "tokens": len(new)}
+        texts = [tokenizer.decode(row[width:].tolist()) for row in out]
+        n_new = out.size(1) - width
+        if batched:
+            return {"text": texts, "tokens": n_new}
+        return {"text": texts[0], "tokens": n_new}
 
     return app
 
