@@ -100,10 +100,14 @@ def create_app(cfg: dict):
         width = max(len(e) for e in enc)
         pad = getattr(tokenizer, "pad_token_id", 0) or 0
         ids = torch.tensor([[pad] * (width - len(e)) + e for e in enc])
+        am = torch.tensor(
+            [[0] * (width - len(e)) + [1] * len(e) for e in enc]
+        ) if batched and any(len(e) != width for e in enc) else None
         if torch.cuda.is_available():
             ids = ids.cuda()
+            am = am.cuda() if am is not None else None
         out = generate(
-            model, ids,
+            model, ids, attention_mask=am,
             max_new_tokens=int(req.get("max_tokens",
                                        cfg.get("max_new_tokens", 64))),
             temperature=float(req.get("temperature", 0.0)),

@@ -17,6 +17,7 @@ def generate(
     top_k: int = 0,
     top_p: float = 0.0,
     use_cache: bool = True,
+    attention_mask: torch.Tensor | None = None,
 ) -> torch.Tensor:
     """input_ids: [b, s]. Returns [b, s + new]. Incremental decode with a
     KV cache when the model supports it (LlamaForCausalLM); full-recompute
@@ -24,6 +25,8 @@ def generate(
     model.eval()
     ids = input_ids
     cache = None
+    if attention_mask is not None:
+        use_cache = False  # padded batch: full recompute honors the mask
     if use_cache and getattr(model, "supports_kv_cache", False) \
             and not getattr(model.cfg, "sequence_parallel", False):
         from ..models.llama import KVCache
@@ -31,8 +34,12 @@ def generate(
         cache = KVCache(len(model.model.layers))
     step_in = ids
     for _ in range(max_new_tokens):
-        logits = model(step_in, kv_cache=cache) if cache is not None \
-            else model(ids)  # [b, s, V] (gathered over TP by the model)
+        if cache is not None:
+            logits = model(step_in, kv_cache=cache)
+        elif attention_mask is not None:
+            logits = model(ids, attention_mask=attention_mask)
+        else:
+            logits = model(ids)  # [b, s, V] (gathered over TP by the model)
         nxt = logits[:, -1].float()
         if temperature and temperature > 0:
             nxt = nxt / temperature
@@ -52,6 +59,10 @@ def generate(
         else:
             tok = nxt.argmax(-1, keepdim=True)
         ids = torch.cat([ids, tok], dim=1)
+        if attention_mask is not None:
+            attention_mask = torch.cat(
+                [attention_mask, torch.ones_like(tok)], dim=1
+            )
         step_in = tok  # cache path feeds only the new token next step
         if eos_token_id is not None and bool((tok == eos_token_id).all()):
             break
