@@ -17,6 +17,10 @@ from .samplers import (
 
 
 class MegatronDataModule(BaseDataModule):
+    # the megatron batch samplers position themselves from
+    # consumed_samples; the trainer must not also skip batches
+    resumes_via_sampler = True
+
     def setup(self):
         d = self.cfg["data"]
         max_steps = int(self.cfg.get("trainer", {}).get("max_steps", 100))

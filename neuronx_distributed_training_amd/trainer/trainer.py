@@ -97,6 +97,22 @@ class Trainer:
         loader = datamodule.train_dataloader()
         it = iter(loader)
         epoch = 0
+        if datamodule.consumed_samples and not getattr(
+            datamodule, "resumes_via_sampler", False
+        ):
+            # fast-forward the map-style loader to the resume point so
+            # post-resume steps see the SAME data as an uninterrupted run
+            gbs = datamodule.global_batch_size
+            n_micro = gbs // (datamodule.dp_size * datamodule.micro_batch_size)
+            batches_done = (datamodule.consumed_samples // gbs) * n_micro
+            bpe = max(len(loader), 1)
+            epoch = batches_done // bpe
+            sampler = getattr(loader, "sampler", None)
+            if hasattr(sampler, "set_epoch"):
+                sampler.set_epoch(epoch)
+                it = iter(loader)
+            for _ in range(batches_done % bpe):
+                next(it)
         while self.global_step < self.max_steps:
             try:
                 micro = list(datamodule.microbatch_iterator(it))

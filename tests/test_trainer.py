@@ -418,3 +418,55 @@ def _broadcast_load(rank, world, tmpdir):
 
 def test_broadcast_load_dp2(tmp_path):
     run_distributed(_broadcast_load, 2, str(tmp_path))
+
+
+def _resume_exact(rank, world, tmpdir, phase):
+    """Interrupted-and-resumed training is BIT-EXACT vs an uninterrupted
+    run (regression: resume restarted the epoch's data from sample 0;
+    the trainer now fast-forwards map-style loaders by consumed samples)."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.trainer import Trainer
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+    from neuronx_distributed_training_amd.trainer.checkpoint import (
+        find_latest_checkpoint,
+    )
+    from neuronx_distributed_training_amd.data.datamodule import build_datamodule
+
+    ps.initialize_model_parallel()
+    torch.manual_seed(0)
+    cfg = {
+        "trainer": {"max_steps": 2 if phase == "first" else 4,
+                    "log_every_n_steps": 100},
+        "data": {"kind": "synthetic", "global_batch_size": 2,
+                 "micro_batch_size": 1, "seq_length": 32,
+                 "num_train_samples": 64, "num_workers": 0},
+        "distributed_strategy": {},
+        "model": {"vocab_size": 128, "hidden_size": 64,
+                  "intermediate_size": 128, "num_layers": 2,
+                  "num_attention_heads": 4, "num_kv_heads": 2,
+                  "grad_clip": 1.0,
+                  "optim": {"lr": 1e-2,
+                            "sched": {"warmup_steps": 1, "max_steps": 4}}},
+        "precision": {"type": "fp32"},
+        "exp_manager": {"exp_dir": tmpdir,
+                        "checkpoint_callback_params": {
+                            "every_n_train_steps": 2, "save_top_k": 3}},
+    }
+    tr = Trainer(cfg)
+    tr.ckpt_dir = os.path.join(
+        tmpdir, "ck_S" if phase == "straight" else "ck_AB"
+    )
+    os.makedirs(tr.ckpt_dir, exist_ok=True)
+    module = LlamaModule(cfg)
+    dm = build_datamodule(cfg)
+    ckpt = find_latest_checkpoint(tr.ckpt_dir) if phase == "second" else None
+    tr.fit(module, dm, ckpt_path=ckpt)
+    return module.model.model.layers[0].self_attn.o_proj.weight.detach()[:, :8].clone()
+
+
+def test_resume_is_bit_exact(tmp_path):
+    d = str(tmp_path)
+    straight = run_distributed(_resume_exact, 1, d, "straight")[0]
+    run_distributed(_resume_exact, 1, d, "first")
+    resumed = run_distributed(_resume_exact, 1, d, "second")[0]
+    assert torch.equal(straight, resumed), (straight - resumed).abs().max()
