@@ -38,7 +38,7 @@ def init_distributed():
 
 
 def build_module(cfg):
-    src = cfg["model"].get("model_source", "hf")
+    src = cfg.get("model_source") or cfg["model"].get("model_source", "hf")
     arch = cfg["model"].get("arch", "llama")
     align = cfg.get("model_alignment_strategy", {})
     if align.get("dpo"):
