@@ -65,8 +65,12 @@ def train(cfg):
         context_parallel_size=int(ds.get("context_parallel_size", 1)),
         expert_model_parallel_size=int(ds.get("expert_model_parallel_size", 1)),
     )
-    seed = int(cfg.get("seed", 1234)) + 100 * ps.get_pipeline_model_parallel_rank()
-    torch.manual_seed(seed)
+    from neuronx_distributed_training_amd.parallel.random import (
+        model_parallel_manual_seed,
+    )
+
+    # per-stage seed + model-parallel RNG tracker (SP/TP dropout streams)
+    model_parallel_manual_seed(int(cfg.get("seed", 1234)))
 
     trainer_obj = Trainer(cfg)
     loggers, ckpt_dir = exp_manager(trainer_obj, cfg.get("exp_manager", {}))
