@@ -7,6 +7,13 @@
 set -x
 mkdir -p gpurun_out/r2
 
+# 0a) distributed-backward parity on the device (gloo spawn; both ranks
+# share the GPU — catches device-specific grad bugs the CPU run can't)
+timeout 600 python tools/check_grad_parity.py --layout tp2_sp \
+    > gpurun_out/r2/grad_parity.log 2>&1 || true
+timeout 600 python tools/check_grad_parity.py --layout lora_tp2_sp \
+    >> gpurun_out/r2/grad_parity.log 2>&1 || true
+
 # 0) baseline sanity: full GPU suite + current bench number
 timeout 900 python -m pytest tests -m gpu -x -q > gpurun_out/r2/pytest_gpu.log 2>&1
 timeout 600 python bench.py --steps 6 --warmup 2 > gpurun_out/r2/bench_base.json 2>&1
