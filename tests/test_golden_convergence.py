@@ -41,3 +41,38 @@ def test_golden_loss_curve():
     for i, (got, want) in enumerate(zip(losses, GOLDEN)):
         # fp32 CPU is deterministic; allow small slack for BLAS variation
         assert abs(got - want) < 0.02, (i, got, want)
+
+
+GOLDEN_MEGATRON = [3.688576, 3.458487, 3.138335, 2.855569, 2.572753,
+                   2.285691, 2.010218, 1.739632, 1.482066, 1.241034]
+
+
+def _run_megatron(rank, world):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.megatron_gpt import (
+        GPTConfig, GPTModel,
+    )
+    from neuronx_distributed_training_amd.optim.zero1 import ZeRO1AdamW
+
+    ps.initialize_model_parallel()
+    torch.manual_seed(1234)
+    m = GPTModel(GPTConfig(vocab_size=128, hidden_size=64, ffn_hidden_size=128,
+                           num_layers=2, num_attention_heads=4,
+                           max_position_embeddings=32))
+    opt = ZeRO1AdamW(list(m.named_parameters()), lr=5e-3, grad_clip=1.0)
+    g = torch.Generator().manual_seed(42)
+    ids = torch.randint(0, 128, (4, 32), generator=g)
+    losses = []
+    for _ in range(10):
+        opt.zero_grad()
+        loss = m(ids, labels=ids.clone())
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    return losses
+
+
+def test_golden_megatron_trajectory():
+    losses = run_distributed(_run_megatron, 1)[0]
+    for got, want in zip(losses, GOLDEN_MEGATRON):
+        assert abs(got - want) < 2e-4, (got, want)
