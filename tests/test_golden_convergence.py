@@ -76,3 +76,41 @@ def test_golden_megatron_trajectory():
     losses = run_distributed(_run_megatron, 1)[0]
     for got, want in zip(losses, GOLDEN_MEGATRON):
         assert abs(got - want) < 2e-4, (got, want)
+
+
+GOLDEN_MIXTRAL = [4.89852, 4.405124, 4.125925, 3.773686, 3.49911,
+                  3.241106, 2.986397, 2.731829, 2.453821, 2.183581]
+
+
+def _run_mixtral(rank, world):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.models.mixtral import (
+        MixtralConfig, MixtralForCausalLM,
+    )
+    from neuronx_distributed_training_amd.optim.zero1 import ZeRO1AdamW
+
+    ps.initialize_model_parallel()
+    torch.manual_seed(1234)
+    m = MixtralForCausalLM(MixtralConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=32, num_local_experts=4,
+        num_experts_per_tok=2))
+    opt = ZeRO1AdamW(list(m.named_parameters()), lr=5e-3, grad_clip=1.0)
+    g = torch.Generator().manual_seed(42)
+    ids = torch.randint(0, 128, (4, 32), generator=g)
+    losses = []
+    for _ in range(10):
+        opt.zero_grad()
+        loss = m(ids, labels=ids.clone())
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    return losses
+
+
+def test_golden_mixtral_trajectory():
+    """Pins router + expert + aux-loss numerics end to end."""
+    losses = run_distributed(_run_mixtral, 1)[0]
+    for got, want in zip(losses, GOLDEN_MIXTRAL):
+        assert abs(got - want) < 2e-4, (got, want)
