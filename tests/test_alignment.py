@@ -259,3 +259,37 @@ def test_lora_grads_exact_vs_tp1(sp):
     a2 = [r for r in run_distributed(_lora_grad_exact, 2, sp) if r is not None][0]
     assert torch.allclose(a1[0], a2[0], atol=1e-5), (a1[0] - a2[0]).abs().max()
     assert torch.allclose(a1[1], a2[1], atol=1e-5), (a1[1] - a2[1]).abs().max()
+
+
+def _dpo_golden(rank, world, tmpdir):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.alignment import DPOModule
+    from neuronx_distributed_training_amd.data.datamodule import build_datamodule
+
+    ps.initialize_model_parallel()
+    torch.manual_seed(0)
+    cfg = _dpo_cfg(tmpdir, os.path.join(tmpdir, "dpo.jsonl"), "dpo")
+    module = DPOModule(cfg)
+    module.setup()
+    module.configure_optimizers(max_steps=6)
+    dm = build_datamodule(cfg)
+    dm.setup()
+    module.on_train_start(dm)
+    loader = iter(dm.train_dataloader())
+    out = []
+    for _ in range(2):
+        micro = list(dm.microbatch_iterator(loader))
+        m = module.training_step(micro)
+        out.append(m["reduced_train_loss"])
+        out.append(m["reward_accuracy"])
+    return out
+
+
+def test_dpo_golden_trajectory(tmp_path):
+    """Pins the DPO path numerics: step 1 must be exactly -logsigmoid(0)
+    (policy == frozen reference ⇒ logits 0), then the policy separates."""
+    _write_dpo(os.path.join(str(tmp_path), "dpo.jsonl"))
+    vals = run_distributed(_dpo_golden, 1, str(tmp_path))[0]
+    assert abs(vals[0] - 0.693147) < 1e-4           # ln 2 at step 1
+    assert abs(vals[2] - 0.396708) < 2e-4           # recorded step-2 loss
+    assert vals[3] == 1.0                           # chosen > rejected
