@@ -293,3 +293,33 @@ def test_dpo_golden_trajectory(tmp_path):
     assert abs(vals[0] - 0.693147) < 1e-4           # ln 2 at step 1
     assert abs(vals[2] - 0.396708) < 2e-4           # recorded step-2 loss
     assert vals[3] == 1.0                           # chosen > rejected
+
+
+def test_orpo_tp2(tmp_path):
+    """ORPO training step under TP=2 matches single-rank (shares the
+    TP-aware sequence-logprob path with DPO but with avg-logp odds)."""
+    path = os.path.join(str(tmp_path), "dpo.jsonl")
+    _write_dpo(path)
+    l1 = run_distributed(_dpo_tp_run_orpo, 1, str(tmp_path))[0]
+    l2 = run_distributed(_dpo_tp_run_orpo, 2, str(tmp_path))
+    assert abs(l1 - l2[0]) < 5e-3, (l1, l2[0])
+
+
+def _dpo_tp_run_orpo(rank, world, tmpdir):
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.alignment import ORPOModule
+    from neuronx_distributed_training_amd.data.datamodule import build_datamodule
+
+    ps.initialize_model_parallel(tensor_model_parallel_size=world)
+    torch.manual_seed(0)
+    cfg = _dpo_cfg(tmpdir, os.path.join(tmpdir, "dpo.jsonl"), "orpo")
+    cfg["distributed_strategy"] = {"tensor_model_parallel_size": world}
+    module = ORPOModule(cfg)
+    module.setup()
+    module.configure_optimizers(max_steps=4)
+    dm = build_datamodule(cfg)
+    dm.setup()
+    module.on_train_start(dm)
+    loader = iter(dm.train_dataloader())
+    micro = list(dm.microbatch_iterator(loader))
+    return module.training_step(micro)["reduced_train_loss"]
