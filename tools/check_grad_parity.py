@@ -25,11 +25,12 @@ LAYOUTS = {
     "tp2_sp": dict(tp=2, sp=True),
     "tp4_sp": dict(tp=4, sp=True),
     "tp2_sp_gqa": dict(tp=2, sp=True, qkv=True),
-    "tp8_sp": dict(tp=8, sp=True),
+    "tp8_sp": dict(tp=8, sp=True, heads=8, kv=8),  # bench head layout
 }
 
 
-def run(rank: int, world: int, tp: int = 1, sp: bool = False, qkv: bool = False):
+def run(rank: int, world: int, tp: int = 1, sp: bool = False,
+        qkv: bool = False, heads: int = 4, kv: int = 2):
     from neuronx_distributed_training_amd.parallel import state as ps
     from neuronx_distributed_training_amd.parallel.layers import (
         allreduce_sequence_parallel_grads,
@@ -42,8 +43,8 @@ def run(rank: int, world: int, tp: int = 1, sp: bool = False, qkv: bool = False)
     torch.manual_seed(7)
     cfg = LlamaConfig(
         vocab_size=128, hidden_size=64, intermediate_size=128,
-        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
-        max_position_embeddings=32,
+        num_hidden_layers=2, num_attention_heads=heads,
+        num_key_value_heads=kv, max_position_embeddings=32,
         sequence_parallel=sp and world > 1,
         qkv_linear=qkv,
         kv_replicator=tp if (qkv and world > 1) else 1,
@@ -106,10 +107,14 @@ def main():
                   "OK" if d < 1e-4 else "MISMATCH")
         sys.exit(0 if ok else 1)
 
-    ref = run_distributed(run, 1, 1, False, lay.get("qkv", False))[0]
+    ref = run_distributed(
+        run, 1, 1, False, lay.get("qkv", False),
+        lay.get("heads", 4), lay.get("kv", 2),
+    )[0]
     world = lay["tp"]
     out = [r for r in run_distributed(
-        run, world, lay["tp"], lay["sp"], lay.get("qkv", False)
+        run, world, lay["tp"], lay["sp"], lay.get("qkv", False),
+        lay.get("heads", 4), lay.get("kv", 2),
     ) if r is not None]
     ok = True
     for name, a, b in (("o_proj", ref[0], out[0][0]),
