@@ -15,8 +15,9 @@ void launch_swiglu_bwd(const void*, const void*, void*, long, int,
                        hipStream_t);
 void launch_rope_fwd(const void*, const void*, const void*, void*, int, int,
                      int, int, const long*, int, float, hipStream_t);
-void launch_adamw(void*, const void*, void*, void*, const void*, long, float,
-                  float, float, float, float, int, hipStream_t);
+void launch_adamw(void*, const void*, void*, void*, const void*, const void*,
+                  void*, long, float, float, float, float, float, int,
+                  hipStream_t);
 void launch_flash_fwd(const void*, const void*, const void*, void*, void*,
                       int, int, int, int, int, bool, float, int, const long*,
                       const long*, const long*, hipStream_t);
@@ -121,12 +122,28 @@ torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cost, torch::Tensor sint,
 
 void adamw_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                 torch::Tensor v, torch::Tensor wd_mask, double lr, double b1,
-                double b2, double eps, double wd, long step) {
+                double b2, double eps, double wd, long step,
+                c10::optional<torch::Tensor> grad_scale,
+                c10::optional<torch::Tensor> p_bf16) {
   CHECK_IN(p);
   CHECK_IN(g);
+  void* gs = nullptr;
+  if (grad_scale.has_value()) {
+    TORCH_CHECK(grad_scale->scalar_type() == torch::kFloat32 &&
+                    grad_scale->is_cuda(),
+                "grad_scale must be a device fp32 scalar");
+    gs = grad_scale->data_ptr();
+  }
+  void* pb = nullptr;
+  if (p_bf16.has_value()) {
+    TORCH_CHECK(p_bf16->scalar_type() == torch::kBFloat16 &&
+                    p_bf16->is_contiguous() && p_bf16->numel() == p.numel(),
+                "p_bf16 must be contiguous bf16 of the same numel");
+    pb = p_bf16->data_ptr();
+  }
   launch_adamw(p.data_ptr(), g.data_ptr(), m.data_ptr(), v.data_ptr(),
-               wd_mask.data_ptr(), p.numel(), (float)lr, (float)b1, (float)b2,
-               (float)eps, (float)wd, (int)step, cur_stream());
+               wd_mask.data_ptr(), gs, pb, p.numel(), (float)lr, (float)b1,
+               (float)b2, (float)eps, (float)wd, (int)step, cur_stream());
 }
 
 static void check_bhsd(const torch::Tensor& t, const char* name) {
@@ -308,7 +325,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("rope_fwd", &rope_fwd);
-  m.def("adamw_step", &adamw_step);
+  m.def("adamw_step", &adamw_step, py::arg("p"), py::arg("g"), py::arg("m"),
+        py::arg("v"), py::arg("wd_mask"), py::arg("lr"), py::arg("b1"),
+        py::arg("b2"), py::arg("eps"), py::arg("wd"), py::arg("step"),
+        py::arg("grad_scale") = py::none(), py::arg("p_bf16") = py::none());
   m.def("flash_attn_fwd", &flash_attn_fwd, pybind11::arg("q"),
         pybind11::arg("k"), pybind11::arg("v"), pybind11::arg("causal"),
         pybind11::arg("scale"), pybind11::arg("window") = 0);

@@ -16,10 +16,31 @@ compare the HIP kernel against this).
 from __future__ import annotations
 
 import math
+import os
 
 import torch
 
 from . import kernels_for
+
+
+def _attn_version() -> int:
+    """Kernel generation: 2 = 16x16x32 MFMA 8-wave (round-1 default),
+    3 = 32x32x16 swapped-operand register-softmax (ROADMAP §1).
+    NXDT_ATTN_V3=1/0 overrides."""
+    env = os.environ.get("NXDT_ATTN_V3", "")
+    if env == "1":
+        return 3
+    if env == "0":
+        return 2
+    return 2  # default until the v3 A/B shows a win on hardware
+
+
+def _fwd_fn(kern):
+    return kern.flash_attn_fwd_v3 if _attn_version() == 3 else kern.flash_attn_fwd
+
+
+def _bwd_fn(kern):
+    return kern.flash_attn_bwd_v3 if _attn_version() == 3 else kern.flash_attn_bwd
 
 
 def _make_mask(sq, sk, causal, window, device):
@@ -54,7 +75,7 @@ class _FlashAttnFn(torch.autograd.Function):
         kern = kernels_for(q)
         if kern is not None:
             # kernel takes arbitrary-strided [b, h, s, d] views (d contig)
-            o, lse = kern.flash_attn_fwd(q, k, v, causal, scale, window)
+            o, lse = _fwd_fn(kern)(q, k, v, causal, scale, window)
         else:
             o, lse = _cpu_ref_fwd(q, k, v, causal, scale, window)
         ctx.save_for_backward(q, k, v, o, lse)
@@ -68,7 +89,7 @@ class _FlashAttnFn(torch.autograd.Function):
         q, k, v, o, lse = ctx.saved_tensors
         kern = kernels_for(q)
         if kern is not None:
-            dq, dk, dv = kern.flash_attn_bwd(
+            dq, dk, dv = _bwd_fn(kern)(
                 do, q, k, v, o, lse, ctx.causal, ctx.scale, ctx.window
             )
             return dq, dk, dv, None, None, None

@@ -145,11 +145,24 @@ class ZeRO1AdamW:
         self.normmask_all = bool(nm.all())
         self.normmask_shard = nm.to(dev).to(torch.float32)
 
-        # map param -> (offset, numel) for grad fill
+        # map param -> (offset, numel) for grad fill. Each param also gets a
+        # Megatron-style ``p.main_grad`` fp32 view into the flat buffer:
+        # post-accumulate-grad hooks cast/accumulate autograd's bf16 grad
+        # directly into it and drop p.grad, so gradient accumulation happens
+        # in fp32 (reference mixed-precision fp32-grad-acc semantics,
+        # model/base.py:118-132) and step() needs no collect/zero passes over
+        # the 32 GB buffer.
         self._grad_views = [
             (p, self.grad_flat[o : o + p.numel()].view(p.shape))
             for (n, p), o in zip(self.named_params, offsets)
         ]
+        self._touched = [False] * len(self.named_params)
+        self._acc_hooks = []
+        for i, (p, gview) in enumerate(self._grad_views):
+            p.main_grad = gview
+            self._acc_hooks.append(
+                p.register_post_accumulate_grad_hook(self._make_acc_hook(i))
+            )
 
         # ---- expert-parallel state (unsharded, synced over expert-DP) ----
         self.expert_state = []
@@ -206,29 +219,30 @@ class ZeRO1AdamW:
                                            "remaining": len(idxs)})
                 for i in idxs:
                     self._param_bucket[i] = bi
-            self._hook_handles = []
-            for i, (n, p) in enumerate(self.named_params):
-                self._hook_handles.append(
-                    p.register_post_accumulate_grad_hook(self._make_hook(i))
-                )
-
-    def _make_hook(self, i):
+    def _make_acc_hook(self, i):
         def hook(p):
-            # fires after EVERY grad accumulation; only the final
-            # microbatch (enable_grad_sync armed) launches communication
-            if not self._sync_enabled:
-                return
-            o = self.offsets[i]
-            self.grad_flat[o : o + p.numel()].copy_(p.grad.reshape(-1))
-            bi = self._param_bucket[i]
-            st = self._bucket_state[bi]
-            st["remaining"] -= 1
-            if st["remaining"] == 0:
-                w = dist.all_reduce(
-                    self.grad_flat[st["lo"] : st["hi"]],
-                    group=self.dp_group, async_op=True,
-                )
-                self._pending_works.append(w)
+            # fires after every backward's grad accumulation for this param:
+            # cast/accumulate into the fp32 flat buffer and free the bf16
+            # grad immediately (fp32 grad accumulation across microbatches).
+            gview = self._grad_views[i][1]
+            if self._touched[i]:
+                gview.add_(p.grad)
+            else:
+                gview.copy_(p.grad)
+                self._touched[i] = True
+            p.grad = None
+            # overlap mode: on the LAST microbatch (armed), launch the
+            # bucket's DP all-reduce as soon as all its grads are in
+            if self._sync_enabled:
+                bi = self._param_bucket[i]
+                st = self._bucket_state[bi]
+                st["remaining"] -= 1
+                if st["remaining"] == 0:
+                    w = dist.all_reduce(
+                        self.grad_flat[st["lo"] : st["hi"]],
+                        group=self.dp_group, async_op=True,
+                    )
+                    self._pending_works.append(w)
         return hook
 
     def enable_grad_sync(self):
@@ -238,14 +252,12 @@ class ZeRO1AdamW:
             self._sync_enabled = True
 
     def _finish_overlap_reduce(self):
-        # flush buckets whose params produced no grad this step
+        # flush buckets whose params produced no grad this step (after
+        # clearing any stale data their flat ranges may hold)
         for st in self._bucket_state:
             if st["remaining"] > 0:
                 for i in st["idxs"]:
-                    p = self.named_params[i][1]
-                    o = self.offsets[i]
-                    if p.grad is not None:
-                        self.grad_flat[o : o + p.numel()].copy_(p.grad.reshape(-1))
+                    self._finalize_one(i)
                 self._pending_works.append(
                     dist.all_reduce(
                         self.grad_flat[st["lo"] : st["hi"]],
@@ -260,8 +272,25 @@ class ZeRO1AdamW:
             st["remaining"] = len(st["idxs"])
         self._sync_enabled = False
 
+    def _finalize_one(self, i: int):
+        """Bring param i's flat-grad range up to date when its hook never
+        fired this step: manual ``p.grad`` assignment (tests / raw-autograd
+        users) is collected; otherwise stale data is zeroed."""
+        if self._touched[i]:
+            return
+        p, gview = self._grad_views[i]
+        if p.grad is not None:
+            gview.copy_(p.grad)
+        else:
+            gview.zero_()
+        self._touched[i] = True
+
     # -- hooks the trainer uses --
     def zero_grad(self, set_to_none: bool = True):
+        # grads live in the fp32 flat buffer; marking all params untouched
+        # makes the next hook fire a copy_ (overwrite) instead of add_, so
+        # no 32 GB zero_() pass is needed.
+        self._touched = [False] * len(self.named_params)
         for _, p in self.named_params:
             if set_to_none:
                 p.grad = None
@@ -272,31 +301,33 @@ class ZeRO1AdamW:
                 p.grad = None
             elif p.grad is not None:
                 p.grad.zero_()
-        self.grad_flat.zero_()
 
     @torch.no_grad()
     def _collect_grads(self):
-        for p, gview in self._grad_views:
-            if p.grad is not None:
-                gview.copy_(p.grad.reshape(gview.shape))
+        for i in range(len(self.named_params)):
+            self._finalize_one(i)
 
     @torch.no_grad()
     def step(self) -> torch.Tensor:
         # 1) grads → fp32, synced over DP: either the backward-overlapped
-        #    bucketed all-reduce (flag) or one reduce-scatter here
+        #    bucketed all-reduce (flag) or one reduce-scatter here.
+        #    The shard holds the DP/CP **sum**; the ÷world divisor is folded
+        #    into the AdamW kernel's grad_scale together with the clip scale
+        #    (one pass instead of separate div_/mul_ sweeps over the buffer).
+        divisor = 1.0
         if self.overlap_grad_reduce:
             self._finish_overlap_reduce()
             shard = self.grad_flat[
                 self.shard_start : self.shard_start + self.shard_size
             ]
-            shard.div_(self.dp_world)
+            divisor *= self.dp_world
         elif self.dp_world > 1:
             self._collect_grads()
             shard = torch.empty(
                 self.shard_size, dtype=torch.float32, device=self.device
             )
             dist.reduce_scatter_tensor(shard, self.grad_flat, group=self.dp_group)
-            shard.div_(self.dp_world)
+            divisor *= self.dp_world
         else:
             self._collect_grads()
             shard = self.grad_flat[
@@ -308,7 +339,7 @@ class ZeRO1AdamW:
         cp_group = ps.get_context_model_parallel_group()
         if ps.get_context_model_parallel_world_size() > 1:
             dist.all_reduce(shard, group=cp_group)
-            shard.div_(ps.get_context_model_parallel_world_size())
+            divisor *= ps.get_context_model_parallel_world_size()
 
         # 1c) SUM sequence-parallel-tagged grads over TP (each rank's grad
         # covers only its sequence shard — reference sequence_parallel_enabled
@@ -344,6 +375,8 @@ class ZeRO1AdamW:
         #    replicated params only on tp rank 0; reduce over DP then TP/PP.
         #    torch.dot keeps this allocation-free (a masked pow() materializes
         #    a full fp32 copy of the 32 GB shard and can OOM at large MBS).
+        #    The shard still holds divisor× the true grad, so sq is
+        #    divisor²× the true square-sum — undone after the reductions.
         sq = torch.zeros((), dtype=torch.float32, device=shard.device)
         cs = 1 << 28  # 256M elements/chunk (BLAS dot has an int32 bound)
         for s0 in range(0, shard.numel(), cs):
@@ -351,6 +384,8 @@ class ZeRO1AdamW:
             if not self.normmask_all:
                 piece = piece * self.normmask_shard[s0 : s0 + cs]
             sq += torch.dot(piece, piece)
+        if divisor != 1.0:
+            sq /= divisor * divisor
         if expert_grads and ps.get_tensor_model_parallel_rank() == 0:
             # each expert set appears expert_dp_world times across DP
             sq = sq + sum(g.pow(2).sum() for g in expert_grads) / self.expert_dp_world
@@ -363,25 +398,47 @@ class ZeRO1AdamW:
         gnorm = sq.sqrt()
         self.grad_norm = gnorm
         if self.grad_clip and self.grad_clip > 0:
-            scale = self.grad_clip / (gnorm + 1e-6)
-            scale = torch.clamp(scale, max=1.0)
-            shard.mul_(scale)
+            clip_scale = torch.clamp(
+                self.grad_clip / (gnorm + 1e-6), max=1.0
+            )
             for g in expert_grads:
-                g.mul_(scale)
+                g.mul_(clip_scale)
+        else:
+            clip_scale = None
+        # combined on-load grad scale for the fused kernel: clip ∘ ÷divisor
+        if clip_scale is not None:
+            kernel_scale = clip_scale if divisor == 1.0 else clip_scale / divisor
+        elif divisor != 1.0:
+            kernel_scale = torch.full(
+                (), 1.0 / divisor, dtype=torch.float32, device=shard.device
+            )
+        else:
+            kernel_scale = None
 
-        # 3) AdamW on the fp32 shard
+        # 3) AdamW on the fp32 shard; the fused kernel also writes the bf16
+        #    params directly into param_flat's shard slice (no separate
+        #    cast/copy passes)
         self.step_count += 1
         b1, b2 = self.betas
         t = self.step_count
         from ..ops import _try_load
 
         k = _try_load() if shard.is_cuda else None
+        shard_slice = self.param_flat[
+            self.shard_start : self.shard_start + self.shard_size
+        ]
         if k is not None and hasattr(k, "adamw_step"):
             k.adamw_step(
                 self.master_shard, shard, self.exp_avg, self.exp_avg_sq,
                 self.wd_shard, self.lr, b1, b2, self.eps, self.weight_decay, t,
+                grad_scale=kernel_scale,
+                p_bf16=shard_slice if self.model_dtype == torch.bfloat16 else None,
             )
+            if self.model_dtype != torch.bfloat16:
+                shard_slice.copy_(self.master_shard)
         else:
+            if kernel_scale is not None:
+                shard.mul_(kernel_scale)
             self.exp_avg.mul_(b1).add_(shard, alpha=1 - b1)
             self.exp_avg_sq.mul_(b2).addcmul_(shard, shard, value=1 - b2)
             bc1 = 1 - b1 ** t

@@ -164,9 +164,14 @@ class PipelineEngine:
         if w is None or self.world == 1:
             return
         grp = ps.get_embedding_group()
-        if grp is None or w.grad is None:
+        if grp is None:
             return
-        dist.all_reduce(w.grad, group=grp)
+        # under ZeRO-1 the fp32 grad lives in w.main_grad (flat-buffer view;
+        # the post-accumulate hook clears w.grad) — reduce whichever holds it
+        g = getattr(w, "main_grad", None) if w.grad is None else w.grad
+        if g is None:
+            return
+        dist.all_reduce(g, group=grp)
 
 
 class InterleavedPipelineEngine:

@@ -421,10 +421,15 @@ def _mega_moe_pp(rank, world):
     ]
     m = mod.training_step(micros)
     if pp > 1 and ps.get_pipeline_model_parallel_rank() == 0:
+        # ZeRO-1 hooks move grads into the fp32 flat buffer (p.main_grad)
+        # and clear p.grad — check whichever holds the router grad
         gsum = sum(
-            float(p.grad.abs().sum())
+            float(
+                (p.grad if p.grad is not None else p.main_grad).abs().sum()
+            )
             for n, p in mod.model.named_parameters()
-            if "router" in n and p.grad is not None
+            if "router" in n
+            and (p.grad is not None or getattr(p, "main_grad", None) is not None)
         )
         assert gsum > 0, "stage-0 megatron router got no grad"
     return m["reduced_train_loss"]
