@@ -72,6 +72,10 @@ class LlamaChunk(nn.Module):
         return (s, b, self.cfg.hidden_size)
 
     def forward(self, x):
+        # mirror LlamaStage's CP handling: the trainer pre-splits the batch
+        # per CP rank (labels pre-shifted), so RoPE positions are offset by
+        # cp_rank*seq and the local next-token shift is skipped under CP
+        cp_rank = ps.get_context_model_parallel_rank()
         if self.is_first:
             x = self.embed_tokens(self._batch["input_ids"]).transpose(0, 1).contiguous()
             if self.cfg.sequence_parallel:
@@ -79,8 +83,10 @@ class LlamaChunk(nn.Module):
                     scatter_to_sequence_parallel_region,
                 )
                 x = scatter_to_sequence_parallel_region(x)
+        seq_full = self._batch["input_ids"].size(1)
+        pos_offset = cp_rank * seq_full
         for layer in self.layers:
-            x = layer(x, self.rope_cos, self.rope_sin, 0)
+            x = layer(x, self.rope_cos, self.rope_sin, pos_offset)
         if not self.is_last:
             return x
         if self.cfg.sequence_parallel:
@@ -91,12 +97,16 @@ class LlamaChunk(nn.Module):
         logits = self.lm_head(x, pre_mapped=self.cfg.sequence_parallel).transpose(0, 1)
         labels = self._batch.get("labels", self._batch["input_ids"])
         loss_mask = self._batch.get("loss_mask")
-        logits = logits[:, :-1]
-        labels = labels[:, 1:]
-        loss_mask = loss_mask[:, 1:] if loss_mask is not None else None
+        if ps.get_context_model_parallel_world_size() == 1:
+            logits = logits[:, :-1]
+            labels = labels[:, 1:]
+            loss_mask = loss_mask[:, 1:] if loss_mask is not None else None
         per_tok = parallel_cross_entropy(logits, labels)
         if loss_mask is not None:
             m = loss_mask.to(per_tok.dtype)
+            denom = self._batch.get("loss_denominator")
+            if denom is not None:
+                return (per_tok * m).sum() / denom
             return (per_tok * m).sum() / m.sum().clamp(min=1)
         return per_tok.mean()
 
@@ -236,5 +246,10 @@ class LlamaStage(nn.Module):
         per_tok = parallel_cross_entropy(logits, labels)
         if loss_mask is not None:
             m = loss_mask.to(per_tok.dtype)
+            denom = self._batch.get("loss_denominator")
+            if denom is not None:
+                # trainer-computed CP-global mean denominator (exact global
+                # mean for non-uniform masks, consistent with non-PP path)
+                return (per_tok * m).sum() / denom
             return (per_tok * m).sum() / m.sum().clamp(min=1)
         return per_tok.mean()

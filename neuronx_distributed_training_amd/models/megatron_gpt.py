@@ -62,6 +62,8 @@ class GPTConfig:
     sequence_parallel: bool = False
     activation_checkpoint: Optional[str] = None
     dtype: str = "float32"
+    sliding_window: Optional[int] = None  # Mistral-style windowed attention
+    # (megatron_mistral_config parity — in-kernel window on the flash path)
     # MoE
     num_moe_experts: int = 0
     moe_top_k: int = 2
@@ -224,7 +226,8 @@ class ParallelAttention(nn.Module):
         self.attn_dropout_p = cfg.attention_dropout
 
     def core_attention(self, q, k, v):
-        return flash_attn_func(q, k, v, causal=True, scale=self.scale)
+        return flash_attn_func(q, k, v, causal=True, scale=self.scale,
+                               window=self.cfg.sliding_window)
 
     def forward(self, x, cos, sin, pos_offset=0):
         s_in, b = x.size(0), x.size(1)

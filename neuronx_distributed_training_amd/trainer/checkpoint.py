@@ -48,15 +48,24 @@ def _optim_shard_name() -> str:
     return f"dp_rank_{dp:02d}_tp_rank_{tp:02d}_pp_rank_{pp:02d}.pt"
 
 
-def _cpu_copy(obj):
+def _cpu_copy_inner(obj):
     if torch.is_tensor(obj):
-        return obj.detach().to("cpu", non_blocking=False)
+        # overlapped D2H: copies queue on the stream; one synchronize at the
+        # end of _cpu_copy instead of a blocking round-trip per tensor
+        return obj.detach().to("cpu", non_blocking=True)
     if isinstance(obj, dict):
-        return {k: _cpu_copy(v) for k, v in obj.items()}
+        return {k: _cpu_copy_inner(v) for k, v in obj.items()}
     if isinstance(obj, (list, tuple)):
         t = type(obj)
-        return t(_cpu_copy(v) for v in obj)
+        return t(_cpu_copy_inner(v) for v in obj)
     return obj
+
+
+def _cpu_copy(obj):
+    out = _cpu_copy_inner(obj)
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    return out
 
 
 def _remote_fs(path: str):
@@ -158,19 +167,76 @@ class CheckpointIO:
         if dist.is_initialized():
             dist.barrier()
         if is_global_zero and not self.async_save:
+            # sync mode: the barrier above already guarantees every rank's
+            # (synchronous) writes are durable
             _touch_done(root, fs)
             if keep_top_k and fs is None:
                 self._prune(ckpt_dir, keep_top_k)
         elif is_global_zero and self.async_save:
-            def _commit(threads=list(self._pending), root=root, k=keep_top_k, d=ckpt_dir):
+            # async mode: OTHER ranks' writer threads may still be in
+            # flight after the barrier — the commit thread must not mark
+            # the checkpoint complete until every expected shard file
+            # exists (all writes land via tmp+atomic-rename, so presence
+            # implies a complete file). Otherwise a crash in that window
+            # leaves a 'done'-marked checkpoint with missing shards that
+            # find_latest_checkpoint would resume from.
+            expected = self._expected_shards(
+                root, fs, has_optim=module.optimizer is not None
+            )
+
+            def _commit(threads=list(self._pending), root=root, k=keep_top_k,
+                        d=ckpt_dir, expected=expected):
                 for t in threads:
                     t.join()
+                self._wait_for_shards(expected, fs)
                 _touch_done(root, fs)
                 if k and fs is None:
                     self._prune(d, k)
             tc = threading.Thread(target=_commit, daemon=False)
             tc.start()
             self._pending = [tc]
+
+    @staticmethod
+    def _expected_shards(root: str, fs, has_optim: bool) -> List[str]:
+        """Every shard path the full job writes for this tag (global-rank-0
+        view): one model shard per (tp, pp) coordinate and, when an
+        optimizer is attached, one optim shard per (dp, tp, pp)."""
+        tp_w = ps.get_tensor_model_parallel_world_size()
+        pp_w = ps.get_pipeline_model_parallel_world_size()
+        dp_w = ps.get_data_parallel_world_size()
+        paths = []
+        for pp in range(pp_w):
+            for tp in range(tp_w):
+                paths.append(_join(
+                    root, fs, "model",
+                    f"dp_rank_00_tp_rank_{tp:02d}_pp_rank_{pp:02d}.pt"))
+                if has_optim:
+                    for dp in range(dp_w):
+                        paths.append(_join(
+                            root, fs, "optim",
+                            f"dp_rank_{dp:02d}_tp_rank_{tp:02d}"
+                            f"_pp_rank_{pp:02d}.pt"))
+        return paths
+
+    @staticmethod
+    def _wait_for_shards(paths: List[str], fs, timeout_s: float = 900.0):
+        import time
+
+        def _exists(p):
+            return fs.exists(p) if fs is not None else os.path.exists(p)
+
+        deadline = time.monotonic() + timeout_s
+        missing = list(paths)
+        while missing:
+            missing = [p for p in missing if not _exists(p)]
+            if not missing:
+                return
+            if time.monotonic() > deadline:
+                raise TimeoutError(
+                    f"checkpoint commit: {len(missing)} shard(s) never "
+                    f"appeared (first: {missing[0]}); refusing to mark done"
+                )
+            time.sleep(0.2)
 
     def finalize(self):
         for t in self._pending:

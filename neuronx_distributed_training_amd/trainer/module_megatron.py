@@ -59,6 +59,7 @@ class MegatronGPTModule(BaseModelModule):
             sequence_parallel=bool(dstr.get("sequence_parallel", False)),
             activation_checkpoint=mcfg.get("activation_checkpoint"),
             dtype=dtype,
+            sliding_window=mcfg.get("sliding_window"),
             num_moe_experts=int(moe.get("num_experts", 0)),
             moe_top_k=int(moe.get("top_k", 2)),
             moe_frequency=int(moe.get("moe_frequency", 1)),

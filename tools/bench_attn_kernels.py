@@ -11,7 +11,11 @@ can be flipped on evidence (VERDICT round-1 item #3).
 
 import argparse
 import math
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
