@@ -254,24 +254,10 @@ class LlamaAttention(nn.Module):
             )
         elif cache is not None:
             k, v = cache.update(layer_idx, k, v)
-            if q.size(2) < k.size(2):
-                # incremental decode: query at global position past+i sees
-                # keys 0..past+i (causal within the new chunk too)
-                rep = self.n_heads_local // self.n_kv_local
-                sq, sk = q.size(2), k.size(2)
-                past = sk - sq
-                am = None
-                if sq > 1:
-                    bad = torch.ones(sq, sk, dtype=torch.bool,
-                                     device=q.device).triu(past + 1)
-                    am = bad.to(q.dtype) * -1e4
-                o = torch.nn.functional.scaled_dot_product_attention(
-                    q, k.repeat_interleave(rep, 1),
-                    v.repeat_interleave(rep, 1),
-                    attn_mask=am, is_causal=False, scale=self.scale,
-                )
-            else:
-                o = self.core_attention(q, k, v)  # prefill
+            # prefill (S_q == S_kv) and incremental decode (S_q < S_kv)
+            # both run the flash kernel: causal is bottom-right aligned, so
+            # query at global position past+i sees keys 0..past+i
+            o = flash_attn_func(q, k, v, causal=True, scale=self.scale)
         elif self.cfg.activation_checkpoint == "selective" and self.training:
             o = _ckpt(self.core_attention, q, k, v, use_reentrant=False)
         else:
@@ -349,8 +335,11 @@ class LlamaModel(nn.Module):
             assert not self.cfg.sequence_parallel, "kv cache requires SP off"
             pos_offset = kv_cache.seq_len
         else:
-            cp_rank = ps.get_context_model_parallel_rank()
-            pos_offset = cp_rank * input_ids.size(1)
+            from ..parallel.cp import cp_offsets
+
+            # zigzag CP layout: the two halves of the local chunk are two
+            # different global chunks (parallel/cp.py)
+            pos_offset = cp_offsets(input_ids.size(1))
         pad_mask = None
         if attention_mask is not None and bool((attention_mask == 0).any()):
             assert ps.get_context_model_parallel_world_size() == 1, \

@@ -73,9 +73,10 @@ class LlamaChunk(nn.Module):
 
     def forward(self, x):
         # mirror LlamaStage's CP handling: the trainer pre-splits the batch
-        # per CP rank (labels pre-shifted), so RoPE positions are offset by
-        # cp_rank*seq and the local next-token shift is skipped under CP
-        cp_rank = ps.get_context_model_parallel_rank()
+        # per CP rank (zigzag layout, labels pre-shifted), so RoPE positions
+        # come from cp_offsets and the local next-token shift is skipped
+        from ..parallel.cp import cp_offsets
+
         if self.is_first:
             x = self.embed_tokens(self._batch["input_ids"]).transpose(0, 1).contiguous()
             if self.cfg.sequence_parallel:
@@ -84,7 +85,7 @@ class LlamaChunk(nn.Module):
                 )
                 x = scatter_to_sequence_parallel_region(x)
         seq_full = self._batch["input_ids"].size(1)
-        pos_offset = cp_rank * seq_full
+        pos_offset = cp_offsets(seq_full)
         for layer in self.layers:
             x = layer(x, self.rope_cos, self.rope_sin, pos_offset)
         if not self.is_last:
@@ -215,7 +216,8 @@ class LlamaStage(nn.Module):
         return (s, b, self.cfg.hidden_size)
 
     def forward(self, x: Optional[torch.Tensor]):
-        cp_rank = ps.get_context_model_parallel_rank()
+        from ..parallel.cp import cp_offsets
+
         if self.is_first:
             ids = self._batch["input_ids"]
             x = self.embed_tokens(ids).transpose(0, 1).contiguous()
@@ -225,7 +227,7 @@ class LlamaStage(nn.Module):
                 )
                 x = scatter_to_sequence_parallel_region(x)
         seq_full = self._batch["input_ids"].size(1)
-        pos_offset = cp_rank * seq_full
+        pos_offset = cp_offsets(seq_full)  # zigzag CP layout
         for layer in self.layers:
             x = layer(x, self.rope_cos, self.rope_sin, pos_offset)
         if not self.is_last:

@@ -384,8 +384,16 @@ class GPTModel(nn.Module):
 
     def forward(self, input_ids, position_ids=None, labels=None, loss_mask=None,
                 loss_denominator=None):
-        cp_rank = ps.get_context_model_parallel_rank()
-        pos_offset = cp_rank * input_ids.size(1)
+        from ..parallel.cp import cp_offsets, cp_position_ids
+
+        pos_offset = cp_offsets(input_ids.size(1))  # zigzag CP layout
+        if (position_ids is None
+                and ps.get_context_model_parallel_world_size() > 1
+                and hasattr(self.embedding, "position_embeddings")):
+            # learned-absolute positions must use the global zigzag ids
+            position_ids = cp_position_ids(
+                input_ids.size(1), device=input_ids.device
+            ).unsqueeze(0).expand(input_ids.size(0), -1)
         x = self.embedding(input_ids, position_ids)
         router_logits = []
         full_ckpt = self.cfg.activation_checkpoint == "full" and self.training

@@ -141,8 +141,9 @@ class MixtralModel(nn.Module):
         self.register_buffer("rope_sin", sin, persistent=False)
 
     def forward(self, input_ids):
-        cp_rank = ps.get_context_model_parallel_rank()
-        pos_offset = cp_rank * input_ids.size(1)
+        from ..parallel.cp import cp_offsets
+
+        pos_offset = cp_offsets(input_ids.size(1))  # zigzag CP layout
         x = self.embed_tokens(input_ids).transpose(0, 1).contiguous()
         if self.cfg.sequence_parallel:
             x = scatter_to_sequence_parallel_region(x)

@@ -14,16 +14,16 @@ void launch_swiglu_fwd(const void*, void*, long, int, hipStream_t);
 void launch_swiglu_bwd(const void*, const void*, void*, long, int,
                        hipStream_t);
 void launch_rope_fwd(const void*, const void*, const void*, void*, int, int,
-                     int, int, const long*, int, float, hipStream_t);
+                     int, int, const long*, int, int, float, hipStream_t);
 void launch_adamw(void*, const void*, void*, void*, const void*, const void*,
                   void*, long, float, float, float, float, float, int,
                   hipStream_t);
 void launch_flash_fwd(const void*, const void*, const void*, void*, void*,
-                      int, int, int, int, int, bool, float, int, const long*,
-                      const long*, const long*, hipStream_t);
+                      int, int, int, int, int, int, bool, float, int,
+                      const long*, const long*, const long*, hipStream_t);
 void launch_flash_bwd(const void*, const void*, const void*, const void*,
                       const void*, const void*, void*, void*, void*, int, int,
-                      int, int, int, bool, float, int, const long*,
+                      int, int, int, int, bool, float, int, const long*,
                       const long*, const long*, const long*, hipStream_t);
 void launch_mfma_probe(const void*, const void*, void*, hipStream_t);
 void launch_mfma_probe32(const void*, const void*, void*, hipStream_t);
@@ -101,10 +101,12 @@ torch::Tensor swiglu_bwd(torch::Tensor dy, torch::Tensor gu) {
 }
 
 torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cost, torch::Tensor sint,
-                       long pos_offset) {
+                       long pos_offset, long pos_offset2) {
   // x: logical [b, h, s, d], any strides with d contiguous (e.g. a view of
   // the fused QKV output). Returns a [s, b, h, d]-contiguous tensor exposed
   // as a [b, h, s, d] view — zero permute copies in the attention path.
+  // pos_offset2 >= 0: zigzag CP layout — rows [s/2, s) take positions
+  // pos_offset2 + i instead of pos_offset + s/2 + i.
   TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.stride(3) == 1,
               "rope: x must be [b, h, s, d] on GPU with d contiguous");
   auto cos_c = cost.contiguous();
@@ -112,11 +114,12 @@ torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cost, torch::Tensor sint,
   TORCH_CHECK(cos_c.scalar_type() == torch::kFloat32, "rope table must be f32");
   int B = (int)x.size(0), H = (int)x.size(1), S = (int)x.size(2),
       D = (int)x.size(3);
+  TORCH_CHECK(pos_offset2 < 0 || S % 2 == 0, "zigzag rope needs even s");
   auto y_mem = torch::empty({S, B, H, D}, x.options());
   long xstr[3] = {x.stride(2), x.stride(0), x.stride(1)};  // s, b, h
   launch_rope_fwd(x.data_ptr(), cos_c.data_ptr(), sin_c.data_ptr(),
-                  y_mem.data_ptr(), S, B, H, D, xstr, (int)pos_offset, 1.0f,
-                  cur_stream());
+                  y_mem.data_ptr(), S, B, H, D, xstr, (int)pos_offset,
+                  (int)pos_offset2, 1.0f, cur_stream());
   return y_mem.permute({1, 2, 0, 3});
 }
 
@@ -163,6 +166,7 @@ std::vector<torch::Tensor> flash_attn_fwd_v3(torch::Tensor q, torch::Tensor k,
       D = (int)q.size(3);
   int HKV = (int)k.size(1);
   TORCH_CHECK(D == 128 || D == 64, "head_dim must be 64 or 128");
+  TORCH_CHECK(k.size(2) == S, "v3 kernel requires S_q == S_kv (use v2)");
   auto o_mem = torch::empty({S, B, HQ, D}, q.options());
   auto lse = torch::empty({B, HQ, S}, q.options().dtype(torch::kFloat32));
   long qs[3] = {q.stride(2), q.stride(0), q.stride(1)};
@@ -215,22 +219,25 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
                                           double scale, long window) {
   // q/k/v: logical [b, h, s, d], arbitrary strides (views of the QKV GEMM
   // output). Returns O as a [b, h, s, d] view of [s, b, h, d] storage.
+  // S_q != S_kv supported; causal is then bottom-right aligned (decode /
+  // ring-attention half-block convention, requires S_kv >= S_q).
   check_bhsd(q, "q");
   check_bhsd(k, "k");
   check_bhsd(v, "v");
-  int B = (int)q.size(0), HQ = (int)q.size(1), S = (int)q.size(2),
+  int B = (int)q.size(0), HQ = (int)q.size(1), SQ = (int)q.size(2),
       D = (int)q.size(3);
-  int HKV = (int)k.size(1);
+  int HKV = (int)k.size(1), SKV = (int)k.size(2);
   TORCH_CHECK(D == 128 || D == 64, "head_dim must be 64 or 128");
   TORCH_CHECK(HQ % HKV == 0, "GQA head mismatch");
-  auto o_mem = torch::empty({S, B, HQ, D}, q.options());
-  auto lse = torch::empty({B, HQ, S}, q.options().dtype(torch::kFloat32));
+  TORCH_CHECK(!causal || SKV >= SQ, "causal needs S_kv >= S_q");
+  auto o_mem = torch::empty({SQ, B, HQ, D}, q.options());
+  auto lse = torch::empty({B, HQ, SQ}, q.options().dtype(torch::kFloat32));
   long qs[3] = {q.stride(2), q.stride(0), q.stride(1)};
   long ks[3] = {k.stride(2), k.stride(0), k.stride(1)};
   long vs[3] = {v.stride(2), v.stride(0), v.stride(1)};
   launch_flash_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o_mem.data_ptr(),
-                   lse.data_ptr(), B, HQ, HKV, S, D, causal, (float)scale,
-                   (int)window, qs, ks, vs, cur_stream());
+                   lse.data_ptr(), B, HQ, HKV, SQ, SKV, D, causal,
+                   (float)scale, (int)window, qs, ks, vs, cur_stream());
   return {o_mem.permute({1, 2, 0, 3}), lse};
 }
 
@@ -242,27 +249,28 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
   if (dout.stride(3) != 1) dout = dout.contiguous();
   check_bhsd(dout, "dout");
   check_bhsd(q, "q");
-  int B = (int)q.size(0), HQ = (int)q.size(1), S = (int)q.size(2),
+  int B = (int)q.size(0), HQ = (int)q.size(1), SQ = (int)q.size(2),
       D = (int)q.size(3);
-  int HKV = (int)k.size(1);
+  int HKV = (int)k.size(1), SKV = (int)k.size(2);
+  TORCH_CHECK(!causal || SKV >= SQ, "causal needs S_kv >= S_q");
   auto delta = (dout.to(torch::kFloat32) * o.to(torch::kFloat32))
                    .sum(-1)
-                   .contiguous();  // [B, HQ, S] f32
-  auto dq_mem = torch::empty({S, B, HQ, D}, q.options());
+                   .contiguous();  // [B, HQ, SQ] f32
+  auto dq_mem = torch::empty({SQ, B, HQ, D}, q.options());
   // dkv writes one fp32 partial slab per q-head of each GQA group (full
   // grid occupancy at high TP); sum + cast here
   int group = HQ / HKV;
   auto f32 = q.options().dtype(torch::kFloat32);
-  auto dk_part = torch::empty({group, S, B, HKV, D}, f32);
-  auto dv_part = torch::empty({group, S, B, HKV, D}, f32);
+  auto dk_part = torch::empty({group, SKV, B, HKV, D}, f32);
+  auto dv_part = torch::empty({group, SKV, B, HKV, D}, f32);
   long qs[3] = {q.stride(2), q.stride(0), q.stride(1)};
   long ks[3] = {k.stride(2), k.stride(0), k.stride(1)};
   long vs[3] = {v.stride(2), v.stride(0), v.stride(1)};
   long ds[3] = {dout.stride(2), dout.stride(0), dout.stride(1)};
   launch_flash_bwd(dout.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                    lse.data_ptr(), delta.data_ptr(), dq_mem.data_ptr(),
-                   dk_part.data_ptr(), dv_part.data_ptr(), B, HQ, HKV, S, D,
-                   causal, (float)scale, (int)window, qs, ks, vs, ds,
+                   dk_part.data_ptr(), dv_part.data_ptr(), B, HQ, HKV, SQ,
+                   SKV, D, causal, (float)scale, (int)window, qs, ks, vs, ds,
                    cur_stream());
   auto dk_mem = dk_part.sum(0).to(torch::kBFloat16);
   auto dv_mem = dv_part.sum(0).to(torch::kBFloat16);
@@ -324,7 +332,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
-  m.def("rope_fwd", &rope_fwd);
+  m.def("rope_fwd", &rope_fwd, py::arg("x"), py::arg("cost"), py::arg("sint"),
+        py::arg("pos_offset"), py::arg("pos_offset2") = -1);
   m.def("adamw_step", &adamw_step, py::arg("p"), py::arg("g"), py::arg("m"),
         py::arg("v"), py::arg("wd_mask"), py::arg("lr"), py::arg("b1"),
         py::arg("b2"), py::arg("eps"), py::arg("wd"), py::arg("step"),

@@ -22,11 +22,13 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
                              // binding sums over dim 0) — one block per
                              // (key-block, b, hkv, g) keeps the grid full
                              // even at TP=8 where HKV_local = 1
-    int S, int Bb, int HQ, int HKV, float scale,
+    int SQ, int SKV, int Bb, int HQ, int HKV, float scale,
     int window, long sQs, long sQb, long sQh, long sKs, long sKb, long sKh,
     long sVs, long sVb, long sVh, long sDs, long sDb, long sDh) {
   constexpr int BN = 256;  // keys per block
   constexpr int BM = 64;   // q tile
+  // bottom-right causal alignment for S_q != S_kv (see flash_attn_fwd.hip)
+  const int coff = SKV - SQ;
   constexpr int KP = D + 8;
   constexpr int VP = BM + 8;
   __shared__ __bf16 q_lds[BM * KP];
@@ -55,7 +57,7 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
 #pragma unroll
   for (int sb = 0; sb < 2; ++sb) {
     const int r = krow_w + sb * 16 + (lane & 15);
-    const long row = (r < S) ? r : (S - 1);
+    const long row = (r < SKV) ? r : (SKV - 1);
 #pragma unroll
     for (int kk = 0; kk < DK; ++kk) {
       *(int4*)&kfrag[sb][kk] =
@@ -85,13 +87,14 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
     const int hq = hkv * group + g;
     const bf16* Qp = Q + b * sQb + hq * sQh;
     const bf16* dOp = dO + b * sDb + hq * sDh;
-    const float* Lp = LSE + ((long)b * HQ + hq) * S;
-    const float* Dp = DELTA + ((long)b * HQ + hq) * S;
+    const float* Lp = LSE + ((long)b * HQ + hq) * SQ;
+    const float* Dp = DELTA + ((long)b * HQ + hq) * SQ;
 
-    const int ib0 = CAUSAL ? (kbase / BM) : 0;
-    int nqb = (S + BM - 1) / BM;
+    // causal: first q tile containing a row with qcol + coff >= kbase
+    const int ib0 = CAUSAL ? max(0, (kbase - coff) / BM) : 0;
+    int nqb = (SQ + BM - 1) / BM;
     if (CAUSAL && window > 0)
-      nqb = min(nqb, (kbase + BN - 1 + window + BM - 1) / BM);
+      nqb = min(nqb, (kbase + BN - 1 + window - coff + BM - 1) / BM);
     for (int ib = ib0; ib < nqb; ++ib) {
       const int qbase = ib * BM;
       {  // stage Q, dO row-major (16B) + transposed (paired b32)
@@ -100,10 +103,10 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
           const int row = t / (D / 8);
           const int col8 = (t % (D / 8)) * 8;
           const int gr = qbase + row;
-          int4 qv = (gr < S) ? *(const int4*)(Qp + (long)gr * sQs + col8)
-                             : int4{0, 0, 0, 0};
-          int4 dv = (gr < S) ? *(const int4*)(dOp + (long)gr * sDs + col8)
-                             : int4{0, 0, 0, 0};
+          int4 qv = (gr < SQ) ? *(const int4*)(Qp + (long)gr * sQs + col8)
+                              : int4{0, 0, 0, 0};
+          int4 dv = (gr < SQ) ? *(const int4*)(dOp + (long)gr * sDs + col8)
+                              : int4{0, 0, 0, 0};
           *(int4*)&q_lds[row * KP + col8] = qv;
           *(int4*)&do_lds[row * KP + col8] = dv;
         }
@@ -112,14 +115,14 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
           const int row = (t / (D / 8)) * 2;
           const int col8 = (t % (D / 8)) * 8;
           const int g0 = qbase + row, g1 = g0 + 1;
-          int4 q0 = (g0 < S) ? *(const int4*)(Qp + (long)g0 * sQs + col8)
-                             : int4{0, 0, 0, 0};
-          int4 q1 = (g1 < S) ? *(const int4*)(Qp + (long)g1 * sQs + col8)
-                             : int4{0, 0, 0, 0};
-          int4 d0 = (g0 < S) ? *(const int4*)(dOp + (long)g0 * sDs + col8)
-                             : int4{0, 0, 0, 0};
-          int4 d1 = (g1 < S) ? *(const int4*)(dOp + (long)g1 * sDs + col8)
-                             : int4{0, 0, 0, 0};
+          int4 q0 = (g0 < SQ) ? *(const int4*)(Qp + (long)g0 * sQs + col8)
+                              : int4{0, 0, 0, 0};
+          int4 q1 = (g1 < SQ) ? *(const int4*)(Qp + (long)g1 * sQs + col8)
+                              : int4{0, 0, 0, 0};
+          int4 d0 = (g0 < SQ) ? *(const int4*)(dOp + (long)g0 * sDs + col8)
+                              : int4{0, 0, 0, 0};
+          int4 d1 = (g1 < SQ) ? *(const int4*)(dOp + (long)g1 * sDs + col8)
+                              : int4{0, 0, 0, 0};
           const __bf16 *qe0 = (const __bf16*)&q0, *qe1 = (const __bf16*)&q1;
           const __bf16 *de0 = (const __bf16*)&d0, *de1 = (const __bf16*)&d1;
 #pragma unroll
@@ -136,7 +139,7 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
       __syncthreads();
 
       // causal: skip waves whose keys are all above this q tile
-      if (!CAUSAL || wkey_min <= qbase + BM - 1) {
+      if (!CAUSAL || wkey_min <= qbase + BM - 1 + coff) {
         __bf16* pw = &pT_lds[wid * 32 * VP];
         __bf16* dw = &dsT_lds[wid * 32 * VP];
 #pragma unroll
@@ -156,16 +159,17 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
             dpt[1] = MFMA_16x16x32(vfrag[1][kk], db, dpt[1]);
           }
           const int qcol = qbase + nq * 16 + (lane & 15);
-          const float lse = (qcol < S) ? Lp[qcol] : 1e30f;
-          const float delta = (qcol < S) ? Dp[qcol] : 0.f;
+          const float lse = (qcol < SQ) ? Lp[qcol] : 1e30f;
+          const float delta = (qcol < SQ) ? Dp[qcol] : 0.f;
 #pragma unroll
           for (int sb = 0; sb < 2; ++sb) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
               const int krow = krow_w + sb * 16 + (lane >> 4) * 4 + r;
-              bool dead =
-                  (qcol >= S) || (krow >= S) || (CAUSAL && qcol < krow);
-              if (CAUSAL && window > 0) dead |= (qcol >= krow + window);
+              bool dead = (qcol >= SQ) || (krow >= SKV) ||
+                          (CAUSAL && qcol + coff < krow);
+              if (CAUSAL && window > 0)
+                dead |= (qcol + coff >= krow + window);
               const float p = dead ? 0.f : __expf(st[sb][r] - lse);
               const float ds = p * (dpt[sb][r] - delta) * scale;
               const int lrow = sb * 16 + (lane >> 4) * 4 + r;
@@ -200,7 +204,7 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
 
   // ---- store fp32 partials: [group][s][b][hkv][d] ----
   const long sOs = (long)Bb * HKV * D;
-  const long slab = (long)g * S * sOs + ((long)b * HKV + hkv) * D;
+  const long slab = (long)g * SKV * sOs + ((long)b * HKV + hkv) * D;
   float* dKp = dK + slab;
   float* dVp = dV + slab;
 #pragma unroll
@@ -210,7 +214,7 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int krow = krow_w + sb * 16 + (lane >> 4) * 4 + r;
-        if (krow < S) {
+        if (krow < SKV) {
           dKp[(long)krow * sOs + nj * 16 + (lane & 15)] = dkacc[sb][nj][r];
           dVp[(long)krow * sOs + nj * 16 + (lane & 15)] = dvacc[sb][nj][r];
         }
@@ -226,10 +230,11 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
     const bf16* __restrict__ K, const bf16* __restrict__ V,
     const float* __restrict__ LSE, const float* __restrict__ DELTA,
     bf16* __restrict__ dQ,  // [s, b, hq, d] contiguous
-    int S, int Bb, int HQ, int HKV, float scale, int window, long sQs,
-    long sQb, long sQh, long sKs, long sKb, long sKh, long sVs, long sVb,
-    long sVh, long sDs, long sDb, long sDh) {
+    int SQ, int SKV, int Bb, int HQ, int HKV, float scale, int window,
+    long sQs, long sQb, long sQh, long sKs, long sKb, long sKh, long sVs,
+    long sVb, long sVh, long sDs, long sDb, long sDh) {
   constexpr int BM = 256, BN = 64;
+  const int coff = SKV - SQ;  // bottom-right causal alignment
   constexpr int KP = D + 8;
   constexpr int VP = BN + 8;
   __shared__ __bf16 k_lds[BN * KP];
@@ -249,8 +254,8 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
   const bf16* dOp = dO + b * sDb + hq * sDh;
   const bf16* Kp = K + b * sKb + hkv * sKh;
   const bf16* Vp = V + b * sVb + hkv * sVh;
-  const float* Lp = LSE + ((long)b * HQ + hq) * S;
-  const float* Dp = DELTA + ((long)b * HQ + hq) * S;
+  const float* Lp = LSE + ((long)b * HQ + hq) * SQ;
+  const float* Dp = DELTA + ((long)b * HQ + hq) * SQ;
 
   const int q0 = qblock * BM;
   const int qrow_w = q0 + wid * 32;
@@ -261,7 +266,7 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
 #pragma unroll
   for (int sb = 0; sb < 2; ++sb) {
     const int r = qrow_w + sb * 16 + (lane & 15);
-    const long row = (r < S) ? r : (S - 1);
+    const long row = (r < SQ) ? r : (SQ - 1);
 #pragma unroll
     for (int kk = 0; kk < DK; ++kk) {
       *(int4*)&qfrag[sb][kk] =
@@ -276,8 +281,8 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
 #pragma unroll
     for (int rr = 0; rr < 4; ++rr) {
       const int qrow = qrow_w + sb * 16 + (lane >> 4) * 4 + rr;
-      lse[sb][rr] = (qrow < S) ? Lp[qrow] : 1e30f;
-      delta[sb][rr] = (qrow < S) ? Dp[qrow] : 0.f;
+      lse[sb][rr] = (qrow < SQ) ? Lp[qrow] : 1e30f;
+      delta[sb][rr] = (qrow < SQ) ? Dp[qrow] : 0.f;
     }
   }
 
@@ -288,10 +293,10 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
 #pragma unroll
     for (int nj = 0; nj < DN; ++nj) dqacc[sb][nj] = f32x4_t{0.f, 0.f, 0.f, 0.f};
 
-  const int kend = CAUSAL ? min(S, q0 + BM) : S;
+  const int kend = CAUSAL ? min(SKV, q0 + BM + coff) : SKV;
   const int nkb = (kend + BN - 1) / BN;
   const int jb0 =
-      (CAUSAL && window > 0) ? max(0, (q0 - window + 1) / BN) : 0;
+      (CAUSAL && window > 0) ? max(0, (q0 + coff - window + 1) / BN) : 0;
   const int wrow_max = qrow_w + 31;
 
   for (int jb = jb0; jb < nkb; ++jb) {
@@ -302,10 +307,10 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
         const int row = t / (D / 8);
         const int col8 = (t % (D / 8)) * 8;
         const int gr = kbase + row;
-        int4 kv = (gr < S) ? *(const int4*)(Kp + (long)gr * sKs + col8)
-                           : int4{0, 0, 0, 0};
-        int4 vv = (gr < S) ? *(const int4*)(Vp + (long)gr * sVs + col8)
-                           : int4{0, 0, 0, 0};
+        int4 kv = (gr < SKV) ? *(const int4*)(Kp + (long)gr * sKs + col8)
+                             : int4{0, 0, 0, 0};
+        int4 vv = (gr < SKV) ? *(const int4*)(Vp + (long)gr * sVs + col8)
+                             : int4{0, 0, 0, 0};
         *(int4*)&k_lds[row * KP + col8] = kv;
         *(int4*)&v_lds[row * KP + col8] = vv;
       }
@@ -314,10 +319,10 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
         const int row = (t / (D / 8)) * 2;
         const int col8 = (t % (D / 8)) * 8;
         const int g0 = kbase + row, g1 = g0 + 1;
-        int4 k0 = (g0 < S) ? *(const int4*)(Kp + (long)g0 * sKs + col8)
-                           : int4{0, 0, 0, 0};
-        int4 k1 = (g1 < S) ? *(const int4*)(Kp + (long)g1 * sKs + col8)
-                           : int4{0, 0, 0, 0};
+        int4 k0 = (g0 < SKV) ? *(const int4*)(Kp + (long)g0 * sKs + col8)
+                             : int4{0, 0, 0, 0};
+        int4 k1 = (g1 < SKV) ? *(const int4*)(Kp + (long)g1 * sKs + col8)
+                             : int4{0, 0, 0, 0};
         const __bf16 *e0 = (const __bf16*)&k0, *e1 = (const __bf16*)&k1;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
@@ -330,7 +335,7 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
     }
     __syncthreads();
 
-    if (!CAUSAL || kbase <= wrow_max) {
+    if (!CAUSAL || kbase <= wrow_max + coff) {
       __bf16* dsw = &ds_lds[wid * 32 * VP];
 #pragma unroll
       for (int nk = 0; nk < 4; ++nk) {
@@ -354,8 +359,9 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
             const int qrow = qrow_w + sb * 16 + (lane >> 4) * 4 + r;
-            bool dead = (kcol >= S) || (CAUSAL && kcol > qrow);
-            if (CAUSAL && window > 0) dead |= (kcol <= qrow - window);
+            bool dead = (kcol >= SKV) || (CAUSAL && kcol > qrow + coff);
+            if (CAUSAL && window > 0)
+              dead |= (kcol <= qrow + coff - window);
             const float p = dead ? 0.f : __expf(st[sb][r] - lse[sb][r]);
             const float ds = p * (dpt[sb][r] - delta[sb][r]) * scale;
             dsw[(sb * 16 + (lane >> 4) * 4 + r) * VP + nk * 16 +
@@ -388,7 +394,7 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int qrow = qrow_w + sb * 16 + (lane >> 4) * 4 + r;
-        if (qrow < S)
+        if (qrow < SQ)
           dQp[(long)qrow * sOs + nj * 16 + (lane & 15)] =
               f2bf(dqacc[sb][nj][r]);
       }
@@ -398,25 +404,26 @@ extern "C" {
 void launch_flash_bwd(const void* dout, const void* q, const void* k,
                       const void* v, const void* lse, const void* delta,
                       void* dq, void* dk, void* dv, int B, int HQ, int HKV,
-                      int S, int D, bool causal, float scale, int window,
-                      const long* qstr, const long* kstr, const long* vstr,
-                      const long* dostr, hipStream_t stream) {
+                      int SQ, int SKV, int D, bool causal, float scale,
+                      int window, const long* qstr, const long* kstr,
+                      const long* vstr, const long* dostr,
+                      hipStream_t stream) {
   dim3 blk(512);
-  dim3 gkv((S + 255) / 256, B * HQ);  // one block per (kblock, b, hkv, g)
-  dim3 gq((S + 255) / 256, B * HQ);
+  dim3 gkv((SKV + 255) / 256, B * HQ);  // one block per (kblock, b, hkv, g)
+  dim3 gq((SQ + 255) / 256, B * HQ);
 #define CASE(DD, CC)                                                          \
   do {                                                                        \
     flash_bwd_dkv_kernel<DD, CC><<<gkv, blk, 0, stream>>>(                    \
         (const bf16*)dout, (const bf16*)q, (const bf16*)k, (const bf16*)v,    \
-        (const float*)lse, (const float*)delta, (float*)dk, (float*)dv, S,   \
-        B, HQ, HKV, scale, window, qstr[0], qstr[1], qstr[2], kstr[0],        \
+        (const float*)lse, (const float*)delta, (float*)dk, (float*)dv, SQ,  \
+        SKV, B, HQ, HKV, scale, window, qstr[0], qstr[1], qstr[2], kstr[0],   \
         kstr[1],                                                              \
         kstr[2], vstr[0], vstr[1], vstr[2], dostr[0], dostr[1], dostr[2]);    \
     flash_bwd_dq_kernel<DD, CC><<<gq, blk, 0, stream>>>(                      \
         (const bf16*)dout, (const bf16*)q, (const bf16*)k, (const bf16*)v,    \
-        (const float*)lse, (const float*)delta, (bf16*)dq, S, B, HQ, HKV,     \
-        scale, window, qstr[0], qstr[1], qstr[2], kstr[0], kstr[1], kstr[2],  \
-        vstr[0], vstr[1], vstr[2], dostr[0], dostr[1], dostr[2]);             \
+        (const float*)lse, (const float*)delta, (bf16*)dq, SQ, SKV, B, HQ,    \
+        HKV, scale, window, qstr[0], qstr[1], qstr[2], kstr[0], kstr[1],      \
+        kstr[2], vstr[0], vstr[1], vstr[2], dostr[0], dostr[1], dostr[2]);    \
   } while (0)
   if (D == 128) {
     if (causal) CASE(128, true); else CASE(128, false);

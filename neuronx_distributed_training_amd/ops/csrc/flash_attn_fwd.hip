@@ -17,8 +17,8 @@ template <int D, bool CAUSAL>
 __global__ __launch_bounds__(512) void flash_fwd_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, bf16* __restrict__ O,
-    float* __restrict__ LSE,  // [B, HQ, S] f32
-    int S, int Bb, int HQ, int HKV, float scale,
+    float* __restrict__ LSE,  // [B, HQ, SQ] f32
+    int SQ, int SKV, int Bb, int HQ, int HKV, float scale,
     int window,                     // sliding window (<=0: disabled)
     long sQs, long sQb, long sQh,   // Q element strides (seq, batch, head)
     long sKs, long sKb, long sKh,   // K strides
@@ -37,6 +37,9 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
   const int hq = bh % HQ;
   const int b = bh / HQ;
   const int hkv = hq / (HQ / HKV);
+  // S_q != S_kv: causal is BOTTOM-RIGHT aligned (query i sees keys
+  // j <= i + SKV - SQ) — the KV-cache decode / ring half-block convention
+  const int coff = SKV - SQ;
 
   const bf16* Qp = Q + b * sQb + hq * sQh;
   const bf16* Kp = K + b * sKb + hkv * sKh;
@@ -50,7 +53,7 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
 #pragma unroll
   for (int sb = 0; sb < 2; ++sb) {
     const int r = qrow_w + sb * 16 + (lane & 15);
-    const long row = (r < S) ? r : (S - 1);
+    const long row = (r < SQ) ? r : (SQ - 1);
 #pragma unroll
     for (int kk = 0; kk < DK; ++kk) {
       const bf16* p = Qp + row * sQs + kk * 32 + (lane >> 4) * 8;
@@ -84,10 +87,10 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
     for (int nj = 0; nj < DN; ++nj) oacc[sb][nj] = f32x4_t{0.f, 0.f, 0.f, 0.f};
   }
 
-  const int kend = CAUSAL ? min(S, q0 + BM) : S;
+  const int kend = CAUSAL ? min(SKV, q0 + BM + coff) : SKV;
   const int nkb = (kend + BN - 1) / BN;
   const int jb0 =
-      (CAUSAL && window > 0) ? max(0, (q0 - window + 1) / BN) : 0;
+      (CAUSAL && window > 0) ? max(0, (q0 + coff - window + 1) / BN) : 0;
   const int wrow_max = qrow_w + 31;
 
   // T5 static form: the younger dispatch half gets priority so it is not
@@ -110,18 +113,18 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
       const int row = t / (D / 8);
       const int col8 = (t % (D / 8)) * 8;
       const int gr = kbase + row;
-      kreg[u] = (gr < S) ? *(const int4*)(Kp + (long)gr * sKs + col8)
-                         : int4{0, 0, 0, 0};
+      kreg[u] = (gr < SKV) ? *(const int4*)(Kp + (long)gr * sKs + col8)
+                           : int4{0, 0, 0, 0};
     }
     {
       const int t = threadIdx.x;  // BN*D/16 == 512 row-pair slices
       const int row = (t / (D / 8)) * 2;
       const int col8 = (t % (D / 8)) * 8;
       const int g0 = kbase + row, g1 = g0 + 1;
-      vreg0 = (g0 < S) ? *(const int4*)(Vp + (long)g0 * sVs + col8)
-                       : int4{0, 0, 0, 0};
-      vreg1 = (g1 < S) ? *(const int4*)(Vp + (long)g1 * sVs + col8)
-                       : int4{0, 0, 0, 0};
+      vreg0 = (g0 < SKV) ? *(const int4*)(Vp + (long)g0 * sVs + col8)
+                         : int4{0, 0, 0, 0};
+      vreg1 = (g1 < SKV) ? *(const int4*)(Vp + (long)g1 * sVs + col8)
+                         : int4{0, 0, 0, 0};
     }
   };
 
@@ -157,7 +160,7 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
     const int kbase = jb * BN;
     if (jb + 1 < nkb) issue_loads((jb + 1) * BN);
 
-    if (!CAUSAL || kbase <= wrow_max) {
+    if (!CAUSAL || kbase <= wrow_max + coff) {
       // ---- S = Q K^T for both sub-blocks (B-frags loaded once) ----
       f32x4_t sacc[2][4];
 #pragma unroll
@@ -178,9 +181,9 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
       for (int sb = 0; sb < 2; ++sb) {
         // interior tiles (every key visible to every row) skip the mask
         const bool full_tile =
-            (kbase + BN <= S) &&
-            (!CAUSAL || (kbase + BN - 1 <= qrow_w + sb * 16)) &&
-            (window <= 0 || kbase >= qrow_w + sb * 16 + 15 - window + 1);
+            (kbase + BN <= SKV) &&
+            (!CAUSAL || (kbase + BN - 1 <= qrow_w + sb * 16 + coff)) &&
+            (window <= 0 || kbase >= qrow_w + sb * 16 + coff + 15 - window + 1);
         float tile_max[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
         float sv[4][4];
         if (full_tile) {
@@ -201,8 +204,8 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
             for (int r = 0; r < 4; ++r) {
               const int qrow = qrow_w + sb * 16 + (lane >> 4) * 4 + r;
               float s = sacc[sb][nk][r];  // scale folded into Q
-              bool dead = (kcol >= S) || (CAUSAL && kcol > qrow);
-              if (CAUSAL && window > 0) dead |= (kcol <= qrow - window);
+              bool dead = (kcol >= SKV) || (CAUSAL && kcol > qrow + coff);
+              if (CAUSAL && window > 0) dead |= (kcol <= qrow + coff - window);
               s = dead ? -1e30f : s;
               sv[nk][r] = s;
               tile_max[r] = fmaxf(tile_max[r], s);
@@ -282,7 +285,7 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
   // ---- epilogue: O /= l (strided [s,b,h,d] store) + LSE ----
   bf16* Op = O + ((long)b * HQ + hq) * D;  // O contiguous [s, b, hq, d]
   const long sOs = (long)Bb * HQ * D;
-  float* Lp = LSE + ((long)b * HQ + hq) * S;
+  float* Lp = LSE + ((long)b * HQ + hq) * SQ;
 #pragma unroll
   for (int sb = 0; sb < 2; ++sb) {
     float inv_l[4];
@@ -294,7 +297,7 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int qrow = qrow_w + sb * 16 + (lane >> 4) * 4 + r;
-        if (qrow < S)
+        if (qrow < SQ)
           Op[(long)qrow * sOs + nj * 16 + (lane & 15)] =
               f2bf(oacc[sb][nj][r] * inv_l[r]);
       }
@@ -303,7 +306,7 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int qrow = qrow_w + sb * 16 + (lane >> 4) * 4 + r;
-        if (qrow < S)
+        if (qrow < SQ)
           Lp[qrow] = m_i[sb][r] + __logf(fmaxf(l_i[sb][r], 1e-30f));
       }
     }
@@ -312,16 +315,17 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
 
 extern "C" {
 void launch_flash_fwd(const void* q, const void* k, const void* v, void* o,
-                      void* lse, int B, int HQ, int HKV, int S, int D,
-                      bool causal, float scale, int window, const long* qstr,
-                      const long* kstr, const long* vstr, hipStream_t stream) {
-  dim3 grid((S + 255) / 256, B * HQ);
+                      void* lse, int B, int HQ, int HKV, int SQ, int SKV,
+                      int D, bool causal, float scale, int window,
+                      const long* qstr, const long* kstr, const long* vstr,
+                      hipStream_t stream) {
+  dim3 grid((SQ + 255) / 256, B * HQ);
   dim3 blk(512);
 #define CASE(DD, CC)                                                          \
   flash_fwd_kernel<DD, CC><<<grid, blk, 0, stream>>>(                         \
       (const bf16*)q, (const bf16*)k, (const bf16*)v, (bf16*)o, (float*)lse, \
-      S, B, HQ, HKV, scale, window, qstr[0], qstr[1], qstr[2], kstr[0],       \
-      kstr[1], kstr[2], vstr[0], vstr[1], vstr[2])
+      SQ, SKV, B, HQ, HKV, scale, window, qstr[0], qstr[1], qstr[2],          \
+      kstr[0], kstr[1], kstr[2], vstr[0], vstr[1], vstr[2])
   if (D == 128) {
     if (causal) CASE(128, true); else CASE(128, false);
   } else if (D == 64) {

@@ -35,21 +35,32 @@ def _attn_version() -> int:
     return 2  # default until the v3 A/B shows a win on hardware
 
 
-def _fwd_fn(kern):
-    return kern.flash_attn_fwd_v3 if _attn_version() == 3 else kern.flash_attn_fwd
+def _fwd_fn(kern, sq=None, skv=None):
+    # v3 covers the training shape (S_q == S_kv); decode / ring half-blocks
+    # (S_q != S_kv) always go through v2, which supports them.
+    if _attn_version() == 3 and (sq is None or sq == skv):
+        return kern.flash_attn_fwd_v3
+    return kern.flash_attn_fwd
 
 
-def _bwd_fn(kern):
-    return kern.flash_attn_bwd_v3 if _attn_version() == 3 else kern.flash_attn_bwd
+def _bwd_fn(kern, sq=None, skv=None):
+    if _attn_version() == 3 and (sq is None or sq == skv):
+        return kern.flash_attn_bwd_v3
+    return kern.flash_attn_bwd
 
 
 def _make_mask(sq, sk, causal, window, device):
+    # S_q != S_kv: bottom-right alignment (query i sees keys
+    # j <= i + sk - sq) — matches the HIP kernel's decode convention
     mask = torch.zeros(sq, sk, dtype=torch.bool, device=device)
     if causal:
-        mask |= torch.ones(sq, sk, dtype=torch.bool, device=device).triu(1)
+        diag = sk - sq
+        mask |= torch.ones(sq, sk, dtype=torch.bool, device=device).triu(
+            1 + diag
+        )
         if window and window > 0:
             mask |= torch.ones(sq, sk, dtype=torch.bool, device=device).tril(
-                -window
+                diag - window
             )
     return mask
 
@@ -75,7 +86,9 @@ class _FlashAttnFn(torch.autograd.Function):
         kern = kernels_for(q)
         if kern is not None:
             # kernel takes arbitrary-strided [b, h, s, d] views (d contig)
-            o, lse = _fwd_fn(kern)(q, k, v, causal, scale, window)
+            o, lse = _fwd_fn(kern, q.size(2), k.size(2))(
+                q, k, v, causal, scale, window
+            )
         else:
             o, lse = _cpu_ref_fwd(q, k, v, causal, scale, window)
         ctx.save_for_backward(q, k, v, o, lse)
@@ -89,7 +102,7 @@ class _FlashAttnFn(torch.autograd.Function):
         q, k, v, o, lse = ctx.saved_tensors
         kern = kernels_for(q)
         if kern is not None:
-            dq, dk, dv = _bwd_fn(kern)(
+            dq, dk, dv = _bwd_fn(kern, q.size(2), k.size(2))(
                 do, q, k, v, o, lse, ctx.causal, ctx.scale, ctx.window
             )
             return dq, dk, dv, None, None, None

@@ -54,17 +54,28 @@ def _rotate_half(x):
     return torch.cat((-x2, x1), dim=-1)
 
 
+def _take_table(tab, off, off2, s):
+    """[s, d] slice of the table for offset off (rows [0, s/2)) and, when
+    off2 >= 0, offset off2 for rows [s/2, s) — the zigzag CP layout."""
+    if off2 is None or off2 < 0:
+        return tab[off : off + s]
+    h = s // 2
+    return torch.cat([tab[off : off + h], tab[off2 : off2 + h]], dim=0)
+
+
 class _RopeFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor, pos_offset: int):
+    def forward(ctx, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+                pos_offset: int, pos_offset2: int):
         ctx.save_for_backward(cos, sin)
         ctx.pos_offset = pos_offset
+        ctx.pos_offset2 = pos_offset2
         k = kernels_for(x)
         if k is not None:
-            return k.rope_fwd(x, cos, sin, pos_offset)
+            return k.rope_fwd(x, cos, sin, pos_offset, pos_offset2)
         s = x.size(-2)
-        c = cos[pos_offset : pos_offset + s].to(x.dtype)
-        sn = sin[pos_offset : pos_offset + s].to(x.dtype)
+        c = _take_table(cos, pos_offset, pos_offset2, s).to(x.dtype)
+        sn = _take_table(sin, pos_offset, pos_offset2, s).to(x.dtype)
         return x * c + _rotate_half(x) * sn
 
     @staticmethod
@@ -72,18 +83,25 @@ class _RopeFn(torch.autograd.Function):
         cos, sin = ctx.saved_tensors
         k = kernels_for(dy)
         if k is not None:
-            dx = k.rope_fwd(dy if dy.stride(-1) == 1 else dy.contiguous(), cos, -sin, ctx.pos_offset)
+            dx = k.rope_fwd(dy if dy.stride(-1) == 1 else dy.contiguous(),
+                            cos, -sin, ctx.pos_offset, ctx.pos_offset2)
         else:
             s = dy.size(-2)
-            c = cos[ctx.pos_offset : ctx.pos_offset + s].to(dy.dtype)
-            sn = -sin[ctx.pos_offset : ctx.pos_offset + s].to(dy.dtype)
+            c = _take_table(cos, ctx.pos_offset, ctx.pos_offset2, s).to(dy.dtype)
+            sn = -_take_table(sin, ctx.pos_offset, ctx.pos_offset2, s).to(dy.dtype)
             dx = dy * c + _rotate_half(dy) * sn
-        return dx, None, None, None
+        return dx, None, None, None, None
 
 
 def apply_rotary_pos_emb(
-    x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor, pos_offset: int = 0
+    x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor, pos_offset=0
 ) -> torch.Tensor:
     """x: [b, h, s, d]; cos/sin: [S, d] tables; pos_offset shifts positions
-    (context-parallel rank offset, reference modeling_llama.py:621-629)."""
-    return _RopeFn.apply(x, cos, sin, pos_offset)
+    (context-parallel rank offset, reference modeling_llama.py:621-629).
+    A (off_lo, off_hi) tuple applies the zigzag CP layout: the two halves
+    of the local sequence are the rank's two global chunks."""
+    if isinstance(pos_offset, (tuple, list)):
+        off, off2 = int(pos_offset[0]), int(pos_offset[1])
+    else:
+        off, off2 = int(pos_offset), -1
+    return _RopeFn.apply(x, cos, sin, off, off2)

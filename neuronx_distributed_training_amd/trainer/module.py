@@ -154,10 +154,12 @@ class BaseModelModule:
             # counts differ (the reference's mean-of-local-means is only
             # approximate there)
             batch["loss_denominator"] = mask.sum() / cp
+        from ..parallel.cp import cp_split
+
         out = {}
         for k, v in batch.items():
             if torch.is_tensor(v) and v.dim() >= 2 and v.size(1) == self.seq_length:
-                out[k] = v.chunk(cp, dim=1)[r].contiguous()
+                out[k] = cp_split(v, dim=1)
             else:
                 out[k] = v
         return out

@@ -112,6 +112,31 @@ def exp_manager(trainer, em_cfg: Dict) -> Tuple[List, Optional[str]]:
     loggers: List = [JsonlLogger(os.path.join(log_dir, "metrics.jsonl"))]
     if em_cfg.get("create_tensorboard_logger"):
         loggers.append(TensorBoardLogger(os.path.join(log_dir, "tb")))
+    if em_cfg.get("create_mlflow_logger"):
+        # reference exp_manager.py creates TB/W&B/MLflow side by side; the
+        # mlflow package is absent from this image, so the logger degrades
+        # to a local mlruns-style jsonl when the import fails.
+        try:
+            import mlflow  # guarded: not installed in this image
+
+            class _M:
+                def __init__(self):
+                    mlcfg = em_cfg.get("mlflow") or {}
+                    mlflow.set_tracking_uri(
+                        mlcfg.get("tracking_uri") or f"file:{log_dir}/mlruns"
+                    )
+                    mlflow.set_experiment(mlcfg.get("experiment_name", name))
+                    mlflow.start_run(run_name=mlcfg.get("run_name"))
+
+                def log_metrics(self, m, s):
+                    mlflow.log_metrics(
+                        {k: v for k, v in m.items() if isinstance(v, (int, float))},
+                        step=s,
+                    )
+
+            loggers.append(_M())
+        except Exception:
+            loggers.append(JsonlLogger(os.path.join(log_dir, "mlflow_fallback.jsonl")))
     if em_cfg.get("create_wandb_logger"):
         try:
             import wandb  # not installed in this image; guarded

@@ -319,3 +319,93 @@ def test_flash_attn_bwd_v3_gpu(ext):
         err = (got.float() - want).abs().max()
         sc = want.abs().max().clamp(min=1)
         assert err / sc < 0.05, f"{name} rel err {err / sc}"
+
+
+@pytest.mark.parametrize("sq,skv", [(1, 512), (128, 512), (257, 768)])
+def test_flash_attn_crosslen_fwd_gpu(ext, sq, skv):
+    """S_q != S_kv (KV-cache decode / ring half-block): bottom-right
+    causal — query i sees keys j <= i + skv - sq."""
+    from neuronx_distributed_training_amd.ops import flash_attn_func
+
+    torch.manual_seed(10)
+    b, hq, hkv, d = 2, 4, 2, 128
+    q = torch.randn(b, hq, sq, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(b, hkv, skv, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(b, hkv, skv, d, device="cuda", dtype=torch.bfloat16)
+    o = flash_attn_func(q, k, v, causal=True)
+    kx = k.repeat_interleave(hq // hkv, 1).float()
+    vx = v.repeat_interleave(hq // hkv, 1).float()
+    s = (q.float() @ kx.transpose(-1, -2)) / d ** 0.5
+    mask = torch.ones(sq, skv, dtype=torch.bool, device="cuda").triu(
+        1 + skv - sq
+    )
+    ref = torch.softmax(s.masked_fill(mask, float("-inf")), -1) @ vx
+    err = (o.float() - ref).abs().max()
+    assert err < 0.02, f"max err {err}"
+
+
+def test_flash_attn_crosslen_bwd_gpu(ext):
+    from neuronx_distributed_training_amd.ops import flash_attn_func
+
+    torch.manual_seed(11)
+    b, hq, hkv, sq, skv, d = 2, 4, 2, 192, 448, 128
+    q = torch.randn(b, hq, sq, d, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn(b, hkv, skv, d, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    v = torch.randn(b, hkv, skv, d, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    o = flash_attn_func(q, k, v, causal=True)
+    g = torch.randn_like(o)
+    o.backward(g)
+    qr = q.detach().float().clone().requires_grad_(True)
+    kr = k.detach().float().clone().requires_grad_(True)
+    vr = v.detach().float().clone().requires_grad_(True)
+    s = (qr @ kr.repeat_interleave(hq // hkv, 1).transpose(-1, -2)) / d ** 0.5
+    mask = torch.ones(sq, skv, dtype=torch.bool, device="cuda").triu(
+        1 + skv - sq
+    )
+    ref = torch.softmax(s.masked_fill(mask, float("-inf")), -1) @ \
+        vr.repeat_interleave(hq // hkv, 1)
+    ref.backward(g.float())
+    for got, want, name in ((q.grad, qr.grad, "dq"), (k.grad, kr.grad, "dk"),
+                            (v.grad, vr.grad, "dv")):
+        err = (got.float() - want).abs().max()
+        sc = want.abs().max().clamp(min=1)
+        assert err / sc < 0.05, f"{name} rel err {err / sc}"
+
+
+def test_flash_attn_longseq_gpu(ext):
+    """seq-8192 d=128 GQA fwd+bwd vs fp32 SDPA — the bench shape's
+    accuracy, not just loss-finite. Tolerances:
+    fwd 0.02 abs (bf16 output rounding ~0.004 at |o|<=1 plus online-
+    softmax re-scale error); bwd 5% of max-|grad| (bf16 P/dS round-trip
+    accumulated over 8k keys)."""
+    from neuronx_distributed_training_amd.ops import flash_attn_func
+
+    torch.manual_seed(12)
+    b, hq, hkv, s, d = 1, 4, 1, 8192, 128
+    q = torch.randn(b, hq, s, d, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn(b, hkv, s, d, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    v = torch.randn(b, hkv, s, d, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    o = flash_attn_func(q, k, v, causal=True)
+    g = torch.randn_like(o)
+    o.backward(g)
+    qr = q.detach().float().clone().requires_grad_(True)
+    kr = k.detach().float().clone().requires_grad_(True)
+    vr = v.detach().float().clone().requires_grad_(True)
+    ref = torch.nn.functional.scaled_dot_product_attention(
+        qr, kr.repeat_interleave(hq // hkv, 1),
+        vr.repeat_interleave(hq // hkv, 1), is_causal=True,
+    )
+    err = (o.float() - ref).abs().max()
+    assert err < 0.02, f"fwd max err {err}"
+    ref.backward(g.float())
+    for got, want, name in ((q.grad, qr.grad, "dq"), (k.grad, kr.grad, "dk"),
+                            (v.grad, vr.grad, "dv")):
+        e = (got.float() - want).abs().max()
+        sc = want.abs().max().clamp(min=1)
+        assert e / sc < 0.05, f"{name} rel err {e / sc}"
