@@ -119,7 +119,11 @@ class _RingFlashAttnFn(torch.autograd.Function):
         assert s_loc % 2 == 0, "zigzag CP needs an even local sequence"
         half = s_loc // 2
         k_cur, v_cur = k.contiguous(), v.contiguous()
-        k_buf, v_buf = torch.empty_like(k_cur), torch.empty_like(v_cur)
+        # TWO fresh recv buffers: the ping-pong must never receive into the
+        # original k/v (they are saved for backward — aliasing them
+        # corrupts the saved tensors from hop 2 on)
+        k_bufs = [torch.empty_like(k_cur), torch.empty_like(k_cur)]
+        v_bufs = [torch.empty_like(v_cur), torch.empty_like(v_cur)]
 
         acc_o = torch.zeros(q.shape, dtype=torch.float32, device=q.device)
         acc_lse = torch.full(
@@ -132,7 +136,9 @@ class _RingFlashAttnFn(torch.autograd.Function):
                 j = (r - t) % R
                 works: List = []
                 if t < R - 1:
-                    works = _post_shift((k_cur, v_cur), (k_buf, v_buf))
+                    works = _post_shift(
+                        (k_cur, v_cur), (k_bufs[t % 2], v_bufs[t % 2])
+                    )
                 if j == r:
                     o, lse = _fwd_block(kern, q, k_cur, v_cur, True, scale)
                     _merge_into(acc_o, acc_lse, o, lse)
@@ -150,8 +156,7 @@ class _RingFlashAttnFn(torch.autograd.Function):
                 for w in works:
                     w.wait()
                 if t < R - 1:
-                    k_cur, k_buf = k_buf, k_cur
-                    v_cur, v_buf = v_buf, v_cur
+                    k_cur, v_cur = k_bufs[t % 2], v_bufs[t % 2]
             out = acc_o.to(q.dtype)
         ctx.save_for_backward(q, k, v, out, acc_lse)
         ctx.scale = scale
@@ -168,7 +173,8 @@ class _RingFlashAttnFn(torch.autograd.Function):
         dout = dout.contiguous()
         out = out.contiguous()
         k_cur, v_cur = k.contiguous(), v.contiguous()
-        k_buf, v_buf = torch.empty_like(k_cur), torch.empty_like(v_cur)
+        k_bufs = [torch.empty_like(k_cur), torch.empty_like(k_cur)]
+        v_bufs = [torch.empty_like(v_cur), torch.empty_like(v_cur)]
         # the K/V-block gradient accumulators ride the ring with their
         # block; ping-pong buffers so the isend source stays untouched
         dk_cur = torch.zeros_like(k_cur)
@@ -184,8 +190,10 @@ class _RingFlashAttnFn(torch.autograd.Function):
             j = (r - t) % R
             kv_works: List = []
             if t < R - 1:
-                # prefetch next hop's K/V and its in-flight accumulator
-                kv_works = _post_shift((k_cur, v_cur), (k_buf, v_buf))
+                # prefetch next hop's K/V while this block computes
+                kv_works = _post_shift(
+                    (k_cur, v_cur), (k_bufs[t % 2], v_bufs[t % 2])
+                )
             lo_only = j < r  # only this block's LOW chunk was visible
             if j == r:
                 dq_j, dk_j, dv_j = _bwd_block(
@@ -222,8 +230,7 @@ class _RingFlashAttnFn(torch.autograd.Function):
             for w in kv_works:
                 w.wait()
             if t < R - 1:
-                k_cur, k_buf = k_buf, k_cur
-                v_cur, v_buf = v_buf, v_cur
+                k_cur, v_cur = k_bufs[t % 2], v_bufs[t % 2]
         for w in acc_works:
             w.wait()
         return dq_acc, dk_buf, dv_buf, None

@@ -13,19 +13,20 @@ def _ring_attn(rank, world):
     from neuronx_distributed_training_amd.ops.ring_attn import ring_flash_attn
 
     ps.initialize_model_parallel(context_parallel_size=world)
+    from neuronx_distributed_training_amd.parallel.cp import cp_split
+
     torch.manual_seed(0)
     b, h, hkv, s, d = 2, 4, 2, 32, 16
     qf = torch.randn(b, h, s, d)
     kf = torch.randn(b, hkv, s, d)
     vf = torch.randn(b, hkv, s, d)
     g = torch.randn(b, h, s, d)
-    r = ps.get_context_model_parallel_rank()
-    sl = s // world
-    q = qf[:, :, r * sl : (r + 1) * sl].clone().requires_grad_(True)
-    k = kf[:, :, r * sl : (r + 1) * sl].clone().requires_grad_(True)
-    v = vf[:, :, r * sl : (r + 1) * sl].clone().requires_grad_(True)
+    # zigzag placement: rank r holds global chunks (r, 2*world-1-r)
+    q = cp_split(qf, dim=2).clone().requires_grad_(True)
+    k = cp_split(kf, dim=2).clone().requires_grad_(True)
+    v = cp_split(vf, dim=2).clone().requires_grad_(True)
     o = ring_flash_attn(q, k, v)
-    o.backward(g[:, :, r * sl : (r + 1) * sl])
+    o.backward(cp_split(g, dim=2))
 
     # full-sequence reference
     qr = qf.clone().requires_grad_(True)
@@ -36,12 +37,12 @@ def _ring_attn(rank, world):
         is_causal=True,
     )
     ref.backward(g)
-    assert torch.allclose(o, ref[:, :, r * sl : (r + 1) * sl], atol=1e-4), (
-        (o - ref[:, :, r * sl : (r + 1) * sl]).abs().max()
+    assert torch.allclose(o, cp_split(ref.detach(), dim=2), atol=1e-4), (
+        (o - cp_split(ref.detach(), dim=2)).abs().max()
     )
-    assert torch.allclose(q.grad, qr.grad[:, :, r * sl : (r + 1) * sl], atol=1e-4)
-    assert torch.allclose(k.grad, kr.grad[:, :, r * sl : (r + 1) * sl], atol=1e-4)
-    assert torch.allclose(v.grad, vr.grad[:, :, r * sl : (r + 1) * sl], atol=1e-4)
+    assert torch.allclose(q.grad, cp_split(qr.grad, dim=2), atol=1e-4)
+    assert torch.allclose(k.grad, cp_split(kr.grad, dim=2), atol=1e-4)
+    assert torch.allclose(v.grad, cp_split(vr.grad, dim=2), atol=1e-4)
     return 0.0
 
 
