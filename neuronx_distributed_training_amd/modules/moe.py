@@ -261,10 +261,15 @@ class ExpertMLPs(nn.Module):
         """x: [T, H] tokens sorted by local expert; counts: [E_local] token
         counts per local expert. Returns same-order outputs.
 
-        Balanced loads take a single-bmm path (one batched hipBLASLt call
-        per projection instead of E_local separate GEMMs — fewer launches,
-        no tail effects); skewed loads fall back to per-expert GEMMs so
-        padding waste stays bounded."""
+        On GPU the grouped-GEMM HIP kernel runs all experts in one launch
+        per projection with DEVICE-side layout math — no per-expert loop,
+        no .tolist() host sync. CPU/odd-shape
+        fallback: balanced loads take a single-bmm path, skewed loads
+        per-expert GEMMs."""
+        from ..ops.moe_gemm import grouped_expert_mlp, grouped_path_supported
+
+        if x.numel() and grouped_path_supported(x, self.gate_up, self.down):
+            return grouped_expert_mlp(x, counts, self.gate_up, self.down)
         cl = counts.tolist()
         total = int(sum(cl))
         if total == 0:

@@ -34,6 +34,10 @@ void launch_flash_bwd_v3(const void*, const void*, const void*, const void*,
                          const void*, const void*, void*, void*, void*, int,
                          int, int, int, int, bool, float, int, const long*,
                          const long*, const long*, const long*, hipStream_t);
+void launch_moe_gemm(const void*, const void*, void*, const int*, const int*,
+                     int, int, int, long, bool, hipStream_t);
+void launch_moe_wgrad(const void*, const void*, void*, const int*, int, int,
+                      int, hipStream_t);
 void launch_ce_fwd(const void*, const void*, void*, void*, void*, long, int,
                    long, hipStream_t);
 void launch_ce_bwd(const void*, const void*, const void*, const void*,
@@ -278,6 +282,50 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
           dv_mem.permute({1, 2, 0, 3})};
 }
 
+torch::Tensor moe_gemm(torch::Tensor a, torch::Tensor w,
+                       torch::Tensor tile_expert, torch::Tensor total_rows,
+                       bool trans_b) {
+  // a: [Tp, K] bf16 padded/sorted tokens; w: [E, N, Kw] bf16 expert slabs.
+  // trans_b=false: C = a @ w[e]^T ([Tp, N]); true: C = a @ w[e] ([Tp, Kw]).
+  CHECK_IN(a);
+  CHECK_IN(w);
+  TORCH_CHECK(a.scalar_type() == torch::kBFloat16 &&
+                  w.scalar_type() == torch::kBFloat16,
+              "moe_gemm: bf16 only");
+  TORCH_CHECK(tile_expert.scalar_type() == torch::kInt &&
+                  total_rows.scalar_type() == torch::kInt,
+              "moe_gemm: int32 maps");
+  long Tp = a.size(0);
+  int K = (int)a.size(1);
+  int N = (int)w.size(1), Kw = (int)w.size(2);
+  TORCH_CHECK(Tp % 256 == 0, "padded rows must be a multiple of 256");
+  int row_tiles = (int)(Tp / 256);
+  int contraction = trans_b ? N : Kw;
+  int out_cols = trans_b ? Kw : N;
+  TORCH_CHECK(K == contraction, "moe_gemm: contraction mismatch");
+  TORCH_CHECK(out_cols % 128 == 0, "moe_gemm: out cols % 128");
+  auto c = torch::empty({Tp, out_cols}, a.options());
+  launch_moe_gemm(a.data_ptr(), w.data_ptr(), c.data_ptr(),
+                  tile_expert.data_ptr<int>(), total_rows.data_ptr<int>(),
+                  row_tiles, K, out_cols, (long)N * Kw, trans_b,
+                  cur_stream());
+  return c;
+}
+
+torch::Tensor moe_wgrad(torch::Tensor dy, torch::Tensor x,
+                        torch::Tensor seg_start, long E) {
+  // dW[e] = dy[seg_e]^T @ x[seg_e] -> [E, N, K] fp32
+  CHECK_IN(dy);
+  CHECK_IN(x);
+  TORCH_CHECK(seg_start.scalar_type() == torch::kInt, "int32 seg bounds");
+  int N = (int)dy.size(1), K = (int)x.size(1);
+  TORCH_CHECK(N % 128 == 0 && K % 128 == 0, "moe_wgrad: dims % 128");
+  auto dw = torch::empty({E, N, K}, dy.options().dtype(torch::kFloat32));
+  launch_moe_wgrad(dy.data_ptr(), x.data_ptr(), dw.data_ptr(),
+                   seg_start.data_ptr<int>(), (int)E, N, K, cur_stream());
+  return dw;
+}
+
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b) {
   CHECK_IN(a);
   CHECK_IN(b);
@@ -356,4 +404,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("scale"), pybind11::arg("window") = 0);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
+  m.def("moe_gemm", &moe_gemm);
+  m.def("moe_wgrad", &moe_wgrad);
 }

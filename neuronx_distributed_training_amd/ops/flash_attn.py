@@ -23,28 +23,29 @@ import torch
 from . import kernels_for
 
 
-def _attn_version() -> int:
-    """Kernel generation: 2 = 16x16x32 MFMA 8-wave (round-1 default),
-    3 = 32x32x16 swapped-operand register-softmax (ROADMAP §1).
-    NXDT_ATTN_V3=1/0 overrides."""
+def _attn_version() -> str:
+    """Kernel generation: '2' = 16x16x32 MFMA 8-wave (default), '3' =
+    32x32x16 swapped-operand register-softmax, 'fwd' = v3 forward with v2
+    backward (r2 A/B: v3 end-to-end measured slower than v2).
+    NXDT_ATTN_V3=1/0/fwd overrides."""
     env = os.environ.get("NXDT_ATTN_V3", "")
     if env == "1":
-        return 3
-    if env == "0":
-        return 2
-    return 2  # default until the v3 A/B shows a win on hardware
+        return "3"
+    if env == "fwd":
+        return "fwd"
+    return "2"
 
 
 def _fwd_fn(kern, sq=None, skv=None):
     # v3 covers the training shape (S_q == S_kv); decode / ring half-blocks
     # (S_q != S_kv) always go through v2, which supports them.
-    if _attn_version() == 3 and (sq is None or sq == skv):
+    if _attn_version() in ("3", "fwd") and (sq is None or sq == skv):
         return kern.flash_attn_fwd_v3
     return kern.flash_attn_fwd
 
 
 def _bwd_fn(kern, sq=None, skv=None):
-    if _attn_version() == 3 and (sq is None or sq == skv):
+    if _attn_version() == "3" and (sq is None or sq == skv):
         return kern.flash_attn_bwd_v3
     return kern.flash_attn_bwd
 

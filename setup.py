@@ -30,6 +30,7 @@ sources = [
         "flash_attn_bwd_v3.hip",
         "flash_attn_bwd.hip",
         "cross_entropy.hip",
+        "moe_gemm.hip",
         "probe.hip",
     )
 ]
