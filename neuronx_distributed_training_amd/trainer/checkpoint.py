@@ -48,24 +48,49 @@ def _optim_shard_name() -> str:
     return f"dp_rank_{dp:02d}_tp_rank_{tp:02d}_pp_rank_{pp:02d}.pt"
 
 
-def _cpu_copy_inner(obj):
+def _cpu_copy_inner(obj, pin: bool):
     if torch.is_tensor(obj):
-        # overlapped D2H: copies queue on the stream; one synchronize at the
-        # end of _cpu_copy instead of a blocking round-trip per tensor
+        if obj.is_cuda and pin:
+            # pinned staging → true async D2H on the side stream (pageable
+            # destinations force a synchronous copy in HIP)
+            dst = torch.empty(
+                obj.shape, dtype=obj.dtype, device="cpu", pin_memory=True
+            )
+            dst.copy_(obj.detach(), non_blocking=True)
+            return dst
         return obj.detach().to("cpu", non_blocking=True)
     if isinstance(obj, dict):
-        return {k: _cpu_copy_inner(v) for k, v in obj.items()}
+        return {k: _cpu_copy_inner(v, pin) for k, v in obj.items()}
     if isinstance(obj, (list, tuple)):
         t = type(obj)
-        return t(_cpu_copy_inner(v) for v in obj)
+        return t(_cpu_copy_inner(v, pin) for v in obj)
     return obj
 
 
+_COPY_STREAM = None
+
+
 def _cpu_copy(obj):
-    out = _cpu_copy_inner(obj)
-    if torch.cuda.is_available():
-        torch.cuda.synchronize()
-    return out
+    """D2H snapshot. On GPU the copies run on a dedicated side stream into
+    pinned buffers and this returns WITHOUT waiting — the returned event
+    gates the writer (reference async_checkpointing: the training loop
+    must not stall for checkpoint IO, known_issues.rst:55-85). Returns
+    (cpu_obj, event_or_None)."""
+    global _COPY_STREAM
+    if not (torch.cuda.is_available() and torch.cuda.is_initialized()):
+        return _cpu_copy_inner(obj, pin=False), None
+    if _COPY_STREAM is None:
+        _COPY_STREAM = torch.cuda.Stream()
+    _COPY_STREAM.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(_COPY_STREAM):
+        out = _cpu_copy_inner(obj, pin=True)
+        ev = torch.cuda.Event()
+        ev.record(_COPY_STREAM)
+    # GPU-side ordering only: later kernels (the next optimizer step
+    # mutates these params in place) queue behind the copies, but the
+    # HOST thread returns immediately instead of torch.cuda.synchronize()
+    torch.cuda.current_stream().wait_event(ev)
+    return out, ev
 
 
 def _remote_fs(path: str):
@@ -95,11 +120,15 @@ def _touch_done(root: str, fs):
 
 
 class CheckpointIO:
-    def __init__(self, async_save: bool = False, save_bf16: bool = False):
+    def __init__(self, async_save: bool = False, save_bf16: bool = False,
+                 writer_process: bool = False):
         self.async_save = async_save
         # reference exp_manager `save_bf16`: cast fp32 model tensors to
         # bf16 on save (halves shard size; optimizer masters stay fp32)
         self.save_bf16 = save_bf16
+        # fork a writer process per save so torch.save's pickling (which
+        # holds the GIL) leaves the training process entirely
+        self.writer_process = writer_process
         self._pending: List[threading.Thread] = []
 
     # ---- save ----
@@ -123,22 +152,29 @@ class CheckpointIO:
 
         dp, tp, pp = _rank_tags()
         work = []
+        events = []
         if dp == 0 and ps.get_context_model_parallel_rank() == 0:
-            msd = _cpu_copy(module.model.state_dict())
+            msd, ev = _cpu_copy(module.model.state_dict())
+            events.append(ev)
             if self.save_bf16:
-                msd = {
-                    k: (v.to(torch.bfloat16)
-                        if torch.is_tensor(v) and v.dtype == torch.float32
-                        else v)
-                    for k, v in msd.items()
-                }
+                # the cast reads the pinned copies — it must run after the
+                # D2H completes, so it moves into the writer
+                def _cast(sd=msd):
+                    return {
+                        k: (v.to(torch.bfloat16)
+                            if torch.is_tensor(v) and v.dtype == torch.float32
+                            else v)
+                        for k, v in sd.items()
+                    }
+                msd = _cast
             work.append(
                 (_join(root, fs, "model", _model_shard_name()), msd)
             )
         if module.optimizer is not None and ps.get_context_model_parallel_rank() == 0:
+            osd, ev = _cpu_copy(module.optimizer.state_dict())
+            events.append(ev)
             work.append(
-                (_join(root, fs, "optim", _optim_shard_name()),
-                 _cpu_copy(module.optimizer.state_dict()))
+                (_join(root, fs, "optim", _optim_shard_name()), osd)
             )
         is_global_zero = (not dist.is_initialized()) or dist.get_rank() == 0
         if is_global_zero:
@@ -147,15 +183,39 @@ class CheckpointIO:
                 uc["scheduler"] = module.scheduler.state_dict()
             work.append((_join(root, fs, "user_content.pt"), uc))
 
+        use_fork = self.writer_process and fs is None and hasattr(os, "fork")
+
         def _write():
-            for path, obj in work:
-                if fs is not None:
-                    with fs.open(path, "wb") as f:
-                        torch.save(obj, f)
-                else:
-                    tmp = path + ".tmp"
-                    torch.save(obj, tmp)
-                    os.replace(tmp, path)
+            for ev in events:
+                if ev is not None:
+                    ev.synchronize()  # pinned copies durable before save
+            items = [(p, o() if callable(o) else o) for p, o in work]
+
+            def _dump():
+                for path, obj in items:
+                    if fs is not None:
+                        with fs.open(path, "wb") as f:
+                            torch.save(obj, f)
+                    else:
+                        tmp = path + ".tmp"
+                        torch.save(obj, tmp)
+                        os.replace(tmp, path)
+
+            if use_fork:
+                # reference async semantics: the saver is a separate
+                # PROCESS (nlp_overrides.py:618-627) — serialization cost
+                # (pickling holds the GIL) leaves the training process.
+                # The forked child only touches CPU memory and files,
+                # never the HIP context.
+                pid = os.fork()
+                if pid == 0:
+                    try:
+                        _dump()
+                    finally:
+                        os._exit(0)
+                os.waitpid(pid, 0)
+            else:
+                _dump()
 
         if self.async_save:
             t = threading.Thread(target=_write, daemon=False)

@@ -51,6 +51,8 @@ class Trainer:
         self.ckpt_io = CheckpointIO(
             async_save=self.async_save,
             save_bf16=bool(em.get("save_bf16", False)),
+            writer_process=bool(em.get("async_checkpointing_use_process",
+                                       False)),
         )
         self.global_step = 0
 

@@ -470,3 +470,45 @@ def test_resume_is_bit_exact(tmp_path):
     run_distributed(_resume_exact, 1, d, "first")
     resumed = run_distributed(_resume_exact, 1, d, "second")[0]
     assert torch.equal(straight, resumed), (straight - resumed).abs().max()
+
+
+def _fit_fork_writer_ckpt(rank, world, tmpdir):
+    """async_checkpointing_use_process: the save serializes in a forked
+    writer process (reference async-saver-process semantics,
+    nlp_overrides.py:618-627); the committed checkpoint must round-trip."""
+    import os
+    import torch
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.trainer import Trainer
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+    from neuronx_distributed_training_amd.trainer.checkpoint import (
+        CheckpointIO, find_latest_checkpoint,
+    )
+    from neuronx_distributed_training_amd.data.datamodule import build_datamodule
+
+    ps.initialize_model_parallel()
+    torch.manual_seed(0)
+    cfg = _cfg(tmpdir, 3)
+    cfg["exp_manager"]["async_checkpointing"] = True
+    cfg["exp_manager"]["async_checkpointing_use_process"] = True
+    tr = Trainer(cfg)
+    tr.ckpt_dir = os.path.join(tmpdir, "checkpoints")
+    os.makedirs(tr.ckpt_dir, exist_ok=True)
+    module = LlamaModule(cfg)
+    dm = build_datamodule(cfg)
+    tr.fit(module, dm)
+    latest = find_latest_checkpoint(tr.ckpt_dir)
+    assert latest is not None
+    m2 = LlamaModule(cfg)
+    m2.setup()
+    m2.configure_optimizers(max_steps=3)
+    CheckpointIO().load(latest, m2)
+    for (n1, p1), (n2, p2) in zip(
+        module.model.named_parameters(), m2.model.named_parameters()
+    ):
+        assert torch.equal(p1, p2), n1
+    return 1
+
+
+def test_fork_writer_checkpoint(tmp_path):
+    assert run_distributed(_fit_fork_writer_ckpt, 1, str(tmp_path))[0] == 1
