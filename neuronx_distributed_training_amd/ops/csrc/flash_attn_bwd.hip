@@ -161,20 +161,43 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
           const int qcol = qbase + nq * 16 + (lane & 15);
           const float lse = (qcol < SQ) ? Lp[qcol] : 1e30f;
           const float delta = (qcol < SQ) ? Dp[qcol] : 0.f;
+          // interior fast path (the fwd kernel's full_tile analog): every
+          // (q, key) of this 16-col x 32-key patch in bounds and causally
+          // alive — skip the 3 compare/selects per element (VALU-bound
+          // kernel)
+          const bool full_patch =
+              (qbase + nq * 16 + 15 < SQ) && (krow_w + 31 < SKV) &&
+              (!CAUSAL || (qbase + nq * 16 + coff >= krow_w + 31)) &&
+              (window <= 0 ||
+               (qbase + nq * 16 + 15 + coff < krow_w + window));
+          if (full_patch) {
 #pragma unroll
-          for (int sb = 0; sb < 2; ++sb) {
+            for (int sb = 0; sb < 2; ++sb) {
 #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-              const int krow = krow_w + sb * 16 + (lane >> 4) * 4 + r;
-              bool dead = (qcol >= SQ) || (krow >= SKV) ||
-                          (CAUSAL && qcol + coff < krow);
-              if (CAUSAL && window > 0)
-                dead |= (qcol + coff >= krow + window);
-              const float p = dead ? 0.f : __expf(st[sb][r] - lse);
-              const float ds = p * (dpt[sb][r] - delta) * scale;
-              const int lrow = sb * 16 + (lane >> 4) * 4 + r;
-              pw[lrow * VP + nq * 16 + (lane & 15)] = (__bf16)p;
-              dw[lrow * VP + nq * 16 + (lane & 15)] = (__bf16)ds;
+              for (int r = 0; r < 4; ++r) {
+                const float p = __expf(st[sb][r] - lse);
+                const float ds = p * (dpt[sb][r] - delta) * scale;
+                const int lrow = sb * 16 + (lane >> 4) * 4 + r;
+                pw[lrow * VP + nq * 16 + (lane & 15)] = (__bf16)p;
+                dw[lrow * VP + nq * 16 + (lane & 15)] = (__bf16)ds;
+              }
+            }
+          } else {
+#pragma unroll
+            for (int sb = 0; sb < 2; ++sb) {
+#pragma unroll
+              for (int r = 0; r < 4; ++r) {
+                const int krow = krow_w + sb * 16 + (lane >> 4) * 4 + r;
+                bool dead = (qcol >= SQ) || (krow >= SKV) ||
+                            (CAUSAL && qcol + coff < krow);
+                if (CAUSAL && window > 0)
+                  dead |= (qcol + coff >= krow + window);
+                const float p = dead ? 0.f : __expf(st[sb][r] - lse);
+                const float ds = p * (dpt[sb][r] - delta) * scale;
+                const int lrow = sb * 16 + (lane >> 4) * 4 + r;
+                pw[lrow * VP + nq * 16 + (lane & 15)] = (__bf16)p;
+                dw[lrow * VP + nq * 16 + (lane & 15)] = (__bf16)ds;
+              }
             }
           }
         }
@@ -354,18 +377,36 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
           dpt[1] = MFMA_16x16x32(dofrag[1][kk], vb, dpt[1]);
         }
         const int kcol = kbase + nk * 16 + (lane & 15);
+        // interior fast path: whole 32-row x 16-key patch alive (see dkv)
+        const bool full_patch =
+            (kbase + nk * 16 + 15 < SKV) &&
+            (!CAUSAL || (kbase + nk * 16 + 15 <= qrow_w + coff)) &&
+            (window <= 0 || (kbase + nk * 16 > qrow_w + 31 + coff - window));
+        if (full_patch) {
 #pragma unroll
-        for (int sb = 0; sb < 2; ++sb) {
+          for (int sb = 0; sb < 2; ++sb) {
 #pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            const int qrow = qrow_w + sb * 16 + (lane >> 4) * 4 + r;
-            bool dead = (kcol >= SKV) || (CAUSAL && kcol > qrow + coff);
-            if (CAUSAL && window > 0)
-              dead |= (kcol <= qrow + coff - window);
-            const float p = dead ? 0.f : __expf(st[sb][r] - lse[sb][r]);
-            const float ds = p * (dpt[sb][r] - delta[sb][r]) * scale;
-            dsw[(sb * 16 + (lane >> 4) * 4 + r) * VP + nk * 16 +
-                (lane & 15)] = (__bf16)ds;
+            for (int r = 0; r < 4; ++r) {
+              const float p = __expf(st[sb][r] - lse[sb][r]);
+              const float ds = p * (dpt[sb][r] - delta[sb][r]) * scale;
+              dsw[(sb * 16 + (lane >> 4) * 4 + r) * VP + nk * 16 +
+                  (lane & 15)] = (__bf16)ds;
+            }
+          }
+        } else {
+#pragma unroll
+          for (int sb = 0; sb < 2; ++sb) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              const int qrow = qrow_w + sb * 16 + (lane >> 4) * 4 + r;
+              bool dead = (kcol >= SKV) || (CAUSAL && kcol > qrow + coff);
+              if (CAUSAL && window > 0)
+                dead |= (kcol <= qrow + coff - window);
+              const float p = dead ? 0.f : __expf(st[sb][r] - lse[sb][r]);
+              const float ds = p * (dpt[sb][r] - delta[sb][r]) * scale;
+              dsw[(sb * 16 + (lane >> 4) * 4 + r) * VP + nk * 16 +
+                  (lane & 15)] = (__bf16)ds;
+            }
           }
         }
       }

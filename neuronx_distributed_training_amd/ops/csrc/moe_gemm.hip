@@ -36,10 +36,11 @@ __global__ __launch_bounds__(512) void moe_gemm_kernel(
     long wstride) {  // elements per expert slab (N * Kw)
   constexpr int BM = 256, BN = 128, BK = 32;
   constexpr int AP = BK + 8;
-  constexpr int BP = BK + 8;   // row-major B tile row pitch
-  constexpr int TP = BN + 8;   // transposed B image pitch
+  // B tile: row-major [BN n-rows][BK k] (fwd) or transposed image
+  // [BN k-rows][BK n] (dgrad) — both BN rows x BP pitch
+  constexpr int BP = BK + 8;
   __shared__ __bf16 a_lds[BM * AP];
-  __shared__ __bf16 b_lds[TRANS_B ? (BK * TP) : (BN * BP)];
+  __shared__ __bf16 b_lds[BN * BP];
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -105,7 +106,7 @@ __global__ __launch_bounds__(512) void moe_gemm_kernel(
           for (int j = 0; j < 8; ++j) {
             __bf16 pr[2] = {e0[j], e1[j]};
             const int r = col8 + j;  // k index = image row
-            *(uint*)((char*)b_lds + tr_swz((uint)(r * TP + row) * 2, r)) =
+            *(uint*)((char*)b_lds + tr_swz((uint)(r * BP + row) * 2, r)) =
                 *(uint*)pr;
           }
         }
@@ -123,7 +124,7 @@ __global__ __launch_bounds__(512) void moe_gemm_kernel(
         if (!TRANS_B)
           bf = load_frag_b_rowmajorT(&b_lds[nj * 16 * BP], BP, kk * 32, lane);
         else
-          bf = load_frag_b_trT_swz(b_lds, TP, nj * 16, kk * 32, lane);
+          bf = load_frag_b_trT_swz(b_lds, BP, nj * 16, kk * 32, lane);
         acc[0][nj] = MFMA_16x16x32(a0, bf, acc[0][nj]);
         acc[1][nj] = MFMA_16x16x32(a1, bf, acc[1][nj]);
       }

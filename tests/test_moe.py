@@ -423,3 +423,33 @@ def test_ep2_dp2_expert_weights_match_single():
     a1 = run_distributed(_ep_dp_weights, 1)[0]
     a2 = [r for r in run_distributed(_ep_dp_weights, 4) if r is not None][0]
     assert torch.allclose(a1, a2, atol=1e-4), (a1 - a2).abs().max()
+
+
+def test_moe_gemm_layout_device_math():
+    """The grouped-GEMM layout helper: padded scatter rows, tile->expert
+    map and totals — all computed without any host sync (CPU check of the
+    same torch ops that run on device)."""
+    import torch
+    from neuronx_distributed_training_amd.ops.moe_gemm import _layout, BM
+
+    counts = torch.tensor([700, 0, 13, 301])
+    T = int(counts.sum())
+    pr, tile_e, pad_off, total, Tp = _layout(counts, T)
+    padded = [(int(c) + BM - 1) // BM * BM for c in counts]
+    assert pad_off.tolist() == [0] + torch.cumsum(
+        torch.tensor(padded), 0
+    ).tolist()
+    assert int(total) == sum(padded)
+    assert Tp >= int(total) and Tp % BM == 0
+    # scatter rows: token i of expert e lands at pad_off[e] + rank-in-e
+    start = 0
+    for e, c in enumerate(counts.tolist()):
+        seg = pr[start:start + c]
+        assert seg.tolist() == list(
+            range(int(pad_off[e]), int(pad_off[e]) + c)
+        )
+        start += c
+    # every valid tile maps to the expert owning its rows
+    for t in range(int(total) // BM):
+        e = int(tile_e[t])
+        assert int(pad_off[e]) <= t * BM < int(pad_off[e + 1])
