@@ -25,6 +25,8 @@ void launch_flash_bwd(const void*, const void*, const void*, const void*,
                       const void*, const void*, void*, void*, void*, int, int,
                       int, int, int, int, bool, float, int, const long*,
                       const long*, const long*, const long*, hipStream_t);
+void launch_attn_delta(const void*, const void*, void*, int, int, int, int,
+                       const long*, const long*, hipStream_t);
 void launch_mfma_probe(const void*, const void*, void*, hipStream_t);
 void launch_mfma_probe32(const void*, const void*, void*, hipStream_t);
 void launch_flash_fwd_v3(const void*, const void*, const void*, void*, void*,
@@ -159,6 +161,20 @@ static void check_bhsd(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
 }
 
+static torch::Tensor attn_delta(torch::Tensor dout, torch::Tensor o) {
+  // fused rowsum(dO . O) -> [B, HQ, SQ] fp32 (no fp32 materialization)
+  auto oc = o.stride(3) == 1 ? o : o.contiguous();
+  int B = (int)dout.size(0), HQ = (int)dout.size(1), SQ = (int)dout.size(2),
+      D = (int)dout.size(3);
+  auto delta = torch::empty(
+      {B, HQ, SQ}, dout.options().dtype(torch::kFloat32));
+  long ds[3] = {dout.stride(2), dout.stride(0), dout.stride(1)};
+  long os_[3] = {oc.stride(2), oc.stride(0), oc.stride(1)};
+  launch_attn_delta(dout.data_ptr(), oc.data_ptr(), delta.data_ptr(), SQ, B,
+                    HQ, D, ds, os_, cur_stream());
+  return delta;
+}
+
 std::vector<torch::Tensor> flash_attn_fwd_v3(torch::Tensor q, torch::Tensor k,
                                              torch::Tensor v, bool causal,
                                              double scale, long window) {
@@ -194,9 +210,7 @@ std::vector<torch::Tensor> flash_attn_bwd_v3(torch::Tensor dout,
   int B = (int)q.size(0), HQ = (int)q.size(1), S = (int)q.size(2),
       D = (int)q.size(3);
   int HKV = (int)k.size(1);
-  auto delta = (dout.to(torch::kFloat32) * o.to(torch::kFloat32))
-                   .sum(-1)
-                   .contiguous();
+  auto delta = attn_delta(dout, o);
   auto dq_mem = torch::empty({S, B, HQ, D}, q.options());
   int group = HQ / HKV;
   auto f32 = q.options().dtype(torch::kFloat32);
@@ -257,9 +271,7 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor dout, torch::Tensor q,
       D = (int)q.size(3);
   int HKV = (int)k.size(1), SKV = (int)k.size(2);
   TORCH_CHECK(!causal || SKV >= SQ, "causal needs S_kv >= S_q");
-  auto delta = (dout.to(torch::kFloat32) * o.to(torch::kFloat32))
-                   .sum(-1)
-                   .contiguous();  // [B, HQ, SQ] f32
+  auto delta = attn_delta(dout, o);  // [B, HQ, SQ] f32, fused
   auto dq_mem = torch::empty({SQ, B, HQ, D}, q.options());
   // dkv writes one fp32 partial slab per q-head of each GQA group (full
   // grid occupancy at high TP); sum + cast here

@@ -441,7 +441,53 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
       }
 }
 
+// ===================== delta = rowsum(dO . O) =====================
+// One wave per (b, h, s) row: fused bf16 reads, fp32 dot, no fp32
+// materialization of dO/O (the torch expression cast+mul+sum moved
+// ~3.5 GB per call at the bench shape).
+template <int D>
+__global__ __launch_bounds__(256) void attn_delta_kernel(
+    const bf16* __restrict__ dO, const bf16* __restrict__ O,
+    float* __restrict__ DELTA, int SQ, int Bb, int HQ, long sDs, long sDb,
+    long sDh, long sOs, long sOb, long sOh) {
+  const int lane = threadIdx.x & 63;
+  const long row = ((long)blockIdx.x * 4 + (threadIdx.x >> 6));
+  const long total = (long)SQ * Bb * HQ;
+  if (row >= total) return;
+  // row order matches DELTA [b][h][s]
+  const int s = (int)(row % SQ);
+  const int h = (int)((row / SQ) % HQ);
+  const int b = (int)(row / ((long)SQ * HQ));
+  const bf16* dp = dO + b * sDb + h * sDh + s * sDs;
+  const bf16* op = O + b * sOb + h * sOh + s * sOs;
+  float acc = 0.f;
+#pragma unroll
+  for (int j = 0; j < D / 64; ++j) {
+    const int c = lane + j * 64;
+    acc += bf2f(dp[c]) * bf2f(op[c]);
+  }
+#pragma unroll
+  for (int off = 1; off < 64; off <<= 1) acc += __shfl_xor(acc, off, 64);
+  if (lane == 0) DELTA[row] = acc;
+}
+
 extern "C" {
+void launch_attn_delta(const void* dout, const void* o, void* delta, int SQ,
+                       int B, int HQ, int D, const long* dstr,
+                       const long* ostr, hipStream_t stream) {
+  long total = (long)SQ * B * HQ;
+  dim3 grid((unsigned)((total + 3) / 4));
+  dim3 blk(256);
+  if (D == 128)
+    attn_delta_kernel<128><<<grid, blk, 0, stream>>>(
+        (const bf16*)dout, (const bf16*)o, (float*)delta, SQ, B, HQ, dstr[0],
+        dstr[1], dstr[2], ostr[0], ostr[1], ostr[2]);
+  else if (D == 64)
+    attn_delta_kernel<64><<<grid, blk, 0, stream>>>(
+        (const bf16*)dout, (const bf16*)o, (float*)delta, SQ, B, HQ, dstr[0],
+        dstr[1], dstr[2], ostr[0], ostr[1], ostr[2]);
+}
+
 void launch_flash_bwd(const void* dout, const void* q, const void* k,
                       const void* v, const void* lse, const void* delta,
                       void* dq, void* dk, void* dv, int B, int HQ, int HKV,
