@@ -26,7 +26,11 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
     int window, long sQs, long sQb, long sQh, long sKs, long sKb, long sKh,
     long sVs, long sVb, long sVh, long sDs, long sDb, long sDh) {
   constexpr int BN = 256;  // keys per block
-  constexpr int BM = 64;   // q tile
+  // q tile 64 at 145 KB LDS = 1 block/CU. A BM=32 variant (79 KB, 2
+  // blocks/CU) measured SLOWER (14.7 vs 12.0 ms) — the kernel is
+  // MFMA-pipe-bound at 2 waves/SIMD already, and halving the tile
+  // doubles per-tile staging/barrier overhead.
+  constexpr int BM = 64;
   // bottom-right causal alignment for S_q != S_kv (see flash_attn_fwd.hip)
   const int coff = SKV - SQ;
   constexpr int KP = D + 8;
@@ -143,7 +147,7 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
         __bf16* pw = &pT_lds[wid * 32 * VP];
         __bf16* dw = &dsT_lds[wid * 32 * VP];
 #pragma unroll
-        for (int nq = 0; nq < 4; ++nq) {
+        for (int nq = 0; nq < BM / 16; ++nq) {
           f32x4_t st[2], dpt[2];
           st[0] = st[1] = f32x4_t{0.f, 0.f, 0.f, 0.f};
           dpt[0] = dpt[1] = f32x4_t{0.f, 0.f, 0.f, 0.f};
@@ -247,7 +251,9 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
 }
 
 // ============================ dQ ============================
-template <int D, bool CAUSAL>
+// NSB: 16-row sub-blocks per wave (see flash_fwd_kernel) — NSB=1 keeps
+// the chip full at TP=8's HQ_local=4
+template <int D, bool CAUSAL, int NSB>
 __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
     const bf16* __restrict__ dO, const bf16* __restrict__ Q,
     const bf16* __restrict__ K, const bf16* __restrict__ V,
@@ -256,14 +262,14 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
     int SQ, int SKV, int Bb, int HQ, int HKV, float scale, int window,
     long sQs, long sQb, long sQh, long sKs, long sKb, long sKh, long sVs,
     long sVb, long sVh, long sDs, long sDb, long sDh) {
-  constexpr int BM = 256, BN = 64;
+  constexpr int BM = NSB * 128, BN = 64;  // BN=32 (2 blk/CU) was slower
   const int coff = SKV - SQ;  // bottom-right causal alignment
   constexpr int KP = D + 8;
   constexpr int VP = BN + 8;
   __shared__ __bf16 k_lds[BN * KP];
   __shared__ __bf16 v_lds[BN * KP];
   __shared__ __bf16 kt_lds[D * VP];
-  __shared__ __bf16 ds_lds[8 * 32 * VP];
+  __shared__ __bf16 ds_lds[8 * NSB * 16 * VP];
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -281,13 +287,13 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
   const float* Dp = DELTA + ((long)b * HQ + hq) * SQ;
 
   const int q0 = qblock * BM;
-  const int qrow_w = q0 + wid * 32;
+  const int qrow_w = q0 + wid * NSB * 16;
 
   constexpr int DK = D / 32;
-  bf16x8_t qfrag[2][DK], dofrag[2][DK];
-  float lse[2][4], delta[2][4];
+  bf16x8_t qfrag[NSB][DK], dofrag[NSB][DK];
+  float lse[NSB][4], delta[NSB][4];
 #pragma unroll
-  for (int sb = 0; sb < 2; ++sb) {
+  for (int sb = 0; sb < NSB; ++sb) {
     const int r = qrow_w + sb * 16 + (lane & 15);
     const long row = (r < SQ) ? r : (SQ - 1);
 #pragma unroll
@@ -310,9 +316,9 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
   }
 
   constexpr int DN = D / 16;
-  f32x4_t dqacc[2][DN];
+  f32x4_t dqacc[NSB][DN];
 #pragma unroll
-  for (int sb = 0; sb < 2; ++sb)
+  for (int sb = 0; sb < NSB; ++sb)
 #pragma unroll
     for (int nj = 0; nj < DN; ++nj) dqacc[sb][nj] = f32x4_t{0.f, 0.f, 0.f, 0.f};
 
@@ -320,7 +326,7 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
   const int nkb = (kend + BN - 1) / BN;
   const int jb0 =
       (CAUSAL && window > 0) ? max(0, (q0 + coff - window + 1) / BN) : 0;
-  const int wrow_max = qrow_w + 31;
+  const int wrow_max = qrow_w + NSB * 16 - 1;
 
   for (int jb = jb0; jb < nkb; ++jb) {
     const int kbase = jb * BN;
@@ -359,22 +365,27 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
     __syncthreads();
 
     if (!CAUSAL || kbase <= wrow_max + coff) {
-      __bf16* dsw = &ds_lds[wid * 32 * VP];
+      __bf16* dsw = &ds_lds[wid * NSB * 16 * VP];
 #pragma unroll
-      for (int nk = 0; nk < 4; ++nk) {
-        f32x4_t st[2], dpt[2];
-        st[0] = st[1] = f32x4_t{0.f, 0.f, 0.f, 0.f};
-        dpt[0] = dpt[1] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+      for (int nk = 0; nk < BN / 16; ++nk) {
+        f32x4_t st[NSB], dpt[NSB];
+#pragma unroll
+        for (int sb = 0; sb < NSB; ++sb) {
+          st[sb] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+          dpt[sb] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+        }
 #pragma unroll
         for (int kk = 0; kk < DK; ++kk) {
           bf16x8_t kb =
               load_frag_b_rowmajorT(&k_lds[nk * 16 * KP], KP, kk * 32, lane);
-          st[0] = MFMA_16x16x32(qfrag[0][kk], kb, st[0]);
-          st[1] = MFMA_16x16x32(qfrag[1][kk], kb, st[1]);
+#pragma unroll
+          for (int sb = 0; sb < NSB; ++sb)
+            st[sb] = MFMA_16x16x32(qfrag[sb][kk], kb, st[sb]);
           bf16x8_t vb =
               load_frag_b_rowmajorT(&v_lds[nk * 16 * KP], KP, kk * 32, lane);
-          dpt[0] = MFMA_16x16x32(dofrag[0][kk], vb, dpt[0]);
-          dpt[1] = MFMA_16x16x32(dofrag[1][kk], vb, dpt[1]);
+#pragma unroll
+          for (int sb = 0; sb < NSB; ++sb)
+            dpt[sb] = MFMA_16x16x32(dofrag[sb][kk], vb, dpt[sb]);
         }
         const int kcol = kbase + nk * 16 + (lane & 15);
         // interior fast path: whole 32-row x 16-key patch alive (see dkv)
@@ -384,7 +395,7 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
             (window <= 0 || (kbase + nk * 16 > qrow_w + 31 + coff - window));
         if (full_patch) {
 #pragma unroll
-          for (int sb = 0; sb < 2; ++sb) {
+          for (int sb = 0; sb < NSB; ++sb) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
               const float p = __expf(st[sb][r] - lse[sb][r]);
@@ -395,7 +406,7 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
           }
         } else {
 #pragma unroll
-          for (int sb = 0; sb < 2; ++sb) {
+          for (int sb = 0; sb < NSB; ++sb) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
               const int qrow = qrow_w + sb * 16 + (lane >> 4) * 4 + r;
@@ -416,10 +427,11 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
         for (int kk = 0; kk < BN / 32; ++kk) {
           bf16x8_t kb2 =
               load_frag_b_trT_swz(kt_lds, VP, nj * 16, kk * 32, lane);
-          bf16x8_t da0 = load_frag_a(dsw, VP, kk * 32, lane);
-          bf16x8_t da1 = load_frag_a(dsw + 16 * VP, VP, kk * 32, lane);
-          dqacc[0][nj] = MFMA_16x16x32(da0, kb2, dqacc[0][nj]);
-          dqacc[1][nj] = MFMA_16x16x32(da1, kb2, dqacc[1][nj]);
+#pragma unroll
+          for (int sb = 0; sb < NSB; ++sb) {
+            bf16x8_t da = load_frag_a(dsw + sb * 16 * VP, VP, kk * 32, lane);
+            dqacc[sb][nj] = MFMA_16x16x32(da, kb2, dqacc[sb][nj]);
+          }
         }
       }
     }
@@ -429,7 +441,7 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
   const long sOs = (long)Bb * HQ * D;
   bf16* dQp = dQ + ((long)b * HQ + hq) * D;
 #pragma unroll
-  for (int sb = 0; sb < 2; ++sb)
+  for (int sb = 0; sb < NSB; ++sb)
 #pragma unroll
     for (int nj = 0; nj < DN; ++nj)
 #pragma unroll
@@ -497,7 +509,9 @@ void launch_flash_bwd(const void* dout, const void* q, const void* k,
                       hipStream_t stream) {
   dim3 blk(512);
   dim3 gkv((SKV + 255) / 256, B * HQ);  // one block per (kblock, b, hkv, g)
-  dim3 gq((SQ + 255) / 256, B * HQ);
+  const bool small = ((long)((SQ + 255) / 256) * B * HQ) < 512;
+  const int bm = small ? 128 : 256;
+  dim3 gq((SQ + bm - 1) / bm, B * HQ);
 #define CASE(DD, CC)                                                          \
   do {                                                                        \
     flash_bwd_dkv_kernel<DD, CC><<<gkv, blk, 0, stream>>>(                    \
@@ -506,11 +520,18 @@ void launch_flash_bwd(const void* dout, const void* q, const void* k,
         SKV, B, HQ, HKV, scale, window, qstr[0], qstr[1], qstr[2], kstr[0],   \
         kstr[1],                                                              \
         kstr[2], vstr[0], vstr[1], vstr[2], dostr[0], dostr[1], dostr[2]);    \
-    flash_bwd_dq_kernel<DD, CC><<<gq, blk, 0, stream>>>(                      \
-        (const bf16*)dout, (const bf16*)q, (const bf16*)k, (const bf16*)v,    \
-        (const float*)lse, (const float*)delta, (bf16*)dq, SQ, SKV, B, HQ,    \
-        HKV, scale, window, qstr[0], qstr[1], qstr[2], kstr[0], kstr[1],      \
-        kstr[2], vstr[0], vstr[1], vstr[2], dostr[0], dostr[1], dostr[2]);    \
+    if (small)                                                                \
+      flash_bwd_dq_kernel<DD, CC, 1><<<gq, blk, 0, stream>>>(                 \
+          (const bf16*)dout, (const bf16*)q, (const bf16*)k, (const bf16*)v,  \
+          (const float*)lse, (const float*)delta, (bf16*)dq, SQ, SKV, B, HQ,  \
+          HKV, scale, window, qstr[0], qstr[1], qstr[2], kstr[0], kstr[1],    \
+          kstr[2], vstr[0], vstr[1], vstr[2], dostr[0], dostr[1], dostr[2]);  \
+    else                                                                      \
+      flash_bwd_dq_kernel<DD, CC, 2><<<gq, blk, 0, stream>>>(                 \
+          (const bf16*)dout, (const bf16*)q, (const bf16*)k, (const bf16*)v,  \
+          (const float*)lse, (const float*)delta, (bf16*)dq, SQ, SKV, B, HQ,  \
+          HKV, scale, window, qstr[0], qstr[1], qstr[2], kstr[0], kstr[1],    \
+          kstr[2], vstr[0], vstr[1], vstr[2], dostr[0], dostr[1], dostr[2]);  \
   } while (0)
   if (D == 128) {
     if (causal) CASE(128, true); else CASE(128, false);

@@ -13,7 +13,10 @@
 //  - LDS ≈ 73 KB → 2 blocks/CU → 4 waves/SIMD.
 #include "attn_common.h"
 
-template <int D, bool CAUSAL>
+// NSB = 16-row sub-blocks per wave: 2 -> BM 256 (the TP<=4 shape), 1 ->
+// BM 128 for small B*HQ launches (TP=8 has HQ_local=4: BM=256 gives 256
+// blocks = 1/CU on the bench shape — half the chip idles).
+template <int D, bool CAUSAL, int NSB>
 __global__ __launch_bounds__(512) void flash_fwd_kernel(
     const bf16* __restrict__ Q, const bf16* __restrict__ K,
     const bf16* __restrict__ V, bf16* __restrict__ O,
@@ -23,12 +26,12 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
     long sQs, long sQb, long sQh,   // Q element strides (seq, batch, head)
     long sKs, long sKb, long sKh,   // K strides
     long sVs, long sVb, long sVh) { // V strides
-  constexpr int BM = 256, BN = 64;
+  constexpr int BM = NSB * 128, BN = 64;
   constexpr int KP = D + 8;
   constexpr int VP = BN + 8;
   __shared__ __bf16 k_lds[BN * KP];
   __shared__ __bf16 vt_lds[D * VP];
-  __shared__ __bf16 p_lds[8 * 32 * VP];
+  __shared__ __bf16 p_lds[8 * NSB * 16 * VP];
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -46,12 +49,12 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
   const bf16* Vp = V + b * sVb + hkv * sVh;
 
   const int q0 = qblock * BM;
-  const int qrow_w = q0 + wid * 32;  // first row of this wave (2 sub-blocks)
+  const int qrow_w = q0 + wid * NSB * 16;  // first row of this wave
 
   constexpr int DK = D / 32;
-  bf16x8_t qfrag[2][DK];
+  bf16x8_t qfrag[NSB][DK];
 #pragma unroll
-  for (int sb = 0; sb < 2; ++sb) {
+  for (int sb = 0; sb < NSB; ++sb) {
     const int r = qrow_w + sb * 16 + (lane & 15);
     const long row = (r < SQ) ? r : (SQ - 1);
 #pragma unroll
@@ -66,8 +69,8 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
     }
   }
 
-  float m_i[2][4], l_i[2][4];
-  float alpha_s[2][4];
+  float m_i[NSB][4], l_i[NSB][4];
+  float alpha_s[NSB][4];
   // all-ones B fragment: one MFMA per 32-key chunk computes the P row-sums
   // into every lane's accumulator (replaces 16 adds + 16 shuffles per
   // sub-block of VALU reduction)
@@ -75,9 +78,9 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
 #pragma unroll
   for (int j = 0; j < 8; ++j) ones_frag[j] = (__bf16)1.0f;
   constexpr int DN = D / 16;
-  f32x4_t oacc[2][DN];
+  f32x4_t oacc[NSB][DN];
 #pragma unroll
-  for (int sb = 0; sb < 2; ++sb) {
+  for (int sb = 0; sb < NSB; ++sb) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       m_i[sb][r] = -1e30f;
@@ -91,7 +94,7 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
   const int nkb = (kend + BN - 1) / BN;
   const int jb0 =
       (CAUSAL && window > 0) ? max(0, (q0 + coff - window + 1) / BN) : 0;
-  const int wrow_max = qrow_w + 31;
+  const int wrow_max = qrow_w + NSB * 16 - 1;
 
   // T5 static form: the younger dispatch half gets priority so it is not
   // starved of VALU issue at segment starts (guide §5.5 T5).
@@ -162,23 +165,25 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
 
     if (!CAUSAL || kbase <= wrow_max + coff) {
       // ---- S = Q K^T for both sub-blocks (B-frags loaded once) ----
-      f32x4_t sacc[2][4];
+      f32x4_t sacc[NSB][4];
 #pragma unroll
       for (int nk = 0; nk < 4; ++nk) {
-        sacc[0][nk] = f32x4_t{0.f, 0.f, 0.f, 0.f};
-        sacc[1][nk] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int sb = 0; sb < NSB; ++sb)
+          sacc[sb][nk] = f32x4_t{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int kk = 0; kk < DK; ++kk) {
           bf16x8_t bfrag =
               load_frag_b_rowmajorT(&k_lds[nk * 16 * KP], KP, kk * 32, lane);
-          sacc[0][nk] = MFMA_16x16x32(qfrag[0][kk], bfrag, sacc[0][nk]);
-          sacc[1][nk] = MFMA_16x16x32(qfrag[1][kk], bfrag, sacc[1][nk]);
+#pragma unroll
+          for (int sb = 0; sb < NSB; ++sb)
+            sacc[sb][nk] = MFMA_16x16x32(qfrag[sb][kk], bfrag, sacc[sb][nk]);
         }
       }
       // ---- mask + online softmax + P→LDS, per sub-block ----
-      __bf16* pw = &p_lds[wid * 32 * VP];
+      __bf16* pw = &p_lds[wid * NSB * 16 * VP];
 #pragma unroll
-      for (int sb = 0; sb < 2; ++sb) {
+      for (int sb = 0; sb < NSB; ++sb) {
         // interior tiles (every key visible to every row) skip the mask
         const bool full_tile =
             (kbase + BN <= SKV) &&
@@ -250,27 +255,26 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
           }
         }
       }
-      // ---- O += P V (V^T B-frags loaded once per sub-block pair) ----
-      f32x4_t racc[2];
-      racc[0] = racc[1] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+      // ---- O += P V (V^T B-frags loaded once per sub-block set) ----
+      f32x4_t racc[NSB];
+#pragma unroll
+      for (int sb = 0; sb < NSB; ++sb) racc[sb] = f32x4_t{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int nj = 0; nj < DN; ++nj) {
 #pragma unroll
         for (int kk = 0; kk < BN / 32; ++kk) {
           bf16x8_t vb =
               load_frag_b_trT_swz(vt_lds, VP, nj * 16, kk * 32, lane);
-          bf16x8_t pa0 = load_frag_a(pw, VP, kk * 32, lane);
-          bf16x8_t pa1 = load_frag_a(pw + 16 * VP, VP, kk * 32, lane);
-          oacc[0][nj] = MFMA_16x16x32(pa0, vb, oacc[0][nj]);
-          oacc[1][nj] = MFMA_16x16x32(pa1, vb, oacc[1][nj]);
-          if (nj == 0) {
-            racc[0] = MFMA_16x16x32(pa0, ones_frag, racc[0]);
-            racc[1] = MFMA_16x16x32(pa1, ones_frag, racc[1]);
+#pragma unroll
+          for (int sb = 0; sb < NSB; ++sb) {
+            bf16x8_t pa = load_frag_a(pw + sb * 16 * VP, VP, kk * 32, lane);
+            oacc[sb][nj] = MFMA_16x16x32(pa, vb, oacc[sb][nj]);
+            if (nj == 0) racc[sb] = MFMA_16x16x32(pa, ones_frag, racc[sb]);
           }
         }
       }
 #pragma unroll
-      for (int sb = 0; sb < 2; ++sb)
+      for (int sb = 0; sb < NSB; ++sb)
 #pragma unroll
         for (int r = 0; r < 4; ++r)
           l_i[sb][r] = l_i[sb][r] * alpha_s[sb][r] + racc[sb][r];
@@ -287,7 +291,7 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
   const long sOs = (long)Bb * HQ * D;
   float* Lp = LSE + ((long)b * HQ + hq) * SQ;
 #pragma unroll
-  for (int sb = 0; sb < 2; ++sb) {
+  for (int sb = 0; sb < NSB; ++sb) {
     float inv_l[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r)
@@ -319,13 +323,24 @@ void launch_flash_fwd(const void* q, const void* k, const void* v, void* o,
                       int D, bool causal, float scale, int window,
                       const long* qstr, const long* kstr, const long* vstr,
                       hipStream_t stream) {
-  dim3 grid((SQ + 255) / 256, B * HQ);
+  // BM=128 when the BM=256 grid would leave CUs idle (TP=8: HQ_local=4)
+  const bool small = ((long)((SQ + 255) / 256) * B * HQ) < 512;
+  const int bm = small ? 128 : 256;
+  dim3 grid((SQ + bm - 1) / bm, B * HQ);
   dim3 blk(512);
 #define CASE(DD, CC)                                                          \
-  flash_fwd_kernel<DD, CC><<<grid, blk, 0, stream>>>(                         \
-      (const bf16*)q, (const bf16*)k, (const bf16*)v, (bf16*)o, (float*)lse, \
-      SQ, SKV, B, HQ, HKV, scale, window, qstr[0], qstr[1], qstr[2],          \
-      kstr[0], kstr[1], kstr[2], vstr[0], vstr[1], vstr[2])
+  do {                                                                        \
+    if (small)                                                                \
+      flash_fwd_kernel<DD, CC, 1><<<grid, blk, 0, stream>>>(                  \
+          (const bf16*)q, (const bf16*)k, (const bf16*)v, (bf16*)o,           \
+          (float*)lse, SQ, SKV, B, HQ, HKV, scale, window, qstr[0], qstr[1],  \
+          qstr[2], kstr[0], kstr[1], kstr[2], vstr[0], vstr[1], vstr[2]);     \
+    else                                                                      \
+      flash_fwd_kernel<DD, CC, 2><<<grid, blk, 0, stream>>>(                  \
+          (const bf16*)q, (const bf16*)k, (const bf16*)v, (bf16*)o,           \
+          (float*)lse, SQ, SKV, B, HQ, HKV, scale, window, qstr[0], qstr[1],  \
+          qstr[2], kstr[0], kstr[1], kstr[2], vstr[0], vstr[1], vstr[2]);     \
+  } while (0)
   if (D == 128) {
     if (causal) CASE(128, true); else CASE(128, false);
   } else if (D == 64) {
