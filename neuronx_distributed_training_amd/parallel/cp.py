@@ -28,13 +28,9 @@ def cp_split(t: torch.Tensor, dim: int = 1) -> torch.Tensor:
     return torch.cat([chunks[r], chunks[2 * cp - 1 - r]], dim=dim).contiguous()
 
 
-def cp_merge(t: torch.Tensor, dim: int = 1) -> torch.Tensor:
-    """Inverse of cp_split given the all-gathered per-rank chunks list —
-    used by tests; takes a list of local tensors in rank order."""
-    raise NotImplementedError("use cp_merge_list")
-
-
 def cp_merge_list(parts, dim: int = 1) -> torch.Tensor:
+    """Inverse of cp_split: reassemble the full sequence from the per-rank
+    local tensors (rank order)."""
     cp = len(parts)
     if cp == 1:
         return parts[0]
