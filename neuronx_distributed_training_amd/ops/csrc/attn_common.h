@@ -55,3 +55,32 @@ DEVINL bf16x8_t load_frag_b_trT_swz(const __bf16* img, int stride, int r0,
   *(int4*)&f = *(const int4*)((const char*)img + byte);
   return f;
 }
+
+// ---- 32x32x16 bf16 MFMA fragments (maps probe-verified on gfx950 by
+// tests/test_gpu_kernels.py::test_mfma32_layout_probe):
+//   A (32x16): lane row = lane&31, k = (lane>>5)*8 + j
+//   B (16x32): lane col = lane&31, k = (lane>>5)*8 + j
+//   C (32x32): col = lane&31, row = (r&3) + 8*(r>>2) + 4*(lane>>5)
+typedef float f32x16_t __attribute__((ext_vector_type(16)));
+#define MFMA_32x32x16(A, B, C) \
+  __builtin_amdgcn_mfma_f32_32x32x16_bf16((A), (B), (C), 0, 0, 0)
+
+// A-fragment from a row-major [32 rows][stride] tile, k-window k0
+DEVINL bf16x8_t load_frag_a32(const __bf16* base, int stride, int k0,
+                              int lane) {
+  const __bf16* p = base + (lane & 31) * stride + k0 + (lane >> 5) * 8;
+  bf16x8_t f;
+  *(int4*)&f = *(const int4*)p;
+  return f;
+}
+
+// B-fragment from a tr_swz image [cols][stride] (col-major source):
+// col = r0 + lane&31 picks the image row, 8 contiguous k's.
+DEVINL bf16x8_t load_frag_b32_trT_swz(const __bf16* img, int stride, int r0,
+                                      int k0, int lane) {
+  const int r = r0 + (lane & 31);
+  const uint byte = tr_swz((uint)(r * stride + k0 + (lane >> 5) * 8) * 2, r);
+  bf16x8_t f;
+  *(int4*)&f = *(const int4*)((const char*)img + byte);
+  return f;
+}

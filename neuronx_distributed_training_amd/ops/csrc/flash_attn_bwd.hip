@@ -75,15 +75,18 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
     }
   }
 
-  constexpr int DN = D / 16;
-  f32x4_t dvacc[2][DN], dkacc[2][DN];
+  // dV/dK accumulate on 32x32x16 MFMAs (the wave's 32 keys = one C-row
+  // block): 25% more FLOP per issue-cycle than 16x16x32 on these pure
+  // GEMM chains. kq-outer loop order: per kq the A-frags load once and
+  // the 8 MFMAs hit 8 DISTINCT accumulators (no dependent-accumulator
+  // back-to-back issue).
+  constexpr int DN2 = D / 32;
+  f32x16_t dvacc[DN2], dkacc[DN2];
 #pragma unroll
-  for (int sb = 0; sb < 2; ++sb)
-#pragma unroll
-    for (int nj = 0; nj < DN; ++nj) {
-      dvacc[sb][nj] = f32x4_t{0.f, 0.f, 0.f, 0.f};
-      dkacc[sb][nj] = f32x4_t{0.f, 0.f, 0.f, 0.f};
-    }
+  for (int nj = 0; nj < DN2; ++nj) {
+    dvacc[nj] = f32x16_t{};
+    dkacc[nj] = f32x16_t{};
+  }
 
   const int wkey_min = krow_w;  // first key of this wave
 
@@ -205,23 +208,20 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
             }
           }
         }
-        // dV += P^T dO ; dK += dS^T Q (B-frags shared by both sub-blocks)
+        // dV += P^T dO ; dK += dS^T Q — 32x32x16, contraction in 16-q
+        // chunks; A = the wave's own P^T/dS^T strip, loaded once per kq
 #pragma unroll
-        for (int nj = 0; nj < DN; ++nj) {
+        for (int kq = 0; kq < BM / 16; ++kq) {
+          bf16x8_t pa = load_frag_a32(pw, VP, kq * 16, lane);
+          bf16x8_t da = load_frag_a32(dw, VP, kq * 16, lane);
 #pragma unroll
-          for (int kk = 0; kk < BM / 32; ++kk) {
+          for (int nj = 0; nj < DN2; ++nj) {
             bf16x8_t dob =
-                load_frag_b_trT_swz(dot_lds, VP, nj * 16, kk * 32, lane);
-            bf16x8_t pa0 = load_frag_a(pw, VP, kk * 32, lane);
-            bf16x8_t pa1 = load_frag_a(pw + 16 * VP, VP, kk * 32, lane);
-            dvacc[0][nj] = MFMA_16x16x32(pa0, dob, dvacc[0][nj]);
-            dvacc[1][nj] = MFMA_16x16x32(pa1, dob, dvacc[1][nj]);
+                load_frag_b32_trT_swz(dot_lds, VP, nj * 32, kq * 16, lane);
+            dvacc[nj] = MFMA_32x32x16(pa, dob, dvacc[nj]);
             bf16x8_t qb2 =
-                load_frag_b_trT_swz(qt_lds, VP, nj * 16, kk * 32, lane);
-            bf16x8_t da0 = load_frag_a(dw, VP, kk * 32, lane);
-            bf16x8_t da1 = load_frag_a(dw + 16 * VP, VP, kk * 32, lane);
-            dkacc[0][nj] = MFMA_16x16x32(da0, qb2, dkacc[0][nj]);
-            dkacc[1][nj] = MFMA_16x16x32(da1, qb2, dkacc[1][nj]);
+                load_frag_b32_trT_swz(qt_lds, VP, nj * 32, kq * 16, lane);
+            dkacc[nj] = MFMA_32x32x16(da, qb2, dkacc[nj]);
           }
         }
       }
@@ -229,22 +229,20 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
     }
   }
 
-  // ---- store fp32 partials: [group][s][b][hkv][d] ----
+  // ---- store fp32 partials: [group][s][b][hkv][d] (32x32 C layout) ----
   const long sOs = (long)Bb * HKV * D;
   const long slab = (long)g * SKV * sOs + ((long)b * HKV + hkv) * D;
   float* dKp = dK + slab;
   float* dVp = dV + slab;
 #pragma unroll
-  for (int sb = 0; sb < 2; ++sb) {
+  for (int nj = 0; nj < DN2; ++nj) {
 #pragma unroll
-    for (int nj = 0; nj < DN; ++nj) {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int krow = krow_w + sb * 16 + (lane >> 4) * 4 + r;
-        if (krow < SKV) {
-          dKp[(long)krow * sOs + nj * 16 + (lane & 15)] = dkacc[sb][nj][r];
-          dVp[(long)krow * sOs + nj * 16 + (lane & 15)] = dvacc[sb][nj][r];
-        }
+    for (int r = 0; r < 16; ++r) {
+      const int krow = krow_w + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+      if (krow < SKV) {
+        const int dcol = nj * 32 + (lane & 31);
+        dKp[(long)krow * sOs + dcol] = dkacc[nj][r];
+        dVp[(long)krow * sOs + dcol] = dvacc[nj][r];
       }
     }
   }
@@ -315,12 +313,23 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
     }
   }
 
+  // NSB==2: dQ accumulates on 32x32x16 MFMAs (wave's 32 q-rows = one
+  // C block, distinct accumulators back-to-back); NSB==1 waves hold
+  // only 16 rows and stay on 16x16x32.
   constexpr int DN = D / 16;
-  f32x4_t dqacc[NSB][DN];
+  constexpr int DN2 = D / 32;
+  f32x4_t dqacc[NSB == 1 ? NSB : 1][NSB == 1 ? DN : 1];
+  f32x16_t dqacc32[NSB == 2 ? DN2 : 1];
+  if constexpr (NSB == 1) {
 #pragma unroll
-  for (int sb = 0; sb < NSB; ++sb)
+    for (int sb = 0; sb < NSB; ++sb)
 #pragma unroll
-    for (int nj = 0; nj < DN; ++nj) dqacc[sb][nj] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+      for (int nj = 0; nj < DN; ++nj)
+        dqacc[sb][nj] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+  } else {
+#pragma unroll
+    for (int nj = 0; nj < DN2; ++nj) dqacc32[nj] = f32x16_t{};
+  }
 
   const int kend = CAUSAL ? min(SKV, q0 + BM + coff) : SKV;
   const int nkb = (kend + BN - 1) / BN;
@@ -421,16 +430,26 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
           }
         }
       }
+      if constexpr (NSB == 1) {
 #pragma unroll
-      for (int nj = 0; nj < DN; ++nj) {
+        for (int nj = 0; nj < DN; ++nj) {
 #pragma unroll
-        for (int kk = 0; kk < BN / 32; ++kk) {
-          bf16x8_t kb2 =
-              load_frag_b_trT_swz(kt_lds, VP, nj * 16, kk * 32, lane);
+          for (int kk = 0; kk < BN / 32; ++kk) {
+            bf16x8_t kb2 =
+                load_frag_b_trT_swz(kt_lds, VP, nj * 16, kk * 32, lane);
+            bf16x8_t da = load_frag_a(dsw, VP, kk * 32, lane);
+            dqacc[0][nj] = MFMA_16x16x32(da, kb2, dqacc[0][nj]);
+          }
+        }
+      } else {
 #pragma unroll
-          for (int sb = 0; sb < NSB; ++sb) {
-            bf16x8_t da = load_frag_a(dsw + sb * 16 * VP, VP, kk * 32, lane);
-            dqacc[sb][nj] = MFMA_16x16x32(da, kb2, dqacc[sb][nj]);
+        for (int kq = 0; kq < BN / 16; ++kq) {
+          bf16x8_t da = load_frag_a32(dsw, VP, kq * 16, lane);
+#pragma unroll
+          for (int nj = 0; nj < DN2; ++nj) {
+            bf16x8_t kb2 =
+                load_frag_b32_trT_swz(kt_lds, VP, nj * 32, kq * 16, lane);
+            dqacc32[nj] = MFMA_32x32x16(da, kb2, dqacc32[nj]);
           }
         }
       }
@@ -440,17 +459,27 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
 
   const long sOs = (long)Bb * HQ * D;
   bf16* dQp = dQ + ((long)b * HQ + hq) * D;
-#pragma unroll
-  for (int sb = 0; sb < NSB; ++sb)
+  if constexpr (NSB == 1) {
 #pragma unroll
     for (int nj = 0; nj < DN; ++nj)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int qrow = qrow_w + sb * 16 + (lane >> 4) * 4 + r;
+        const int qrow = qrow_w + (lane >> 4) * 4 + r;
         if (qrow < SQ)
           dQp[(long)qrow * sOs + nj * 16 + (lane & 15)] =
-              f2bf(dqacc[sb][nj][r]);
+              f2bf(dqacc[0][nj][r]);
       }
+  } else {
+#pragma unroll
+    for (int nj = 0; nj < DN2; ++nj)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qrow = qrow_w + (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+        if (qrow < SQ)
+          dQp[(long)qrow * sOs + nj * 32 + (lane & 31)] =
+              f2bf(dqacc32[nj][r]);
+      }
+  }
 }
 
 // ===================== delta = rowsum(dO . O) =====================
