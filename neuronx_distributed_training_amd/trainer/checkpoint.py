@@ -183,7 +183,15 @@ class CheckpointIO:
                 uc["scheduler"] = module.scheduler.state_dict()
             work.append((_join(root, fs, "user_content.pt"), uc))
 
-        use_fork = self.writer_process and fs is None and hasattr(os, "fork")
+        # forking with a live HIP runtime deadlocks the child (allocator
+        # locks held by threads that don't exist post-fork — observed
+        # hanging on MI355X, r2); the writer process is therefore only
+        # used before CUDA init (CPU jobs / tests). GPU saves rely on the
+        # pinned side-stream staging + thread writer, which never blocks
+        # the training thread.
+        use_fork = (self.writer_process and fs is None
+                    and hasattr(os, "fork")
+                    and not torch.cuda.is_initialized())
 
         def _write():
             for ev in events:
