@@ -44,9 +44,9 @@ __global__ __launch_bounds__(512) void flash_bwd_dkv_kernel(
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int kblock = blockIdx.x;
   const int group = HQ / HKV;
-  const int bhg = blockIdx.y;  // (b * HKV + hkv) * group + g
+  int kblock, bhg;  // (b * HKV + hkv) * group + g
+  xcd_remap(kblock, bhg);
   const int g = bhg % group;
   const int hkv = (bhg / group) % HKV;
   const int b = bhg / (group * HKV);
@@ -271,8 +271,8 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int qblock = blockIdx.x;
-  const int bh = blockIdx.y;  // b * HQ + hq
+  int qblock, bh;  // b * HQ + hq
+  xcd_remap(qblock, bh);
   const int hq = bh % HQ;
   const int b = bh / HQ;
   const int hkv = hq / (HQ / HKV);

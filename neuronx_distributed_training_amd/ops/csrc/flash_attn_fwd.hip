@@ -35,8 +35,8 @@ __global__ __launch_bounds__(512) void flash_fwd_kernel(
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int qblock = blockIdx.x;
-  const int bh = blockIdx.y;  // b * HQ + hq
+  int qblock, bh;  // b * HQ + hq
+  xcd_remap(qblock, bh);
   const int hq = bh % HQ;
   const int b = bh / HQ;
   const int hkv = hq / (HQ / HKV);
