@@ -226,6 +226,15 @@ class ParallelAttention(nn.Module):
         self.attn_dropout_p = cfg.attention_dropout
 
     def core_attention(self, q, k, v):
+        if ps.get_context_model_parallel_world_size() > 1:
+            from ..ops.ring_attn import ring_flash_attn
+
+            assert not getattr(self.cfg, "sliding_window", None), (
+                "sliding_window attention is not supported with "
+                "context_parallel_size > 1 (ring attention computes full "
+                "causal attention); disable CP or the window"
+            )
+            return ring_flash_attn(q, k, v, scale=self.scale)
         return flash_attn_func(q, k, v, causal=True, scale=self.scale,
                                window=self.cfg.sliding_window)
 

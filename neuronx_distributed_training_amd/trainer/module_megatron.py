@@ -87,6 +87,34 @@ class MegatronGPTModule(BaseModelModule):
             return GPTStage(cfg, pipeline_cuts=mcfg.get("pipeline_cuts"))
         return GPTModel(cfg)
 
+    def get_batch_on_this_context_parallel_rank(self, batch):
+        """Megatron batches carry PRE-SHIFTED labels (gpt_dataset emits
+        labels = tokens[1:]), so the base class's next-token roll would
+        double-shift them under CP. Only the zigzag split and the exact
+        CP loss denominator apply here."""
+        cp = ps.get_context_model_parallel_world_size()
+        if cp == 1:
+            return batch
+        from ..parallel.cp import cp_split
+
+        batch = dict(batch)
+        if "labels" in batch and torch.is_tensor(batch["labels"]):
+            mask = batch.get("loss_mask")
+            if mask is None:
+                mask = torch.ones_like(
+                    batch["labels"], dtype=torch.float32
+                )
+            batch["loss_mask"] = mask
+            batch["loss_denominator"] = mask.sum() / cp
+        out = {}
+        for k, v in batch.items():
+            if torch.is_tensor(v) and v.dim() >= 2 \
+                    and v.size(1) == self.seq_length:
+                out[k] = cp_split(v, dim=1)
+            else:
+                out[k] = v
+        return out
+
     def model_fwd_calc_loss(self, batch):
         # megatron batches carry pre-shifted labels + position_ids
         return self.model(

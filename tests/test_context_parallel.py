@@ -239,3 +239,79 @@ def test_zigzag_offsets_match_split():
     """cp_offsets/cp_position_ids agree with the actual token positions
     cp_split selects (CP=2, gloo)."""
     run_distributed(_zz_offsets_chk, 2)
+
+
+def _mixtral_cp_loss(rank, world):
+    """Mixtral MoE under zigzag CP: loss matches CP=1 (router + experts
+    see the same tokens, just placed zigzag)."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module_mixtral import (
+        MixtralModule,
+    )
+
+    ps.initialize_model_parallel(context_parallel_size=world)
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 2, "seq_length": 32},
+        "distributed_strategy": {"context_parallel_size": world},
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 2, "num_attention_heads": 4, "num_kv_heads": 2,
+            "moe": {"num_experts": 4, "top_k": 2, "aux_loss_coef": 0.02},
+            "grad_clip": 1.0, "optim": {"lr": 1e-3, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(3)
+    mod = MixtralModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=10)
+    g = torch.Generator().manual_seed(5)
+    ids = torch.randint(0, 128, (2, 32), generator=g)
+    m = mod.training_step([{"input_ids": ids, "labels": ids.clone()}])
+    return m["reduced_train_loss"]
+
+
+def test_mixtral_cp2_loss_matches_cp1():
+    l1 = run_distributed(_mixtral_cp_loss, 1)[0]
+    l2 = run_distributed(_mixtral_cp_loss, 2)
+    assert abs(l2[0] - l2[1]) < 1e-6
+    assert abs(l1 - l2[0]) < 0.05, (l1, l2[0])
+
+
+def _megatron_abs_cp_loss(rank, world):
+    """Megatron GPT with LEARNED-ABSOLUTE positions under zigzag CP:
+    cp_position_ids must hand each rank its true global positions."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module_megatron import (
+        MegatronGPTModule,
+    )
+
+    ps.initialize_model_parallel(context_parallel_size=world)
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 2, "seq_length": 32},
+        "distributed_strategy": {"context_parallel_size": world},
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "ffn_hidden_size": 128,
+            "num_layers": 2, "num_attention_heads": 4,
+            "position_embedding_type": "learned_absolute",
+            "grad_clip": 1.0, "optim": {"lr": 1e-3, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(3)
+    mod = MegatronGPTModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=10)
+    g = torch.Generator().manual_seed(5)
+    ids = torch.randint(0, 128, (2, 32), generator=g)
+    m = mod.training_step([{"input_ids": ids, "labels": ids.clone()}])
+    return m["reduced_train_loss"]
+
+
+def test_megatron_learned_abs_cp2_matches_cp1():
+    l1 = run_distributed(_megatron_abs_cp_loss, 1)[0]
+    l2 = run_distributed(_megatron_abs_cp_loss, 2)
+    assert abs(l2[0] - l2[1]) < 1e-6
+    assert abs(l1 - l2[0]) < 0.05, (l1, l2[0])
