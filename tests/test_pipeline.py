@@ -430,3 +430,96 @@ def _ref8_loss(rank, world):
         ids = torch.randint(0, 128, (1, 32), generator=g)
         tot += float(m(ids, labels=ids))
     return tot / 8
+
+
+def _pp_cp_train(rank, world):
+    """PP2 x CP2 on 4 ranks: LlamaStage's zigzag pos offsets + pre-shifted
+    labels + loss_denominator under CP, through the full trainer module."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+
+    pp = 2 if world == 4 else 1
+    cp = 2 if world == 4 else 1
+    ps.initialize_model_parallel(pipeline_model_parallel_size=pp,
+                                 context_parallel_size=cp)
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 1,
+                 "seq_length": 32},
+        "distributed_strategy": {"pipeline_model_parallel_size": pp,
+                                 "context_parallel_size": cp},
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 4, "num_attention_heads": 4, "num_kv_heads": 2,
+            "grad_clip": 1.0,
+            "optim": {"lr": 1e-3, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(3)
+    mod = LlamaModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=10)
+    g = torch.Generator().manual_seed(5)
+    micros = [
+        {"input_ids": (ids := torch.randint(0, 128, (1, 32), generator=g)),
+         "labels": ids.clone()}
+        for _ in range(2)
+    ]
+    m = mod.training_step(micros)
+    return m["reduced_train_loss"]
+
+
+def test_pp2_cp2_loss_matches_single():
+    ref = run_distributed(_pp_cp_train, 1)[0]
+    res = run_distributed(_pp_cp_train, 4)
+    assert max(abs(r - res[0]) for r in res) < 1e-6
+    assert abs(ref - res[0]) < 0.05, (ref, res[0])
+
+
+def _vp_cp_train(rank, world):
+    """VP2 on PP2 x CP2 (4 ranks): LlamaChunk's zigzag pos offsets under
+    CP (VP+CP previously used pos_offset=0 silently)."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+
+    pp = 2 if world == 4 else 1
+    cp = 2 if world == 4 else 1
+    ps.initialize_model_parallel(pipeline_model_parallel_size=pp,
+                                 context_parallel_size=cp)
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 1,
+                 "seq_length": 32},
+        "distributed_strategy": {
+            "pipeline_model_parallel_size": pp,
+            "context_parallel_size": cp,
+            "virtual_pipeline_model_parallel_size": 2 if pp > 1 else 1,
+        },
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 4, "num_attention_heads": 4, "num_kv_heads": 2,
+            "grad_clip": 1.0,
+            "optim": {"lr": 1e-3, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(3)
+    mod = LlamaModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=10)
+    g = torch.Generator().manual_seed(5)
+    micros = [
+        {"input_ids": (ids := torch.randint(0, 128, (1, 32), generator=g)),
+         "labels": ids.clone()}
+        for _ in range(2)
+    ]
+    m = mod.training_step(micros)
+    return m["reduced_train_loss"]
+
+
+def test_vp2_cp2_loss_matches_single():
+    ref = run_distributed(_vp_cp_train, 1)[0]
+    res = run_distributed(_vp_cp_train, 4)
+    assert max(abs(r - res[0]) for r in res) < 1e-6
+    assert abs(ref - res[0]) < 0.05, (ref, res[0])
