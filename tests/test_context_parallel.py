@@ -315,3 +315,45 @@ def test_megatron_learned_abs_cp2_matches_cp1():
     l2 = run_distributed(_megatron_abs_cp_loss, 2)
     assert abs(l2[0] - l2[1]) < 1e-6
     assert abs(l1 - l2[0]) < 0.05, (l1, l2[0])
+
+
+def _cp_sp_ckpt_train(rank, world):
+    """CP2 x TP2 with sequence parallel AND full activation checkpointing
+    (ring P2P re-runs inside the recompute): loss matches single-rank."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module import LlamaModule
+
+    tp = 2 if world == 4 else 1
+    cp = 2 if world == 4 else 1
+    ps.initialize_model_parallel(tensor_model_parallel_size=tp,
+                                 context_parallel_size=cp)
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 2,
+                 "seq_length": 32},
+        "distributed_strategy": {"tensor_model_parallel_size": tp,
+                                 "context_parallel_size": cp,
+                                 "sequence_parallel": tp > 1},
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 2, "num_attention_heads": 4, "num_kv_heads": 2,
+            "activation_checkpoint": "full", "grad_clip": 1.0,
+            "optim": {"lr": 1e-3, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(3)
+    mod = LlamaModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=10)
+    g = torch.Generator().manual_seed(5)
+    ids = torch.randint(0, 128, (2, 32), generator=g)
+    m = mod.training_step([{"input_ids": ids, "labels": ids.clone()}])
+    return m["reduced_train_loss"]
+
+
+def test_cp2_tp2_sp_full_ckpt_matches_single():
+    l1 = run_distributed(_cp_sp_ckpt_train, 1)[0]
+    l4 = run_distributed(_cp_sp_ckpt_train, 4)
+    assert max(abs(l - l4[0]) for l in l4) < 1e-6
+    assert abs(l1 - l4[0]) < 0.05, (l1, l4[0])
