@@ -257,7 +257,10 @@ class LlamaAttention(nn.Module):
             # prefill (S_q == S_kv) and incremental decode (S_q < S_kv)
             # both run the flash kernel: causal is bottom-right aligned, so
             # query at global position past+i sees keys 0..past+i
-            o = flash_attn_func(q, k, v, causal=True, scale=self.scale)
+            o = flash_attn_func(
+                q, k, v, causal=True, scale=self.scale,
+                window=getattr(self.cfg, "sliding_window", None),
+            )
         elif self.cfg.activation_checkpoint == "selective" and self.training:
             o = _ckpt(self.core_attention, q, k, v, use_reentrant=False)
         else:

@@ -456,3 +456,26 @@ def test_moe_grouped_gemm_gpu(ext):
         err = (got.float() - want).abs().max()
         sc = want.abs().max().clamp(min=1e-3)
         assert err / sc < 0.06, f"{name} rel err {err / sc}"
+
+
+def test_flash_attn_crosslen_window_gpu(ext):
+    """S_q != S_kv WITH a sliding window (Mistral-style decode): query at
+    global position past+i sees keys (past+i-w, past+i]."""
+    from neuronx_distributed_training_amd.ops import flash_attn_func
+
+    torch.manual_seed(14)
+    b, hq, hkv, sq, skv, d, w = 1, 4, 2, 128, 512, 128, 160
+    q = torch.randn(b, hq, sq, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(b, hkv, skv, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(b, hkv, skv, d, device="cuda", dtype=torch.bfloat16)
+    o = flash_attn_func(q, k, v, causal=True, window=w)
+    kx = k.repeat_interleave(hq // hkv, 1).float()
+    vx = v.repeat_interleave(hq // hkv, 1).float()
+    s = (q.float() @ kx.transpose(-1, -2)) / d ** 0.5
+    diag = skv - sq
+    mask = torch.ones(sq, skv, dtype=torch.bool, device="cuda").triu(
+        1 + diag
+    ) | torch.ones(sq, skv, dtype=torch.bool, device="cuda").tril(diag - w)
+    ref = torch.softmax(s.masked_fill(mask, float("-inf")), -1) @ vx
+    err = (o.float() - ref).abs().max()
+    assert err < 0.02, f"max err {err}"
