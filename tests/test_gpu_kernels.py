@@ -479,3 +479,35 @@ def test_flash_attn_crosslen_window_gpu(ext):
     ref = torch.softmax(s.masked_fill(mask, float("-inf")), -1) @ vx
     err = (o.float() - ref).abs().max()
     assert err < 0.02, f"max err {err}"
+
+
+def test_flash_attn_noncausal_gpu(ext):
+    """causal=False instantiation (the zigzag ring's off-diagonal
+    half-blocks run it) fwd+bwd vs fp32 reference."""
+    from neuronx_distributed_training_amd.ops import flash_attn_func
+
+    torch.manual_seed(15)
+    b, hq, hkv, sq, skv, d = 1, 4, 2, 256, 384, 128
+    q = torch.randn(b, hq, sq, d, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn(b, hkv, skv, d, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    v = torch.randn(b, hkv, skv, d, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    o = flash_attn_func(q, k, v, causal=False)
+    g = torch.randn_like(o)
+    o.backward(g)
+    qr = q.detach().float().clone().requires_grad_(True)
+    kr = k.detach().float().clone().requires_grad_(True)
+    vr = v.detach().float().clone().requires_grad_(True)
+    ref = torch.nn.functional.scaled_dot_product_attention(
+        qr, kr.repeat_interleave(hq // hkv, 1),
+        vr.repeat_interleave(hq // hkv, 1), is_causal=False,
+    )
+    assert (o.float() - ref).abs().max() < 0.02
+    ref.backward(g.float())
+    for got, want, name in ((q.grad, qr.grad, "dq"), (k.grad, kr.grad, "dk"),
+                            (v.grad, vr.grad, "dv")):
+        err = (got.float() - want).abs().max()
+        sc = want.abs().max().clamp(min=1)
+        assert err / sc < 0.05, f"{name} rel err {err / sc}"
