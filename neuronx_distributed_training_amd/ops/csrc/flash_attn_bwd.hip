@@ -264,9 +264,13 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
   const int coff = SKV - SQ;  // bottom-right causal alignment
   constexpr int KP = D + 8;
   constexpr int VP = BN + 8;
-  __shared__ __bf16 k_lds[BN * KP];
-  __shared__ __bf16 v_lds[BN * KP];
-  __shared__ __bf16 kt_lds[D * VP];
+  // double-buffered K/V/K^T staging: tile jb+1 streams global->LDS into
+  // the other buffer set DURING tile jb's MFMAs, and each tile needs one
+  // barrier instead of two (the kernel was already 1 block/CU at 90 KB,
+  // so the extra LDS costs no occupancy)
+  __shared__ __bf16 k_lds[2][BN * KP];
+  __shared__ __bf16 v_lds[2][BN * KP];
+  __shared__ __bf16 kt_lds[2][D * VP];
   __shared__ __bf16 ds_lds[8 * NSB * 16 * VP];
 
   const int lane = threadIdx.x & 63;
@@ -337,41 +341,49 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
       (CAUSAL && window > 0) ? max(0, (q0 + coff - window + 1) / BN) : 0;
   const int wrow_max = qrow_w + NSB * 16 - 1;
 
-  for (int jb = jb0; jb < nkb; ++jb) {
+  auto stage_kv = [&](int jb, int buf) {
     const int kbase = jb * BN;
-    {  // stage K, V row-major + K^T paired
-      constexpr int KVECS = BN * D / 8;
-      for (int t = threadIdx.x; t < KVECS; t += 512) {
-        const int row = t / (D / 8);
-        const int col8 = (t % (D / 8)) * 8;
-        const int gr = kbase + row;
-        int4 kv = (gr < SKV) ? *(const int4*)(Kp + (long)gr * sKs + col8)
-                             : int4{0, 0, 0, 0};
-        int4 vv = (gr < SKV) ? *(const int4*)(Vp + (long)gr * sVs + col8)
-                             : int4{0, 0, 0, 0};
-        *(int4*)&k_lds[row * KP + col8] = kv;
-        *(int4*)&v_lds[row * KP + col8] = vv;
-      }
-      constexpr int TVECS = BN * D / 16;
-      for (int t = threadIdx.x; t < TVECS; t += 512) {
-        const int row = (t / (D / 8)) * 2;
-        const int col8 = (t % (D / 8)) * 8;
-        const int g0 = kbase + row, g1 = g0 + 1;
-        int4 k0 = (g0 < SKV) ? *(const int4*)(Kp + (long)g0 * sKs + col8)
-                             : int4{0, 0, 0, 0};
-        int4 k1 = (g1 < SKV) ? *(const int4*)(Kp + (long)g1 * sKs + col8)
-                             : int4{0, 0, 0, 0};
-        const __bf16 *e0 = (const __bf16*)&k0, *e1 = (const __bf16*)&k1;
+    constexpr int KVECS = BN * D / 8;
+    for (int t = threadIdx.x; t < KVECS; t += 512) {
+      const int row = t / (D / 8);
+      const int col8 = (t % (D / 8)) * 8;
+      const int gr = kbase + row;
+      int4 kv = (gr < SKV) ? *(const int4*)(Kp + (long)gr * sKs + col8)
+                           : int4{0, 0, 0, 0};
+      int4 vv = (gr < SKV) ? *(const int4*)(Vp + (long)gr * sVs + col8)
+                           : int4{0, 0, 0, 0};
+      *(int4*)&k_lds[buf][row * KP + col8] = kv;
+      *(int4*)&v_lds[buf][row * KP + col8] = vv;
+    }
+    constexpr int TVECS = BN * D / 16;
+    for (int t = threadIdx.x; t < TVECS; t += 512) {
+      const int row = (t / (D / 8)) * 2;
+      const int col8 = (t % (D / 8)) * 8;
+      const int g0 = kbase + row, g1 = g0 + 1;
+      int4 k0 = (g0 < SKV) ? *(const int4*)(Kp + (long)g0 * sKs + col8)
+                           : int4{0, 0, 0, 0};
+      int4 k1 = (g1 < SKV) ? *(const int4*)(Kp + (long)g1 * sKs + col8)
+                           : int4{0, 0, 0, 0};
+      const __bf16 *e0 = (const __bf16*)&k0, *e1 = (const __bf16*)&k1;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          __bf16 pr[2] = {e0[j], e1[j]};
-          const int r = col8 + j;
-          *(uint*)((char*)kt_lds + tr_swz((uint)(r * VP + row) * 2, r)) =
-              *(uint*)pr;
-        }
+      for (int j = 0; j < 8; ++j) {
+        __bf16 pr[2] = {e0[j], e1[j]};
+        const int r = col8 + j;
+        *(uint*)((char*)kt_lds[buf] + tr_swz((uint)(r * VP + row) * 2, r)) =
+            *(uint*)pr;
       }
     }
-    __syncthreads();
+  };
+
+  stage_kv(jb0, jb0 & 1);
+  __syncthreads();
+
+  for (int jb = jb0; jb < nkb; ++jb) {
+    const int kbase = jb * BN;
+    const int cur = jb & 1;
+    // stream tile jb+1 into the other buffer set (last read before the
+    // barrier that ended tile jb-1) while this tile's MFMAs run
+    if (jb + 1 < nkb) stage_kv(jb + 1, cur ^ 1);
 
     if (!CAUSAL || kbase <= wrow_max + coff) {
       __bf16* dsw = &ds_lds[wid * NSB * 16 * VP];
@@ -385,13 +397,13 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
         }
 #pragma unroll
         for (int kk = 0; kk < DK; ++kk) {
-          bf16x8_t kb =
-              load_frag_b_rowmajorT(&k_lds[nk * 16 * KP], KP, kk * 32, lane);
+          bf16x8_t kb = load_frag_b_rowmajorT(&k_lds[cur][nk * 16 * KP],
+                                              KP, kk * 32, lane);
 #pragma unroll
           for (int sb = 0; sb < NSB; ++sb)
             st[sb] = MFMA_16x16x32(qfrag[sb][kk], kb, st[sb]);
-          bf16x8_t vb =
-              load_frag_b_rowmajorT(&v_lds[nk * 16 * KP], KP, kk * 32, lane);
+          bf16x8_t vb = load_frag_b_rowmajorT(&v_lds[cur][nk * 16 * KP],
+                                              KP, kk * 32, lane);
 #pragma unroll
           for (int sb = 0; sb < NSB; ++sb)
             dpt[sb] = MFMA_16x16x32(dofrag[sb][kk], vb, dpt[sb]);
@@ -436,7 +448,7 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
 #pragma unroll
           for (int kk = 0; kk < BN / 32; ++kk) {
             bf16x8_t kb2 =
-                load_frag_b_trT_swz(kt_lds, VP, nj * 16, kk * 32, lane);
+                load_frag_b_trT_swz(kt_lds[cur], VP, nj * 16, kk * 32, lane);
             bf16x8_t da = load_frag_a(dsw, VP, kk * 32, lane);
             dqacc[0][nj] = MFMA_16x16x32(da, kb2, dqacc[0][nj]);
           }
@@ -447,14 +459,14 @@ __global__ __launch_bounds__(512) void flash_bwd_dq_kernel(
           bf16x8_t da = load_frag_a32(dsw, VP, kq * 16, lane);
 #pragma unroll
           for (int nj = 0; nj < DN2; ++nj) {
-            bf16x8_t kb2 =
-                load_frag_b32_trT_swz(kt_lds, VP, nj * 32, kq * 16, lane);
+            bf16x8_t kb2 = load_frag_b32_trT_swz(kt_lds[cur], VP, nj * 32,
+                                                 kq * 16, lane);
             dqacc32[nj] = MFMA_32x32x16(da, kb2, dqacc32[nj]);
           }
         }
       }
     }
-    __syncthreads();
+    __syncthreads();  // tile jb reads done AND tile jb+1 stage visible
   }
 
   const long sOs = (long)Bb * HQ * D;

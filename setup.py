@@ -26,6 +26,7 @@ sources = [
         "rope.hip",
         "adamw.hip",
         "flash_attn_fwd.hip",
+        "flash_attn_fwd_dbuf.hip",
         "flash_attn_fwd_v3.hip",
         "flash_attn_bwd_v3.hip",
         "flash_attn_bwd.hip",
