@@ -267,11 +267,12 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k,
   return {o_mem.permute({1, 2, 0, 3}), lse};
 }
 
-std::vector<torch::Tensor> flash_attn_fwd_dbuf(torch::Tensor q,
+std::vector<torch::Tensor> flash_attn_fwd_sbuf(torch::Tensor q,
                                                torch::Tensor k,
                                                torch::Tensor v, bool causal,
                                                double scale, long window) {
-  // dark A/B variant: double-buffered K/V staging, one barrier per tile
+  // A/B reference: the SINGLE-buffered staging kernel (two barriers per
+  // tile) that the double-buffered default replaced — bit-identical
   check_bhsd(q, "q");
   check_bhsd(k, "k");
   check_bhsd(v, "v");
@@ -431,7 +432,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("v"), py::arg("wd_mask"), py::arg("lr"), py::arg("b1"),
         py::arg("b2"), py::arg("eps"), py::arg("wd"), py::arg("step"),
         py::arg("grad_scale") = py::none(), py::arg("p_bf16") = py::none());
-  m.def("flash_attn_fwd_dbuf", &flash_attn_fwd_dbuf, pybind11::arg("q"),
+  m.def("flash_attn_fwd_sbuf", &flash_attn_fwd_sbuf, pybind11::arg("q"),
         pybind11::arg("k"), pybind11::arg("v"), pybind11::arg("causal"),
         pybind11::arg("scale"), pybind11::arg("window") = 0);
   m.def("flash_attn_fwd", &flash_attn_fwd, pybind11::arg("q"),

@@ -511,3 +511,17 @@ def test_flash_attn_noncausal_gpu(ext):
         err = (got.float() - want).abs().max()
         sc = want.abs().max().clamp(min=1)
         assert err / sc < 0.05, f"{name} rel err {err / sc}"
+
+
+def test_flash_fwd_dbuf_matches_sbuf(ext):
+    """The double-buffered default and the single-buffered A/B reference
+    stage identical data — outputs must be BIT-identical."""
+    torch.manual_seed(16)
+    b, hq, hkv, s, d = 2, 8, 2, 1024, 128
+    q = torch.randn(b, hq, s, d, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(b, hkv, s, d, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(b, hkv, s, d, device="cuda", dtype=torch.bfloat16)
+    scale = 1.0 / d ** 0.5
+    o1, l1 = ext.flash_attn_fwd(q, k, v, True, scale)
+    o2, l2 = ext.flash_attn_fwd_sbuf(q, k, v, True, scale)
+    assert torch.equal(o1, o2) and torch.equal(l1, l2)
