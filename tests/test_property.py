@@ -173,3 +173,37 @@ def test_sp_overlap_divisibility_gate(s_loc, chunks):
         assert y.shape == (s_loc, 2, 16)
     finally:
         L._SP_OVERLAP_CHUNKS = old
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    cp=st.sampled_from([2, 4, 8]),
+    chunks_per=st.integers(min_value=1, max_value=6),
+    b=st.integers(min_value=1, max_value=3),
+)
+def test_zigzag_partition_properties(cp, chunks_per, b):
+    """Zigzag CP placement: every token appears exactly once across
+    ranks, merge inverts split, and each rank's two halves are its
+    global chunks r and 2cp-1-r (hand-computed, no dist init needed)."""
+    import torch
+    from neuronx_distributed_training_amd.parallel.cp import cp_merge_list
+
+    s = 2 * cp * chunks_per
+    t = torch.arange(b * s).reshape(b, s)
+    chunks = t.chunk(2 * cp, dim=1)
+    parts = [
+        torch.cat([chunks[r], chunks[2 * cp - 1 - r]], dim=1)
+        for r in range(cp)
+    ]
+    # partition: all tokens exactly once
+    allv = torch.cat(parts, dim=1).flatten().sort().values
+    assert torch.equal(allv, t.flatten().sort().values)
+    # merge inverts
+    assert torch.equal(cp_merge_list(parts, dim=1), t)
+    # per-rank halves are contiguous global chunks
+    c = s // (2 * cp)
+    for r, p in enumerate(parts):
+        lo, hi = p[:, :c], p[:, c:]
+        assert torch.equal(lo[0], torch.arange(r * c, (r + 1) * c))
+        hi0 = (2 * cp - 1 - r) * c
+        assert torch.equal(hi[0], torch.arange(hi0, hi0 + c))
