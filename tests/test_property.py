@@ -207,3 +207,36 @@ def test_zigzag_partition_properties(cp, chunks_per, b):
         assert torch.equal(lo[0], torch.arange(r * c, (r + 1) * c))
         hi0 = (2 * cp - 1 - r) * c
         assert torch.equal(hi[0], torch.arange(hi0, hi0 + c))
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    counts=st.lists(st.integers(min_value=0, max_value=1000), min_size=1,
+                    max_size=8),
+)
+def test_moe_gemm_layout_properties(counts):
+    """Grouped-GEMM layout (ops/moe_gemm._layout): scatter rows are
+    unique, land inside their expert's padded segment in order, tiles
+    map to the segment covering them, and totals are BM-aligned."""
+    import torch
+    from neuronx_distributed_training_amd.ops.moe_gemm import _layout, BM
+
+    cts = torch.tensor(counts)
+    T = int(cts.sum())
+    pr, tile_e, pad_off, total, Tp = _layout(cts, T)
+    E = len(counts)
+    assert int(total) % BM == 0 and Tp % BM == 0 and Tp >= int(total)
+    if T:
+        assert pr.unique().numel() == T  # injective scatter
+    start = 0
+    for e, c in enumerate(counts):
+        seg = pr[start:start + c]
+        if c:
+            assert int(seg.min()) == int(pad_off[e])
+            assert int(seg.max()) < int(pad_off[e + 1])
+            assert torch.equal(seg, torch.arange(int(pad_off[e]),
+                                                 int(pad_off[e]) + c))
+        start += c
+    for t in range(int(total) // BM):
+        e = int(tile_e[t])
+        assert int(pad_off[e]) <= t * BM < int(pad_off[e + 1])
