@@ -240,3 +240,45 @@ def test_moe_gemm_layout_properties(counts):
     for t in range(int(total) // BM):
         e = int(tile_e[t])
         assert int(pad_off[e]) <= t * BM < int(pad_off[e + 1])
+
+
+@settings(max_examples=30, deadline=None)
+@given(
+    R=st.integers(min_value=2, max_value=8),
+    c=st.integers(min_value=1, max_value=4),
+)
+def test_zigzag_ring_case_table(R, c):
+    """The ring's per-hop visibility rules (ops/ring_attn.py: diagonal
+    causal / j<r low-chunk-full / j>r high-queries-full) reproduce the
+    EXACT global causal mask for every rank at every ring size up to the
+    node maximum R=8 — brute-force over global positions."""
+    import torch
+
+    s = 2 * R * c
+    full = ~torch.ones(s, s, dtype=torch.bool).triu(1)  # causal: q>=k
+
+    def positions(rank):
+        lo = torch.arange(rank * c, (rank + 1) * c)
+        hi0 = (2 * R - 1 - rank) * c
+        return torch.cat([lo, torch.arange(hi0, hi0 + c)])
+
+    for r in range(R):
+        qpos = positions(r)
+        got = torch.zeros(2 * c, s, dtype=torch.bool)
+        for t in range(R):
+            j = (r - t) % R
+            kpos = positions(j)
+            if j == r:
+                # causal over the local concat (positions are sorted)
+                blk = qpos.unsqueeze(1) >= kpos.unsqueeze(0)
+            elif j < r:
+                # low chunk fully visible, high chunk fully masked
+                blk = torch.zeros(2 * c, 2 * c, dtype=torch.bool)
+                blk[:, :c] = True
+            else:
+                # only the high-half queries see this block (fully)
+                blk = torch.zeros(2 * c, 2 * c, dtype=torch.bool)
+                blk[c:, :] = True
+            assert not (got[:, kpos] & blk).any()  # no double-compute
+            got[:, kpos] |= blk
+        assert torch.equal(got, full[qpos]), (R, c, r)
