@@ -46,7 +46,7 @@ def _ring_attn(rank, world):
     return 0.0
 
 
-@pytest.mark.parametrize("world", [2, 4])
+@pytest.mark.parametrize("world", [2, 4, 8])
 def test_ring_attention_matches_full(world):
     run_distributed(_ring_attn, world)
 
