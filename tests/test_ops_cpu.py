@@ -102,3 +102,26 @@ def test_flash_attn_sliding_window_cpu():
     assert torch.allclose(o, ref, atol=1e-4), (o - ref).abs().max()
     o.sum().backward()
     assert torch.isfinite(q.grad).all()
+
+
+def test_flash_attn_crosslen_cpu_bottom_right():
+    """CPU reference path with S_q != S_kv uses BOTTOM-RIGHT causal
+    alignment (query i sees keys j <= i + skv - sq) — the decode / ring
+    half-block convention the HIP kernel implements."""
+    import torch
+    from neuronx_distributed_training_amd.ops import flash_attn_func
+
+    torch.manual_seed(0)
+    b, h, sq, skv, d = 1, 2, 5, 9, 16
+    q = torch.randn(b, h, sq, d)
+    k = torch.randn(b, h, skv, d)
+    v = torch.randn(b, h, skv, d)
+    o = flash_attn_func(q, k, v, causal=True)
+    s = (q @ k.transpose(-1, -2)) / d ** 0.5
+    mask = torch.ones(sq, skv, dtype=torch.bool).triu(1 + skv - sq)
+    ref = torch.softmax(s.masked_fill(mask, float("-inf")), -1) @ v
+    assert torch.allclose(o, ref, atol=1e-5), (o - ref).abs().max()
+    # last query row sees ALL keys; first row sees skv-sq+1 of them
+    p = torch.softmax(s.masked_fill(mask, float("-inf")), -1)
+    assert (p[0, 0, -1] > 0).all()
+    assert int((p[0, 0, 0] > 0).sum()) == skv - sq + 1
