@@ -525,3 +525,28 @@ def test_flash_fwd_dbuf_matches_sbuf(ext):
     o1, l1 = ext.flash_attn_fwd(q, k, v, True, scale)
     o2, l2 = ext.flash_attn_fwd_sbuf(q, k, v, True, scale)
     assert torch.equal(o1, o2) and torch.equal(l1, l2)
+
+
+def test_rope_zigzag_offsets_gpu(ext):
+    """rope kernel's pos_offset2 branch (zigzag CP: the two halves of
+    the local sequence take different global offsets) vs the fp32 table
+    reference."""
+    from neuronx_distributed_training_amd.ops.rope import (
+        apply_rotary_pos_emb, build_rope_cache,
+    )
+
+    torch.manual_seed(17)
+    cos, sin = build_rope_cache(1024, 128, base=500000.0, device="cuda")
+    x = torch.randn(2, 4, 64, 128, device="cuda", dtype=torch.bfloat16)
+    off_lo, off_hi = 96, 896  # rank 3 of CP 8 with chunk 32
+    y = apply_rotary_pos_emb(x, cos, sin, (off_lo, off_hi))
+
+    def rotate_half(t):
+        t1, t2 = t.chunk(2, dim=-1)
+        return torch.cat((-t2, t1), dim=-1)
+
+    xf = x.float()
+    tab_c = torch.cat([cos[off_lo:off_lo + 32], cos[off_hi:off_hi + 32]])
+    tab_s = torch.cat([sin[off_lo:off_lo + 32], sin[off_hi:off_hi + 32]])
+    ref = xf * tab_c + rotate_half(xf) * tab_s
+    assert (y.float() - ref).abs().max() < 0.05
