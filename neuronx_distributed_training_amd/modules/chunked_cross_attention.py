@@ -8,8 +8,8 @@ causal shift of ``chunk_size − 1`` positions. Q is Column-parallel, fused
 KV over the retrieved states is Column-parallel (stride 2), the output
 projection is Row-parallel — same TP decomposition as self-attention.
 
-The attention itself runs per-chunk through torch SDPA (it is not the
-pretraining hot path; the fused flash kernel covers self-attention).
+The attention runs through the flash kernel's non-causal S_q != S_kv
+path on GPU (bf16, head_dim 64/128); torch SDPA otherwise.
 """
 
 from __future__ import annotations
@@ -101,7 +101,12 @@ class ParallelChunkedCrossAttention(nn.Module):
             .permute(2, 0, 3, 1, 4)
             .reshape(b * num_chunks, nh, r_tot, d)
         )
-        o = F.scaled_dot_product_attention(q, k, v, scale=self.scale)
+        if q.is_cuda and q.dtype == torch.bfloat16 and d in (64, 128):
+            from ..ops import flash_attn_func
+
+            o = flash_attn_func(q, k, v, causal=False, scale=self.scale)
+        else:
+            o = F.scaled_dot_product_attention(q, k, v, scale=self.scale)
         o = (
             o.view(b, num_chunks, nh, m, d)
             .permute(1, 3, 0, 2, 4)

@@ -550,3 +550,28 @@ def test_rope_zigzag_offsets_gpu(ext):
     tab_s = torch.cat([sin[off_lo:off_lo + 32], sin[off_hi:off_hi + 32]])
     ref = xf * tab_c + rotate_half(xf) * tab_s
     assert (y.float() - ref).abs().max() < 0.05
+
+
+def test_chunked_cross_attention_kernel_path_gpu(ext):
+    """RETRO chunked cross-attention dispatches to the non-causal
+    S_q != S_kv flash kernel on GPU — compare against the module's own
+    SDPA fallback in fp32."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.modules.chunked_cross_attention import (
+        ParallelChunkedCrossAttention,
+    )
+
+    ps.destroy_model_parallel()
+    ps.initialize_model_parallel()
+    torch.manual_seed(18)
+    mod = ParallelChunkedCrossAttention(
+        hidden_size=256, num_attention_heads=2, chunk_size=64,
+        dtype=torch.bfloat16, init_seed=1,
+    ).cuda()
+    s, b, nc, rt = 192, 2, 3, 96
+    hidden = torch.randn(s, b, 256, device="cuda", dtype=torch.bfloat16)
+    retrieved = torch.randn(nc, rt, b, 256, device="cuda",
+                            dtype=torch.bfloat16)
+    out = mod(hidden, retrieved)
+    ref = mod.float()(hidden.float(), retrieved.float())  # SDPA fallback
+    assert (out.float() - ref).abs().max() < 0.05
