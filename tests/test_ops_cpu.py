@@ -125,3 +125,28 @@ def test_flash_attn_crosslen_cpu_bottom_right():
     p = torch.softmax(s.masked_fill(mask, float("-inf")), -1)
     assert (p[0, 0, -1] > 0).all()
     assert int((p[0, 0, 0] > 0).sum()) == skv - sq + 1
+
+
+def test_flash_attn_noncausal_crosslen_cpu():
+    """Non-causal S_q != S_kv CPU path (chunked cross-attention / ring
+    half-blocks) against plain softmax attention."""
+    import torch
+    from neuronx_distributed_training_amd.ops import flash_attn_func
+
+    torch.manual_seed(1)
+    b, h, sq, skv, d = 2, 3, 7, 13, 8
+    q = torch.randn(b, h, sq, d, requires_grad=True)
+    k = torch.randn(b, h, skv, d, requires_grad=True)
+    v = torch.randn(b, h, skv, d, requires_grad=True)
+    o = flash_attn_func(q, k, v, causal=False)
+    s = (q @ k.transpose(-1, -2)) / d ** 0.5
+    ref = torch.softmax(s, -1) @ v
+    assert torch.allclose(o, ref, atol=1e-5)
+    g = torch.randn_like(o)
+    o.backward(g)
+    gq = q.grad.clone()
+    q.grad = None
+    qr = q.detach().clone().requires_grad_(True)
+    s2 = (qr @ k.detach().transpose(-1, -2)) / d ** 0.5
+    (torch.softmax(s2, -1) @ v.detach()).backward(g)
+    assert torch.allclose(gq, qr.grad, atol=1e-5)
