@@ -282,3 +282,38 @@ def test_zigzag_ring_case_table(R, c):
             assert not (got[:, kpos] & blk).any()  # no double-compute
             got[:, kpos] |= blk
         assert torch.equal(got, full[qpos]), (R, c, r)
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    sq=st.integers(min_value=1, max_value=24),
+    extra=st.integers(min_value=0, max_value=24),
+    window=st.integers(min_value=0, max_value=30),
+    causal=st.booleans(),
+)
+def test_flash_mask_properties(sq, extra, window, causal):
+    """_make_mask (the CPU reference's masking, mirrored by the HIP
+    kernels): bottom-right alignment invariants for any S_q <= S_kv,
+    window and causality — checked against a brute-force definition."""
+    import torch
+    from neuronx_distributed_training_amd.ops.flash_attn import _make_mask
+
+    skv = sq + extra
+    m = _make_mask(sq, skv, causal, window, torch.device("cpu"))
+    diag = skv - sq
+    for i in range(sq):
+        for j in range(skv):
+            dead = False
+            if causal:
+                dead = j > i + diag
+                if window > 0:
+                    dead = dead or (j <= i + diag - window)
+            assert bool(m[i, j]) == dead, (i, j, sq, skv, window, causal)
+    if causal:
+        # every query sees at least its aligned key (window >= 1 case)
+        if window != 0:
+            assert not bool(m[sq - 1, skv - 1])
+        # the last query row sees all of the last min(window or skv, skv) keys
+        alive_last = (~m[sq - 1]).sum()
+        expect = min(window, skv) if window > 0 else skv
+        assert int(alive_last) == expect
