@@ -421,9 +421,7 @@ class ZeRO1AdamW:
         self.step_count += 1
         b1, b2 = self.betas
         t = self.step_count
-        from ..ops import _try_load
-
-        k = _try_load() if shard.is_cuda else None
+        k = self._kernel_for(shard)
         shard_slice = self.param_flat[
             self.shard_start : self.shard_start + self.shard_size
         ]
@@ -479,6 +477,13 @@ class ZeRO1AdamW:
         elif not wrote_params:
             self.param_flat.copy_(self.master_shard.to(self.model_dtype))
         return gnorm
+
+    def _kernel_for(self, shard):
+        """Fused-kernel lookup — a seam so CPU tests can inject a mock
+        and exercise the wrote_params/in-place-all-gather step path."""
+        from ..ops import _try_load
+
+        return _try_load() if shard.is_cuda else None
 
     # -- LR schedule hook --
     def set_lr(self, lr: float):

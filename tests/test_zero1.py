@@ -192,3 +192,59 @@ def test_overlap_grad_reduce_tp2_dp2():
     over = [x for x in run_distributed(_overlap_tp_dp, 4, True) if x is not None][0]
     assert torch.allclose(plain[0], over[0], atol=1e-5), (plain[0] - over[0]).abs().max()
     assert torch.allclose(plain[1], over[1], atol=1e-5), (plain[1] - over[1]).abs().max()
+
+
+class _FakeKern:
+    """CPU mirror of adamw.hip's fused step (incl. the in-place model-
+    dtype param write) for gate-injection tests."""
+
+    @staticmethod
+    def adamw_step(p, g, m, v, wd_mask, lr, b1, b2, eps, wd, t,
+                   grad_scale=None, p_bf16=None):
+        gg = g * grad_scale if grad_scale is not None else g
+        m.mul_(b1).add_(gg, alpha=1 - b1)
+        v.mul_(b2).addcmul_(gg, gg, value=1 - b2)
+        denom = (v / (1 - b2 ** t)).sqrt_().add_(eps)
+        p.mul_(torch.where(wd_mask, torch.tensor(1.0 - lr * wd),
+                           torch.tensor(1.0)))
+        p.addcdiv_(m, denom, value=-(lr / (1 - b1 ** t)))
+        if p_bf16 is not None:
+            p_bf16.copy_(p.to(p_bf16.dtype))
+
+
+def _zero1_fused_path(rank, world):
+    """The fused-kernel step path (wrote_params=True): at DP>1 the
+    all-gather runs IN PLACE from the shard slice of param_flat — the
+    branch only real multi-GPU RCCL would otherwise reach. A mock kernel
+    injected through the _kernel_for seam runs it on gloo and the result
+    must match the plain (no-kernel) path."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.optim.zero1 import ZeRO1AdamW
+
+    class _Opt(ZeRO1AdamW):
+        def _kernel_for(self, shard):
+            return _FakeKern
+
+    ps.initialize_model_parallel()
+    m = _make_model()
+    opt = _Opt(list(m.named_parameters()), lr=1e-2, grad_clip=1.0)
+    torch.manual_seed(50)
+    for step in range(3):
+        xfull = torch.randn(8, 16)
+        yfull = torch.randn(8, 4)
+        per = 8 // world
+        x = xfull[rank * per : (rank + 1) * per]
+        y = yfull[rank * per : (rank + 1) * per]
+        opt.zero_grad()
+        ((m(x) - y) ** 2).mean().backward()
+        opt.step()
+    return torch.cat([p.detach().reshape(-1) for p in m.parameters()])
+
+
+def test_zero1_fused_kernel_path_matches_plain():
+    plain = run_distributed(_zero1_dp, 2)
+    fused = run_distributed(_zero1_fused_path, 2)
+    assert torch.allclose(plain[0], fused[0], atol=1e-5), (
+        (plain[0] - fused[0]).abs().max()
+    )
+    assert torch.allclose(fused[0], fused[1])
