@@ -357,3 +357,51 @@ def test_cp2_tp2_sp_full_ckpt_matches_single():
     l4 = run_distributed(_cp_sp_ckpt_train, 4)
     assert max(abs(l - l4[0]) for l in l4) < 1e-6
     assert abs(l1 - l4[0]) < 0.05, (l1, l4[0])
+
+
+def _mixtral_ep_cp_loss(rank, world):
+    """EP2 x CP2 on 4 ranks: expert-parallel all-to-all under the zigzag
+    CP token split — loss matches single-rank."""
+    from neuronx_distributed_training_amd.parallel import state as ps
+    from neuronx_distributed_training_amd.trainer.module_mixtral import (
+        MixtralModule,
+    )
+
+    ep = 2 if world == 4 else 1
+    cp = 2 if world == 4 else 1
+    ps.initialize_model_parallel(expert_model_parallel_size=ep,
+                                 context_parallel_size=cp)
+    dp = ps.get_data_parallel_world_size()
+    cfg = {
+        "data": {"global_batch_size": 4, "micro_batch_size": 2, "seq_length": 32},
+        "distributed_strategy": {"expert_model_parallel_size": ep,
+                                 "context_parallel_size": cp},
+        "model": {
+            "vocab_size": 128, "hidden_size": 64, "intermediate_size": 128,
+            "num_layers": 2, "num_attention_heads": 4, "num_kv_heads": 2,
+            "moe": {"num_experts": 4, "top_k": 2, "aux_loss_coef": 0.02},
+            "grad_clip": 1.0, "optim": {"lr": 1e-3, "sched": {"warmup_steps": 1}},
+        },
+        "precision": {"type": "fp32"},
+        "exp_manager": {},
+    }
+    torch.manual_seed(3)
+    mod = MixtralModule(cfg)
+    mod.setup()
+    mod.configure_optimizers(max_steps=10)
+    g = torch.Generator().manual_seed(5)
+    glob = [torch.randint(0, 128, (2, 32), generator=g) for _ in range(2)]
+    r = ps.get_data_parallel_rank()
+    if dp == 2:
+        micros = [{"input_ids": glob[r], "labels": glob[r].clone()}]
+    else:
+        micros = [{"input_ids": x, "labels": x.clone()} for x in glob]
+    m = mod.training_step(micros)
+    return m["reduced_train_loss"]
+
+
+def test_mixtral_ep2_cp2_loss_matches_single():
+    l1 = run_distributed(_mixtral_ep_cp_loss, 1)[0]
+    l4 = run_distributed(_mixtral_ep_cp_loss, 4)
+    assert max(abs(l - l4[0]) for l in l4) < 1e-6
+    assert abs(l1 - l4[0]) < 0.05, (l1, l4[0])
