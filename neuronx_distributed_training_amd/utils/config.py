@@ -106,8 +106,13 @@ def validate_config(cfg: Dict):
         if moe.get("dropless") and moe.get("capacity_factor", 0):
             raise ValueError("dropless MoE excludes capacity_factor")
     seq = int(d.get("seq_length", 1))
-    if cp > 1 and seq % cp != 0:
-        raise ValueError(f"seq_length {seq} % context_parallel {cp} != 0")
+    if cp > 1 and seq % (2 * cp) != 0:
+        # zigzag CP placement: each rank holds TWO global chunks of
+        # seq/(2*cp) (parallel/cp.py), so 2*cp must divide the sequence
+        raise ValueError(
+            f"seq_length {seq} must be divisible by 2*context_parallel "
+            f"(= {2 * cp}) for the zigzag CP layout"
+        )
 
 
 def derive_env(cfg: Dict):

@@ -68,3 +68,21 @@ def test_mistral_recipe_builds():
     ids = torch.randint(0, 128, (2, 32), generator=torch.Generator().manual_seed(1))
     m = mod.training_step([{"input_ids": ids, "labels": ids.clone()}])
     assert m["reduced_train_loss"] == m["reduced_train_loss"]
+
+
+def test_seq_must_divide_2cp():
+    """Zigzag CP needs seq % (2*cp) == 0 — seq divisible by cp alone is
+    rejected (it would produce ragged torch.chunk splits)."""
+    import pytest
+    from neuronx_distributed_training_amd.utils.config import validate_config
+
+    cfg = {
+        "data": {"global_batch_size": 2, "micro_batch_size": 1,
+                 "seq_length": 6},
+        "distributed_strategy": {"context_parallel_size": 2},
+        "model": {"num_layers": 2},
+    }
+    with pytest.raises(ValueError, match="2\\*context_parallel"):
+        validate_config(cfg)
+    cfg["data"]["seq_length"] = 8
+    validate_config(cfg)  # ok
