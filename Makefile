@@ -1,7 +1,7 @@
 # Convenience targets (see docs/general/installation_guide.md)
 PY ?= python
 
-.PHONY: build test test-gpu bench smoke parity clean
+.PHONY: build test test-gpu bench smoke parity bench-attn bench-decode family-smoke clean
 
 build:
 	PYTORCH_ROCM_ARCH=gfx950 $(PY) setup.py build_ext --inplace
@@ -21,6 +21,15 @@ smoke:
 parity:
 	$(PY) tools/check_grad_parity.py --layout tp2_sp
 	$(PY) tools/check_grad_parity.py --layout lora_tp2_sp
+
+bench-attn:
+	$(PY) tools/bench_attn_kernels.py --iters 10
+
+bench-decode:
+	$(PY) tools/bench_decode.py
+
+family-smoke:
+	$(PY) tools/gpu_family_smoke.py
 
 clean:
 	rm -rf build neuronx_distributed_training_amd/ops/csrc/*.o
